@@ -1,0 +1,57 @@
+#!/usr/bin/env python3
+"""Data-parallel training entrypoint (strategy 1).
+
+CLI/CSV parity with the reference's ``data_parallel_train.py`` (flags
+``--world_size --epochs --sample_size``, per-worker CSVs + combined CSV in
+``data_parallel_logs/``), rebuilt on the MI355X-native engine:
+one process per GPU, bf16 bucketed RCCL all-reduce over xGMI, gfx950 HIP
+kernels for the ResNet hot path.  On CPU-only hosts it runs the gloo
+plumbing configuration (BASELINE.json config #1).
+"""
+from __future__ import annotations
+
+import argparse
+
+from horizonml_amd.engine.dp import dp_worker
+from horizonml_amd.runtime.launcher import run_workers
+
+
+def run_data_parallel(world_size: int, epochs: int, sample_size: int,
+                      logs_dir: str = "data_parallel_logs",
+                      batch_size: int = 64, model_name: str = "resnet18",
+                      backend=None, synthetic=None, lr: float = 1e-3,
+                      optimizer_name: str = "adam"):
+    """Launcher parity with reference ``run_data_parallel``
+    (``data_parallel_train.py:233-291``). Returns the combined DataFrame."""
+    return run_workers(dp_worker, world_size, epochs, sample_size, logs_dir,
+                       timeout_base=120,
+                       extra_args=(batch_size, model_name, backend, synthetic,
+                                   lr, optimizer_name))
+
+
+def main():
+    ap = argparse.ArgumentParser(description="Data-parallel training")
+    ap.add_argument("--world_size", type=int, default=5)
+    ap.add_argument("--epochs", type=int, default=5)
+    ap.add_argument("--sample_size", type=int, default=1000)
+    ap.add_argument("--logs_dir", type=str, default="data_parallel_logs")
+    ap.add_argument("--batch_size", type=int, default=64)
+    ap.add_argument("--model", type=str, default="resnet18")
+    ap.add_argument("--backend", type=str, default=None,
+                    choices=[None, "nccl", "gloo"], nargs="?")
+    ap.add_argument("--synthetic", action="store_true", default=None,
+                    help="force synthetic CIFAR-shaped data")
+    ap.add_argument("--lr", type=float, default=1e-3)
+    ap.add_argument("--optimizer", type=str, default="adam",
+                    choices=["adam", "sgd"])
+    args = ap.parse_args()
+    df = run_data_parallel(args.world_size, args.epochs, args.sample_size,
+                           args.logs_dir, args.batch_size, args.model,
+                           args.backend, args.synthetic, args.lr,
+                           args.optimizer)
+    if df is not None:
+        print(df.tail(args.world_size).to_string(index=False))
+
+
+if __name__ == "__main__":
+    main()

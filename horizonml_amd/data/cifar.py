@@ -1,0 +1,122 @@
+"""CIFAR-10 data layer.
+
+The reference downloads CIFAR-10 via torchvision with a rank-0 gate
+(``data_parallel_train.py:51-55``) and subsamples it with an unseeded
+per-rank randperm.  Here:
+
+* default is **synthetic CIFAR-shaped data** (BASELINE.json requires it —
+  there is no network on the target boxes): uint8 images generated once from
+  a fixed seed, normalized on the fly with the reference's (0.5,0.5,0.5)
+  mean/std (``data_parallel_train.py:44-47``);
+* if a real CIFAR-10 python-pickle directory exists (``cifar-10-batches-py``)
+  it is used instead — no download is ever attempted;
+* the random subset is derived from a *shared* seed so every rank sees the
+  same subset (Q1 fix), and DP shards it with ``DistributedSampler``.
+"""
+from __future__ import annotations
+
+import os
+import pickle
+from typing import Optional, Tuple
+
+import numpy as np
+import torch
+from torch.utils.data import DataLoader, Dataset, Subset
+from torch.utils.data.distributed import DistributedSampler
+
+from ..utils.seed import DEFAULT_SEED, shared_subset_indices
+
+_MEAN = 0.5
+_STD = 0.5
+
+
+class SyntheticCIFAR10(Dataset):
+    """Deterministic random CIFAR-shaped dataset (uint8 in memory)."""
+
+    def __init__(self, n: int = 50000, num_classes: int = 10,
+                 seed: int = DEFAULT_SEED):
+        g = torch.Generator().manual_seed(seed)
+        self.images = torch.randint(0, 256, (n, 3, 32, 32),
+                                    dtype=torch.uint8, generator=g)
+        self.labels = torch.randint(0, num_classes, (n,),
+                                    dtype=torch.long, generator=g)
+
+    def __len__(self):
+        return self.images.shape[0]
+
+    def __getitem__(self, i: int):
+        x = self.images[i].to(torch.float32).div_(255.0).sub_(_MEAN).div_(_STD)
+        return x, self.labels[i]
+
+
+class CIFAR10Local(Dataset):
+    """Real CIFAR-10 from a local ``cifar-10-batches-py`` directory."""
+
+    def __init__(self, root: str, train: bool = True):
+        base = root
+        if os.path.isdir(os.path.join(root, "cifar-10-batches-py")):
+            base = os.path.join(root, "cifar-10-batches-py")
+        files = ([f"data_batch_{i}" for i in range(1, 6)] if train
+                 else ["test_batch"])
+        imgs, labels = [], []
+        for f in files:
+            path = os.path.join(base, f)
+            with open(path, "rb") as fh:
+                d = pickle.load(fh, encoding="bytes")
+            imgs.append(np.asarray(d[b"data"], dtype=np.uint8))
+            labels.extend(d[b"labels"])
+        data = np.concatenate(imgs).reshape(-1, 3, 32, 32)
+        self.images = torch.from_numpy(data)
+        self.labels = torch.tensor(labels, dtype=torch.long)
+
+    def __len__(self):
+        return self.images.shape[0]
+
+    def __getitem__(self, i: int):
+        x = self.images[i].to(torch.float32).div_(255.0).sub_(_MEAN).div_(_STD)
+        return x, self.labels[i]
+
+
+def build_dataset(data_dir: str = "./data", synthetic: Optional[bool] = None,
+                  n: int = 50000, seed: int = DEFAULT_SEED) -> Dataset:
+    """Real CIFAR-10 if present under ``data_dir`` (unless ``synthetic=True``),
+    else synthetic."""
+    if synthetic is not True:
+        for cand in (data_dir, os.path.join(data_dir, "cifar-10-batches-py")):
+            if os.path.isfile(os.path.join(cand, "data_batch_1")):
+                return CIFAR10Local(data_dir, train=True)
+        if synthetic is False:
+            raise FileNotFoundError(
+                f"no CIFAR-10 batches under {data_dir!r} and synthetic=False")
+    return SyntheticCIFAR10(n=n, seed=seed)
+
+
+def get_dataloader(rank: int, world_size: int, batch_size: int = 64,
+                   sample_size: int = 1000, strategy: str = "dp",
+                   data_dir: str = "./data", synthetic: Optional[bool] = None,
+                   seed: int = DEFAULT_SEED,
+                   drop_last: bool = False) -> Tuple[DataLoader, Optional[DistributedSampler]]:
+    """Reference-parity dataloader.
+
+    * ``dp``: shared random subset + DistributedSampler shard per rank
+      (``data_parallel_train.py:59-71``).
+    * ``mp``/``tp``: every rank iterates the *same* full subset in the same
+      order (the reference intends this but breaks it — Q1; fixed here),
+      ``shuffle=False`` like ``layer_model_parallel_train.py:103-131``.
+    """
+    ds = build_dataset(data_dir, synthetic,
+                       n=max(50000, sample_size), seed=seed)
+    idx = shared_subset_indices(len(ds), sample_size, seed=seed)
+    subset = Subset(ds, idx.tolist())
+    if strategy == "dp":
+        sampler = DistributedSampler(subset, num_replicas=world_size,
+                                     rank=rank, shuffle=True, seed=seed,
+                                     drop_last=drop_last)
+        loader = DataLoader(subset, batch_size=batch_size, sampler=sampler,
+                            num_workers=0, pin_memory=torch.cuda.is_available(),
+                            drop_last=drop_last)
+        return loader, sampler
+    loader = DataLoader(subset, batch_size=batch_size, shuffle=False,
+                        num_workers=0, pin_memory=torch.cuda.is_available(),
+                        drop_last=drop_last)
+    return loader, None

@@ -1,0 +1,85 @@
+"""Shared training-loop machinery for all three strategy engines."""
+from __future__ import annotations
+
+from typing import Iterable, List, Optional
+
+import torch
+
+
+class Meters:
+    """Loss/accuracy accumulators that stay on-device until epoch end (the
+    reference's per-step ``.item()`` reads would force a hipStreamSynchronize
+    every step on GPU)."""
+
+    def __init__(self, device: Optional[torch.device]):
+        dev = device if device is not None else torch.device("cpu")
+        self.loss_sum = torch.zeros((), dtype=torch.float32, device=dev)
+        self.correct = torch.zeros((), dtype=torch.float32, device=dev)
+        self.count = 0
+
+    def update(self, loss: torch.Tensor, logits: torch.Tensor,
+               labels: torch.Tensor):
+        bs = labels.shape[0]
+        self.loss_sum += loss.detach() * bs
+        self.correct += (logits.detach().argmax(1) == labels).sum()
+        self.count += bs
+
+    def epoch_values(self):
+        n = max(1, self.count)
+        loss = (self.loss_sum / n).item()
+        acc = 100.0 * (self.correct / n).item()
+        self.loss_sum.zero_()
+        self.correct.zero_()
+        self.count = 0
+        return loss, acc
+
+
+class GradDivergenceProbe:
+    """‖g_t − g_{t−1}‖₂ over all gradients (reference
+    ``data_parallel_train.py:132-145``; kernel K11 in SURVEY.md §2.4).
+
+    Keeps the previous flat gradient on-device; per-step results accumulate
+    on-device and are read once per epoch (no per-step sync).
+    """
+
+    def __init__(self, params: Iterable[torch.Tensor]):
+        self.params: List[torch.Tensor] = [p for p in params if p.requires_grad]
+        if not self.params:
+            raise ValueError("no trainable params for divergence probe")
+        dev = self.params[0].device
+        total = sum(p.numel() for p in self.params)
+        self.prev = torch.zeros(total, dtype=torch.float32, device=dev)
+        self.sum = torch.zeros((), dtype=torch.float32, device=dev)
+        self.n = 0
+        self.first = True
+
+    @torch.no_grad()
+    def step(self):
+        grads = [(p.grad if p.grad is not None
+                  else torch.zeros_like(p)).flatten().float()
+                 for p in self.params]
+        flat = torch.cat(grads)
+        if not self.first:
+            self.sum += torch.linalg.vector_norm(flat - self.prev)
+            self.n += 1
+        self.prev.copy_(flat)
+        self.first = False
+
+    def epoch_value(self) -> float:
+        v = (self.sum / max(1, self.n)).item()
+        self.sum.zero_()
+        self.n = 0
+        return v
+
+
+def build_optimizer(params, name: str = "adam", lr: float = 1e-3,
+                    momentum: float = 0.9, weight_decay: float = 0.0):
+    """Reference default: Adam(lr=1e-3) (``data_parallel_train.py:205``).
+    The north star also names the SGD step — both supported."""
+    name = name.lower()
+    if name == "adam":
+        return torch.optim.Adam(params, lr=lr, weight_decay=weight_decay)
+    if name == "sgd":
+        return torch.optim.SGD(params, lr=lr, momentum=momentum,
+                               weight_decay=weight_decay)
+    raise ValueError(f"unknown optimizer {name!r}")
