@@ -90,7 +90,11 @@ def train_dp(ctx: DistContext, epochs: int, sample_size: int, logs_dir: str,
             with prof.compute():
                 optimizer.zero_grad(set_to_none=False)
                 logits = ddp(x)
-                loss = F.cross_entropy(logits.float(), y)
+                if logits.is_cuda:
+                    from ..models._functional_gpu import cross_entropy
+                    loss = cross_entropy(logits, y)
+                else:
+                    loss = F.cross_entropy(logits.float(), y)
                 loss.backward()
             with prof.comm():
                 ddp.finalize_backward()

@@ -1,0 +1,366 @@
+// Torch bindings for the horizonml_amd gfx950 kernel set.
+// Compiled host-side (g++); all device code lives in *.hip translation units
+// linked as extra objects (no hipify, no CUDA compat).
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+
+#include <vector>
+
+struct ConvP {
+  int Nb, H, W, C, K;
+  int Ho, Wo, R, S;
+  int str, pad;
+  int M, Kd;
+};
+
+extern "C" {
+void launch_conv_fwd(const void*, const void*, void*, float*, ConvP,
+                     hipStream_t);
+void launch_conv_dgrad(const void*, const void*, void*, ConvP, hipStream_t);
+void launch_gemm_bf16(const void*, const void*, void*, int, int, int,
+                      hipStream_t);
+void launch_wgrad(const void*, const void*, float*, ConvP, hipStream_t);
+void launch_bn_apply(const void*, const void*, void*, const float*,
+                     const float*, const float*, float*, float*, float*,
+                     float*, long, int, float, float, int, int, hipStream_t);
+void launch_bnact_bwd_reduce(const void*, const void*, const void*,
+                             const float*, const float*, float*, float*, long,
+                             int, int, hipStream_t);
+void launch_bn_bwd_apply(const void*, const void*, const void*, const float*,
+                         const float*, const float*, const float*,
+                         const float*, void*, void*, long, int, int,
+                         hipStream_t);
+void launch_maxpool_fwd(const void*, void*, unsigned char*, int, int, int,
+                        int, int, int, hipStream_t);
+void launch_maxpool_bwd(const void*, const unsigned char*, void*, int, int,
+                        int, int, int, int, hipStream_t);
+void launch_avgpool_fwd(const void*, void*, int, int, int, hipStream_t);
+void launch_avgpool_bwd(const void*, void*, int, int, int, hipStream_t);
+void launch_linear_fwd(const void*, const void*, const float*, float*, int,
+                       int, int, hipStream_t);
+void launch_linear_bwd(const float*, const void*, const void*, void*, float*,
+                       float*, int, int, int, hipStream_t);
+void launch_ce_fwd_bwd(const float*, const long*, float*, float*, int, int,
+                       hipStream_t);
+void launch_adam_step(float*, float*, float*, float*, void*, const float*,
+                      long, float, float, float, float, float, int,
+                      hipStream_t);
+void launch_sgd_step(float*, float*, float*, void*, long, float, float, float,
+                     int, hipStream_t);
+void launch_permute_krsc_rsck(const void*, void*, const int*, int, int,
+                              hipStream_t);
+void launch_grad_divergence(const float*, float*, float*, float*, long, int,
+                            hipStream_t);
+}
+
+namespace {
+
+using torch::Tensor;
+
+hipStream_t cur_stream() { return c10::hip::getCurrentHIPStream().stream(); }
+
+void check_cl(const Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.scalar_type() == torch::kBFloat16, name, " must be bf16");
+  TORCH_CHECK(t.dim() == 4, name, " must be 4-D");
+  bool cl = t.is_contiguous(at::MemoryFormat::ChannelsLast);
+  bool hw1 = t.size(2) == 1 && t.size(3) == 1 && t.is_contiguous();
+  TORCH_CHECK(cl || hw1, name, " must be channels_last contiguous");
+}
+
+void check_f32(const Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda() && t.scalar_type() == torch::kFloat32 &&
+                  t.is_contiguous(),
+              name, " must be contiguous f32 on GPU");
+}
+
+Tensor empty_cl_bf16(int64_t n, int64_t c, int64_t h, int64_t w,
+                     const Tensor& like) {
+  return at::empty({n, c, h, w},
+                   like.options().dtype(torch::kBFloat16),
+                   at::MemoryFormat::ChannelsLast);
+}
+
+ConvP make_convp(const Tensor& x, int K, int R, int S, int stride, int pad) {
+  ConvP p{};
+  p.Nb = (int)x.size(0);
+  p.C = (int)x.size(1);
+  p.H = (int)x.size(2);
+  p.W = (int)x.size(3);
+  p.K = K;
+  p.R = R;
+  p.S = S;
+  p.str = stride;
+  p.pad = pad;
+  p.Ho = (p.H + 2 * pad - R) / stride + 1;
+  p.Wo = (p.W + 2 * pad - S) / stride + 1;
+  p.M = p.Nb * p.Ho * p.Wo;
+  p.Kd = R * S * p.C;
+  return p;
+}
+
+// ---------------------------------------------------------------- conv+BN --
+std::vector<Tensor> conv_bn_act_fwd(
+    Tensor x, Tensor w, Tensor gamma, Tensor beta, Tensor running_mean,
+    Tensor running_var, int64_t stride, int64_t pad, double momentum,
+    double eps, bool training, bool act,
+    c10::optional<Tensor> residual) {
+  check_cl(x, "x");
+  TORCH_CHECK(w.is_cuda() && w.scalar_type() == torch::kBFloat16 &&
+                  w.is_contiguous() && w.dim() == 4,
+              "w must be contiguous bf16 KRSC");
+  check_f32(gamma, "gamma");
+  check_f32(beta, "beta");
+  int K = (int)w.size(0), R = (int)w.size(1), S = (int)w.size(2);
+  TORCH_CHECK(w.size(3) == x.size(1), "w in-channels mismatch");
+  ConvP p = make_convp(x, K, R, S, (int)stride, (int)pad);
+
+  Tensor convout = empty_cl_bf16(p.Nb, K, p.Ho, p.Wo, x);
+  Tensor y = empty_cl_bf16(p.Nb, K, p.Ho, p.Wo, x);
+  auto fopt = x.options().dtype(torch::kFloat32);
+  Tensor stats, smean, sinvstd;
+  float* stats_ptr = nullptr;
+  if (training) {
+    stats = at::zeros({2L * K}, fopt);
+    smean = at::empty({K}, fopt);
+    sinvstd = at::empty({K}, fopt);
+    stats_ptr = stats.data_ptr<float>();
+  } else {
+    smean = at::empty({0}, fopt);
+    sinvstd = at::empty({0}, fopt);
+  }
+  const void* res_ptr = nullptr;
+  if (residual.has_value()) {
+    check_cl(*residual, "residual");
+    res_ptr = residual->data_ptr();
+  }
+  auto st = cur_stream();
+  launch_conv_fwd(x.data_ptr(), w.data_ptr(), convout.data_ptr(), stats_ptr,
+                  p, st);
+  launch_bn_apply(convout.data_ptr(), res_ptr, y.data_ptr(), stats_ptr,
+                  gamma.data_ptr<float>(), beta.data_ptr<float>(),
+                  running_mean.data_ptr<float>(),
+                  running_var.data_ptr<float>(),
+                  training ? smean.data_ptr<float>() : nullptr,
+                  training ? sinvstd.data_ptr<float>() : nullptr,
+                  (long)p.M, K, (float)momentum, (float)eps, training ? 1 : 0,
+                  act ? 1 : 0, st);
+  return {y, convout, smean, sinvstd};
+}
+
+std::vector<Tensor> conv_bn_act_bwd(
+    Tensor dy, Tensor y, Tensor x, Tensor w, Tensor w_rsck, Tensor convout,
+    Tensor gamma, Tensor save_mean, Tensor save_invstd, int64_t stride,
+    int64_t pad, bool act, bool need_dx, bool has_res) {
+  check_cl(dy, "dy");
+  check_cl(y, "y");
+  check_cl(x, "x");
+  check_cl(convout, "convout");
+  int K = (int)w.size(0), R = (int)w.size(1), S = (int)w.size(2);
+  ConvP p = make_convp(x, K, R, S, (int)stride, (int)pad);
+  auto fopt = x.options().dtype(torch::kFloat32);
+  auto st = cur_stream();
+
+  Tensor sum_dz = at::empty({K}, fopt);
+  Tensor sum_dzx = at::empty({K}, fopt);
+  launch_bnact_bwd_reduce(dy.data_ptr(), y.data_ptr(), convout.data_ptr(),
+                          save_mean.data_ptr<float>(),
+                          save_invstd.data_ptr<float>(),
+                          sum_dz.data_ptr<float>(), sum_dzx.data_ptr<float>(),
+                          (long)p.M, K, act ? 1 : 0, st);
+
+  Tensor dconv = empty_cl_bf16(p.Nb, K, p.Ho, p.Wo, x);
+  Tensor dres;
+  void* dres_ptr = nullptr;
+  if (has_res) {
+    dres = empty_cl_bf16(p.Nb, K, p.Ho, p.Wo, x);
+    dres_ptr = dres.data_ptr();
+  }
+  launch_bn_bwd_apply(dy.data_ptr(), y.data_ptr(), convout.data_ptr(),
+                      save_mean.data_ptr<float>(),
+                      save_invstd.data_ptr<float>(), gamma.data_ptr<float>(),
+                      sum_dz.data_ptr<float>(), sum_dzx.data_ptr<float>(),
+                      dconv.data_ptr(), dres_ptr, (long)p.M, K, act ? 1 : 0,
+                      st);
+
+  Tensor dw = at::zeros({(int64_t)K, R, S, (int64_t)p.C}, fopt);
+  launch_wgrad(x.data_ptr(), dconv.data_ptr(), dw.data_ptr<float>(), p, st);
+
+  Tensor dx;
+  if (need_dx) {
+    TORCH_CHECK(w_rsck.is_contiguous() &&
+                    w_rsck.scalar_type() == torch::kBFloat16,
+                "w_rsck must be contiguous bf16");
+    dx = empty_cl_bf16(p.Nb, p.C, p.H, p.W, x);
+    ConvP pd = p;
+    pd.M = p.Nb * p.H * p.W;
+    pd.Kd = R * S * K;
+    launch_conv_dgrad(dconv.data_ptr(), w_rsck.data_ptr(), dx.data_ptr(), pd,
+                      st);
+  }
+  // dgamma = Σ dz·xhat, dbeta = Σ dz
+  return {dx, dw, sum_dzx, sum_dz, dres};
+}
+
+// ----------------------------------------------------------------- pools --
+std::vector<Tensor> maxpool_fwd(Tensor x) {
+  check_cl(x, "x");
+  int Nb = (int)x.size(0), C = (int)x.size(1), H = (int)x.size(2),
+      W = (int)x.size(3);
+  int Hp = (H + 2 - 3) / 2 + 1, Wp = (W + 2 - 3) / 2 + 1;
+  Tensor y = empty_cl_bf16(Nb, C, Hp, Wp, x);
+  Tensor idx = at::empty({Nb, Hp, Wp, C}, x.options().dtype(torch::kUInt8));
+  launch_maxpool_fwd(x.data_ptr(), y.data_ptr(), idx.data_ptr<unsigned char>(),
+                     Nb, H, W, C, Hp, Wp, cur_stream());
+  return {y, idx};
+}
+
+Tensor maxpool_bwd(Tensor dy, Tensor idx, int64_t H, int64_t W) {
+  check_cl(dy, "dy");
+  int Nb = (int)dy.size(0), C = (int)dy.size(1), Hp = (int)dy.size(2),
+      Wp = (int)dy.size(3);
+  Tensor dx = empty_cl_bf16(Nb, C, H, W, dy);
+  launch_maxpool_bwd(dy.data_ptr(), idx.data_ptr<unsigned char>(),
+                     dx.data_ptr(), Nb, (int)H, (int)W, C, Hp, Wp,
+                     cur_stream());
+  return dx;
+}
+
+Tensor avgpool_fwd(Tensor x) {
+  check_cl(x, "x");
+  int Nb = (int)x.size(0), C = (int)x.size(1);
+  int HW = (int)(x.size(2) * x.size(3));
+  Tensor y = at::empty({Nb, C}, x.options());
+  launch_avgpool_fwd(x.data_ptr(), y.data_ptr(), Nb, HW, C, cur_stream());
+  return y;
+}
+
+Tensor avgpool_bwd(Tensor dy, int64_t H, int64_t W) {
+  TORCH_CHECK(dy.is_cuda() && dy.is_contiguous() &&
+              dy.scalar_type() == torch::kBFloat16);
+  int Nb = (int)dy.size(0), C = (int)dy.size(1);
+  Tensor dx = empty_cl_bf16(Nb, C, H, W, dy);
+  launch_avgpool_bwd(dy.data_ptr(), dx.data_ptr(), Nb, (int)(H * W), C,
+                     cur_stream());
+  return dx;
+}
+
+// -------------------------------------------------------------- classifier --
+Tensor linear_fwd(Tensor x, Tensor w, c10::optional<Tensor> bias) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() &&
+              x.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(w.is_contiguous() && w.scalar_type() == torch::kBFloat16);
+  int B = (int)x.size(0), In = (int)x.size(1), Out = (int)w.size(0);
+  Tensor y = at::empty({B, Out}, x.options().dtype(torch::kFloat32));
+  const float* bptr = nullptr;
+  if (bias.has_value()) {
+    check_f32(*bias, "bias");
+    bptr = bias->data_ptr<float>();
+  }
+  launch_linear_fwd(x.data_ptr(), w.data_ptr(), bptr, y.data_ptr<float>(), B,
+                    In, Out, cur_stream());
+  return y;
+}
+
+std::vector<Tensor> linear_bwd(Tensor dy, Tensor x, Tensor w, bool need_dx,
+                               bool need_db) {
+  check_f32(dy, "dy");
+  int B = (int)x.size(0), In = (int)x.size(1), Out = (int)w.size(0);
+  auto fopt = x.options().dtype(torch::kFloat32);
+  Tensor dx, dw = at::empty({Out, In}, fopt);
+  Tensor db = need_db ? at::empty({Out}, fopt) : Tensor();
+  void* dx_ptr = nullptr;
+  if (need_dx) {
+    dx = at::empty({B, In}, x.options());
+    dx_ptr = dx.data_ptr();
+  }
+  launch_linear_bwd(dy.data_ptr<float>(), x.data_ptr(), w.data_ptr(), dx_ptr,
+                    dw.data_ptr<float>(),
+                    need_db ? db.data_ptr<float>() : nullptr, B, In, Out,
+                    cur_stream());
+  return {dx, dw, db};
+}
+
+std::vector<Tensor> cross_entropy_fwd_bwd(Tensor logits, Tensor target) {
+  check_f32(logits, "logits");
+  TORCH_CHECK(target.scalar_type() == torch::kLong && target.is_contiguous());
+  int B = (int)logits.size(0), NC = (int)logits.size(1);
+  Tensor loss = at::zeros({}, logits.options());
+  Tensor dlogits = at::empty({B, NC}, logits.options());
+  launch_ce_fwd_bwd(logits.data_ptr<float>(), target.data_ptr<long>(),
+                    loss.data_ptr<float>(), dlogits.data_ptr<float>(), B, NC,
+                    cur_stream());
+  return {loss, dlogits};
+}
+
+// ------------------------------------------------------------------ gemm ---
+Tensor gemm_bf16(Tensor a, Tensor b) {
+  // C[M,N] = A[M,K] · B[N,K]^T
+  TORCH_CHECK(a.is_cuda() && a.is_contiguous() &&
+              a.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(b.is_contiguous() && b.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(a.size(1) == b.size(1), "K mismatch");
+  int M = (int)a.size(0), K = (int)a.size(1), N = (int)b.size(0);
+  Tensor c = at::empty({M, N}, a.options());
+  launch_gemm_bf16(a.data_ptr(), b.data_ptr(), c.data_ptr(), M, N, K,
+                   cur_stream());
+  return c;
+}
+
+// ------------------------------------------------------------- optimizers --
+void adam_step(Tensor master, Tensor grad, Tensor m, Tensor v,
+               c10::optional<Tensor> shadow, Tensor step_t, double lr,
+               double b1, double b2, double eps, double wd, bool zero_grad) {
+  check_f32(master, "master");
+  launch_adam_step(master.data_ptr<float>(), grad.data_ptr<float>(),
+                   m.data_ptr<float>(), v.data_ptr<float>(),
+                   shadow.has_value() ? shadow->data_ptr() : nullptr,
+                   step_t.data_ptr<float>(), master.numel(), (float)lr,
+                   (float)b1, (float)b2, (float)eps, (float)wd,
+                   zero_grad ? 1 : 0, cur_stream());
+}
+
+void sgd_step(Tensor master, Tensor grad, c10::optional<Tensor> mom,
+              c10::optional<Tensor> shadow, double lr, double mu, double wd,
+              bool zero_grad) {
+  check_f32(master, "master");
+  launch_sgd_step(master.data_ptr<float>(), grad.data_ptr<float>(),
+                  mom.has_value() ? mom->data_ptr<float>() : nullptr,
+                  shadow.has_value() ? shadow->data_ptr() : nullptr,
+                  master.numel(), (float)lr, (float)mu, (float)wd,
+                  zero_grad ? 1 : 0, cur_stream());
+}
+
+void permute_krsc_rsck(Tensor src, Tensor dst, Tensor meta,
+                       int64_t max_elem) {
+  TORCH_CHECK(meta.scalar_type() == torch::kInt32 && meta.is_cuda());
+  launch_permute_krsc_rsck(src.data_ptr(), dst.data_ptr(),
+                           meta.data_ptr<int>(), (int)meta.size(0),
+                           (int)max_elem, cur_stream());
+}
+
+void grad_divergence(Tensor g, Tensor prev, Tensor sumsq, Tensor out,
+                     bool skip_first) {
+  launch_grad_divergence(g.data_ptr<float>(), prev.data_ptr<float>(),
+                         sumsq.data_ptr<float>(), out.data_ptr<float>(),
+                         g.numel(), skip_first ? 1 : 0, cur_stream());
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("conv_bn_act_fwd", &conv_bn_act_fwd);
+  m.def("conv_bn_act_bwd", &conv_bn_act_bwd);
+  m.def("maxpool_fwd", &maxpool_fwd);
+  m.def("maxpool_bwd", &maxpool_bwd);
+  m.def("avgpool_fwd", &avgpool_fwd);
+  m.def("avgpool_bwd", &avgpool_bwd);
+  m.def("linear_fwd", &linear_fwd);
+  m.def("linear_bwd", &linear_bwd);
+  m.def("cross_entropy_fwd_bwd", &cross_entropy_fwd_bwd);
+  m.def("gemm_bf16", &gemm_bf16);
+  m.def("adam_step", &adam_step);
+  m.def("sgd_step", &sgd_step);
+  m.def("permute_krsc_rsck", &permute_krsc_rsck);
+  m.def("grad_divergence", &grad_divergence);
+}

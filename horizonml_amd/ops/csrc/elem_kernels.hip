@@ -1,0 +1,514 @@
+// BatchNorm / pooling / classifier / loss / optimizer kernels for gfx950.
+// All activation tensors are bf16 NHWC (flattened [M][C], channels innermost,
+// coalesced along C); statistics and parameters are f32.
+#include "common.h"
+
+// ------------------------------------------------------------- BN forward --
+// y = gamma*(x-mean)*invstd + beta  [+ residual] [ReLU]
+// Training: mean/var come from the conv epilogue's (Σy, Σy²) in `stats`;
+// block 0 additionally writes save_mean/save_invstd and updates running
+// stats.  Eval: running stats are used and nothing is written back.
+__global__ __launch_bounds__(256) void k_bn_apply(
+    const bf16* __restrict__ x, const bf16* __restrict__ res,
+    bf16* __restrict__ y, const float* __restrict__ stats,
+    const float* __restrict__ gamma, const float* __restrict__ beta,
+    float* __restrict__ running_mean, float* __restrict__ running_var,
+    float* __restrict__ save_mean, float* __restrict__ save_invstd,
+    long M, int C, float momentum, float eps, int training, int act) {
+  const float invM = 1.f / (float)M;
+  if (training && blockIdx.x == 0) {
+    for (int c = threadIdx.x; c < C; c += blockDim.x) {
+      float mean = stats[c] * invM;
+      float var = fmaxf(stats[C + c] * invM - mean * mean, 0.f);
+      save_mean[c] = mean;
+      save_invstd[c] = rsqrtf(var + eps);
+      // unbiased running var like torch.nn.BatchNorm2d
+      float ub = (M > 1) ? var * (float)M / (float)(M - 1) : var;
+      running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * mean;
+      running_var[c] = (1.f - momentum) * running_var[c] + momentum * ub;
+    }
+  }
+  long total = M * C;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int c = (int)(i % C);
+    float mean, invstd;
+    if (training) {
+      mean = stats[c] * invM;
+      float var = fmaxf(stats[C + c] * invM - mean * mean, 0.f);
+      invstd = rsqrtf(var + eps);
+    } else {
+      mean = running_mean[c];
+      invstd = rsqrtf(running_var[c] + eps);
+    }
+    float v = (b2f(x[i]) - mean) * invstd * gamma[c] + beta[c];
+    if (res != nullptr) v += b2f(res[i]);
+    if (act) v = fmaxf(v, 0.f);
+    y[i] = f2b(v);
+  }
+}
+
+// ------------------------------------------------------ BN+act backward ----
+// Pass 1: per-channel Σdz and Σ(dz·xhat) where dz = dy·relu'(y).
+// grid: (cdiv(C,64), msplit); block 256 = 4 m-lanes × 64 channels.
+__global__ __launch_bounds__(256) void k_bnact_bwd_reduce(
+    const bf16* __restrict__ dy, const bf16* __restrict__ yout,
+    const bf16* __restrict__ x, const float* __restrict__ save_mean,
+    const float* __restrict__ save_invstd, float* __restrict__ sum_dz,
+    float* __restrict__ sum_dzx, long M, int C, int act, long mchunk) {
+  __shared__ float sdz[4][64];
+  __shared__ float sdzx[4][64];
+  const int c = blockIdx.x * 64 + (threadIdx.x & 63);
+  const int mlane = threadIdx.x >> 6;
+  const long mbeg = (long)blockIdx.y * mchunk;
+  const long mend = min((long)M, mbeg + mchunk);
+  float a_dz = 0.f, a_dzx = 0.f;
+  if (c < C) {
+    const float mean = save_mean[c], invstd = save_invstd[c];
+    for (long m = mbeg + mlane; m < mend; m += 4) {
+      long i = m * C + c;
+      float g = b2f(dy[i]);
+      if (act && b2f(yout[i]) <= 0.f) g = 0.f;
+      float xh = (b2f(x[i]) - mean) * invstd;
+      a_dz += g;
+      a_dzx += g * xh;
+    }
+  }
+  sdz[mlane][threadIdx.x & 63] = a_dz;
+  sdzx[mlane][threadIdx.x & 63] = a_dzx;
+  __syncthreads();
+  if (mlane == 0 && c < C) {
+    float t1 = sdz[0][threadIdx.x] + sdz[1][threadIdx.x] +
+               sdz[2][threadIdx.x] + sdz[3][threadIdx.x];
+    float t2 = sdzx[0][threadIdx.x] + sdzx[1][threadIdx.x] +
+               sdzx[2][threadIdx.x] + sdzx[3][threadIdx.x];
+    if (gridDim.y == 1) {
+      sum_dz[c] = t1;
+      sum_dzx[c] = t2;
+    } else {
+      atomicAdd(&sum_dz[c], t1);
+      atomicAdd(&sum_dzx[c], t2);
+    }
+  }
+}
+
+// Pass 2: dconv = gamma·invstd·(dz − Σdz/M − xhat·Σdzx/M); optional dres = dz.
+__global__ __launch_bounds__(256) void k_bn_bwd_apply(
+    const bf16* __restrict__ dy, const bf16* __restrict__ yout,
+    const bf16* __restrict__ x, const float* __restrict__ save_mean,
+    const float* __restrict__ save_invstd, const float* __restrict__ gamma,
+    const float* __restrict__ sum_dz, const float* __restrict__ sum_dzx,
+    bf16* __restrict__ dconv, bf16* __restrict__ dres, long M, int C,
+    int act) {
+  const float invM = 1.f / (float)M;
+  long total = M * C;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int c = (int)(i % C);
+    float g = b2f(dy[i]);
+    if (act && b2f(yout[i]) <= 0.f) g = 0.f;
+    if (dres != nullptr) dres[i] = f2b(g);
+    float mean = save_mean[c], invstd = save_invstd[c];
+    float xh = (b2f(x[i]) - mean) * invstd;
+    float v = gamma[c] * invstd *
+              (g - sum_dz[c] * invM - xh * sum_dzx[c] * invM);
+    dconv[i] = f2b(v);
+  }
+}
+
+// ----------------------------------------------------------------- pooling --
+// 3x3/2 pad1 max-pool, NHWC; argmax index (0..8) saved as u8 for backward.
+__global__ __launch_bounds__(256) void k_maxpool_fwd(
+    const bf16* __restrict__ x, bf16* __restrict__ y,
+    unsigned char* __restrict__ idx, int Nb, int H, int W, int C, int Hp,
+    int Wp) {
+  long total = (long)Nb * Hp * Wp * C;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int c = (int)(i % C);
+    long t = i / C;
+    int wo = (int)(t % Wp);
+    t /= Wp;
+    int ho = (int)(t % Hp);
+    int n = (int)(t / Hp);
+    float best = -1e30f;
+    int barg = 0;
+    for (int r = 0; r < 3; r++) {
+      int hi = ho * 2 - 1 + r;
+      if (hi < 0 || hi >= H) continue;
+      for (int s = 0; s < 3; s++) {
+        int wi = wo * 2 - 1 + s;
+        if (wi < 0 || wi >= W) continue;
+        float v = b2f(x[((long)(n * H + hi) * W + wi) * C + c]);
+        if (v > best) {
+          best = v;
+          barg = r * 3 + s;
+        }
+      }
+    }
+    y[i] = f2b(best);
+    idx[i] = (unsigned char)barg;
+  }
+}
+
+__global__ __launch_bounds__(256) void k_maxpool_bwd(
+    const bf16* __restrict__ dy, const unsigned char* __restrict__ idx,
+    bf16* __restrict__ dx, int Nb, int H, int W, int C, int Hp, int Wp) {
+  long total = (long)Nb * H * W * C;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int c = (int)(i % C);
+    long t = i / C;
+    int wi = (int)(t % W);
+    t /= W;
+    int hi = (int)(t % H);
+    int n = (int)(t / H);
+    float acc = 0.f;
+    for (int r = 0; r < 3; r++) {
+      int hs = hi + 1 - r;
+      if (hs < 0 || (hs & 1)) continue;
+      int ho = hs >> 1;
+      if (ho >= Hp) continue;
+      for (int s = 0; s < 3; s++) {
+        int ws = wi + 1 - s;
+        if (ws < 0 || (ws & 1)) continue;
+        int wo = ws >> 1;
+        if (wo >= Wp) continue;
+        long j = ((long)(n * Hp + ho) * Wp + wo) * C + c;
+        if (idx[j] == r * 3 + s) acc += b2f(dy[j]);
+      }
+    }
+    dx[i] = f2b(acc);
+  }
+}
+
+// Global average pool [N,H,W,C] -> [N,C] (f32 out for the classifier).
+__global__ __launch_bounds__(256) void k_avgpool_fwd(
+    const bf16* __restrict__ x, bf16* __restrict__ y, int Nb, int HW, int C) {
+  long total = (long)Nb * C;
+  float inv = 1.f / (float)HW;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int c = (int)(i % C), n = (int)(i / C);
+    float s = 0.f;
+    const bf16* base = x + (long)n * HW * C + c;
+    for (int t = 0; t < HW; t++) s += b2f(base[(long)t * C]);
+    y[i] = f2b(s * inv);
+  }
+}
+
+__global__ __launch_bounds__(256) void k_avgpool_bwd(
+    const bf16* __restrict__ dy, bf16* __restrict__ dx, int Nb, int HW,
+    int C) {
+  long total = (long)Nb * HW * C;
+  float inv = 1.f / (float)HW;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int c = (int)(i % C);
+    long n = i / ((long)HW * C);
+    dx[i] = f2b(b2f(dy[n * C + c]) * inv);
+  }
+}
+
+// ------------------------------------------------------------- classifier --
+// Small fc: x[B,In] bf16 · w[Out,In] bf16 + b[Out] f32 -> logits[B,Out] f32.
+__global__ __launch_bounds__(256) void k_linear_fwd(
+    const bf16* __restrict__ x, const bf16* __restrict__ w,
+    const float* __restrict__ b, float* __restrict__ y, int B, int In,
+    int Out) {
+  long total = (long)B * Out;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int j = (int)(i % Out), bi = (int)(i / Out);
+    const bf16* xr = x + (long)bi * In;
+    const bf16* wr = w + (long)j * In;
+    float s = (b != nullptr) ? b[j] : 0.f;
+    for (int t = 0; t < In; t++) s = fmaf(b2f(xr[t]), b2f(wr[t]), s);
+    y[i] = s;
+  }
+}
+
+// dx[B,In] bf16 = dy[B,Out] f32 · w[Out,In]
+__global__ __launch_bounds__(256) void k_linear_bwd_dx(
+    const float* __restrict__ dy, const bf16* __restrict__ w,
+    bf16* __restrict__ dx, int B, int In, int Out) {
+  long total = (long)B * In;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int t = (int)(i % In), bi = (int)(i / In);
+    float s = 0.f;
+    for (int j = 0; j < Out; j++)
+      s = fmaf(dy[(long)bi * Out + j], b2f(w[(long)j * In + t]), s);
+    dx[i] = f2b(s);
+  }
+}
+
+// dw[Out,In] f32 += Σ_b dy[b,j]·x[b,t];  db[Out] f32 = Σ_b dy[b,j]
+__global__ __launch_bounds__(256) void k_linear_bwd_dw(
+    const float* __restrict__ dy, const bf16* __restrict__ x,
+    float* __restrict__ dw, float* __restrict__ db, int B, int In, int Out) {
+  long total = (long)Out * In;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int t = (int)(i % In), j = (int)(i / In);
+    float s = 0.f;
+    for (int bi = 0; bi < B; bi++)
+      s = fmaf(dy[(long)bi * Out + j], b2f(x[(long)bi * In + t]), s);
+    dw[i] = s;
+    if (t == 0 && db != nullptr) {
+      float sb = 0.f;
+      for (int bi = 0; bi < B; bi++) sb += dy[(long)bi * Out + j];
+      db[j] = sb;
+    }
+  }
+}
+
+// ------------------------------------------------------- fused CE loss -----
+// logits[B,NC] f32 -> mean NLL loss (atomic into loss[0]) + dlogits f32
+// (softmax − onehot)/B.  One thread per row (NC ≤ 32).
+__global__ __launch_bounds__(256) void k_ce_fwd_bwd(
+    const float* __restrict__ logits, const long* __restrict__ target,
+    float* __restrict__ loss, float* __restrict__ dlogits, int B, int NC) {
+  for (int b = blockIdx.x * blockDim.x + threadIdx.x; b < B;
+       b += gridDim.x * blockDim.x) {
+    const float* row = logits + (long)b * NC;
+    float mx = row[0];
+    for (int j = 1; j < NC; j++) mx = fmaxf(mx, row[j]);
+    float se = 0.f;
+    for (int j = 0; j < NC; j++) se += __expf(row[j] - mx);
+    float lse = __logf(se) + mx;
+    int t = (int)target[b];
+    atomicAdd(loss, (lse - row[t]) / (float)B);
+    float invB = 1.f / (float)B;
+    for (int j = 0; j < NC; j++) {
+      float p = __expf(row[j] - lse);
+      dlogits[(long)b * NC + j] = (p - (j == t ? 1.f : 0.f)) * invB;
+    }
+  }
+}
+
+// ------------------------------------------------------- fused optimizers --
+// Flat-buffer Adam: master f32, grad f32 (zeroed after), m/v f32, bf16 shadow
+// emitted for every element.  `step_t` is a device scalar so the kernel is
+// hipGraph-replayable (bias correction computed on device).
+__global__ __launch_bounds__(256) void k_adam_step(
+    float* __restrict__ master, float* __restrict__ grad,
+    float* __restrict__ m, float* __restrict__ v, bf16* __restrict__ shadow,
+    const float* __restrict__ step_t, long n, float lr, float b1, float b2,
+    float eps, float wd, int zero_grad) {
+  float t = step_t[0];
+  float bc1 = 1.f - __powf(b1, t), bc2 = 1.f - __powf(b2, t);
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (long)gridDim.x * blockDim.x) {
+    float g = grad[i];
+    float w = master[i];
+    if (wd != 0.f) g += wd * w;
+    float mi = b1 * m[i] + (1.f - b1) * g;
+    float vi = b2 * v[i] + (1.f - b2) * g * g;
+    m[i] = mi;
+    v[i] = vi;
+    w -= lr * (mi / bc1) / (sqrtf(vi / bc2) + eps);
+    master[i] = w;
+    if (shadow != nullptr) shadow[i] = f2b(w);
+    if (zero_grad) grad[i] = 0.f;
+  }
+}
+
+__global__ __launch_bounds__(256) void k_sgd_step(
+    float* __restrict__ master, float* __restrict__ grad,
+    float* __restrict__ mom, bf16* __restrict__ shadow, long n, float lr,
+    float mu, float wd, int zero_grad) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (long)gridDim.x * blockDim.x) {
+    float g = grad[i];
+    float w = master[i];
+    if (wd != 0.f) g += wd * w;
+    float u = (mom != nullptr) ? (mu * mom[i] + g) : g;
+    if (mom != nullptr) mom[i] = u;
+    w -= lr * u;
+    master[i] = w;
+    if (shadow != nullptr) shadow[i] = f2b(w);
+    if (zero_grad) grad[i] = 0.f;
+  }
+}
+
+__global__ void k_inc_step(float* step_t) { step_t[0] += 1.f; }
+
+// ------------------------------------------------ dgrad weight transpose ---
+// Batched KRSC -> RSCK permute of every conv weight shadow (one launch per
+// step).  meta[i*4 + {0:src_off,1:dst_off,2:nelem,3:packed}] with packed =
+// (K<<16)|(C) ... K,C ≤ 65535; rs count derived from nelem.
+__global__ __launch_bounds__(256) void k_permute_krsc_rsck(
+    const bf16* __restrict__ src, bf16* __restrict__ dst,
+    const int* __restrict__ meta, int nconv) {
+  int ci = blockIdx.y;
+  if (ci >= nconv) return;
+  const int soff = meta[ci * 4 + 0], doff = meta[ci * 4 + 1];
+  const int nelem = meta[ci * 4 + 2];
+  const int K = meta[ci * 4 + 3] >> 16, C = meta[ci * 4 + 3] & 0xffff;
+  const int RS = nelem / (K * C);
+  for (int e = blockIdx.x * blockDim.x + threadIdx.x; e < nelem;
+       e += gridDim.x * blockDim.x) {
+    // e indexes dst in RSCK order: e = ((rs*C)+c)*K + ko
+    int ko = e % K;
+    int rem = e / K;
+    int c = rem % C;
+    int rs = rem / C;
+    dst[doff + e] = src[soff + ((long)ko * RS + rs) * C + c];
+  }
+}
+
+// --------------------------------------------------- gradient divergence ---
+// sumsq += Σ (g−prev)²; prev ← g   (flat f32), then finalize adds sqrt.
+__global__ __launch_bounds__(256) void k_gdiv_partial(
+    const float* __restrict__ g, float* __restrict__ prev,
+    float* __restrict__ sumsq, long n) {
+  float a = 0.f;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (long)gridDim.x * blockDim.x) {
+    float d = g[i] - prev[i];
+    prev[i] = g[i];
+    a += d * d;
+  }
+  a = wave_reduce_sum(a);
+  __shared__ float ws[4];
+  if ((threadIdx.x & 63) == 0) ws[threadIdx.x >> 6] = a;
+  __syncthreads();
+  if (threadIdx.x == 0)
+    atomicAdd(sumsq, ws[0] + ws[1] + ws[2] + ws[3]);
+}
+
+__global__ void k_gdiv_finalize(float* __restrict__ sumsq,
+                                float* __restrict__ out, int skip_first) {
+  if (!skip_first) out[0] += sqrtf(sumsq[0]);
+  sumsq[0] = 0.f;
+}
+
+// ------------------------------------------------------------- launchers --
+static inline int gsz(long total, int block = 256, int cap = 4096) {
+  long g = (total + block - 1) / block;
+  if (g > cap) g = cap;
+  if (g < 1) g = 1;
+  return (int)g;
+}
+
+extern "C" {
+
+void launch_bn_apply(const void* x, const void* res, void* y,
+                     const float* stats, const float* gamma,
+                     const float* beta, float* rmean, float* rvar,
+                     float* smean, float* sinvstd, long M, int C,
+                     float momentum, float eps, int training, int act,
+                     hipStream_t st) {
+  k_bn_apply<<<gsz(M * (long)C), 256, 0, st>>>(
+      (const bf16*)x, (const bf16*)res, (bf16*)y, stats, gamma, beta, rmean,
+      rvar, smean, sinvstd, M, C, momentum, eps, training, act);
+}
+
+void launch_bnact_bwd_reduce(const void* dy, const void* yout, const void* x,
+                             const float* smean, const float* sinvstd,
+                             float* sum_dz, float* sum_dzx, long M, int C,
+                             int act, hipStream_t st) {
+  int cblocks = (C + 63) / 64;
+  int msplit = (int)min((long)64, max((long)1, (long)(256 / cblocks)));
+  long mchunk = (M + msplit - 1) / msplit;
+  msplit = (int)((M + mchunk - 1) / mchunk);
+  if (msplit > 1) {
+    hipMemsetAsync(sum_dz, 0, C * sizeof(float), st);
+    hipMemsetAsync(sum_dzx, 0, C * sizeof(float), st);
+  }
+  dim3 grid(cblocks, msplit);
+  k_bnact_bwd_reduce<<<grid, 256, 0, st>>>(
+      (const bf16*)dy, (const bf16*)yout, (const bf16*)x, smean, sinvstd,
+      sum_dz, sum_dzx, M, C, act, mchunk);
+}
+
+void launch_bn_bwd_apply(const void* dy, const void* yout, const void* x,
+                         const float* smean, const float* sinvstd,
+                         const float* gamma, const float* sum_dz,
+                         const float* sum_dzx, void* dconv, void* dres,
+                         long M, int C, int act, hipStream_t st) {
+  k_bn_bwd_apply<<<gsz(M * (long)C), 256, 0, st>>>(
+      (const bf16*)dy, (const bf16*)yout, (const bf16*)x, smean, sinvstd,
+      gamma, sum_dz, sum_dzx, (bf16*)dconv, (bf16*)dres, M, C, act);
+}
+
+void launch_maxpool_fwd(const void* x, void* y, unsigned char* idx, int Nb,
+                        int H, int W, int C, int Hp, int Wp, hipStream_t st) {
+  k_maxpool_fwd<<<gsz((long)Nb * Hp * Wp * C), 256, 0, st>>>(
+      (const bf16*)x, (bf16*)y, idx, Nb, H, W, C, Hp, Wp);
+}
+
+void launch_maxpool_bwd(const void* dy, const unsigned char* idx, void* dx,
+                        int Nb, int H, int W, int C, int Hp, int Wp,
+                        hipStream_t st) {
+  k_maxpool_bwd<<<gsz((long)Nb * H * W * C), 256, 0, st>>>(
+      (const bf16*)dy, idx, (bf16*)dx, Nb, H, W, C, Hp, Wp);
+}
+
+void launch_avgpool_fwd(const void* x, void* y, int Nb, int HW, int C,
+                        hipStream_t st) {
+  k_avgpool_fwd<<<gsz((long)Nb * C), 256, 0, st>>>((const bf16*)x, (bf16*)y,
+                                                   Nb, HW, C);
+}
+
+void launch_avgpool_bwd(const void* dy, void* dx, int Nb, int HW, int C,
+                        hipStream_t st) {
+  k_avgpool_bwd<<<gsz((long)Nb * HW * C), 256, 0, st>>>((const bf16*)dy,
+                                                        (bf16*)dx, Nb, HW, C);
+}
+
+void launch_linear_fwd(const void* x, const void* w, const float* b, float* y,
+                       int B, int In, int Out, hipStream_t st) {
+  k_linear_fwd<<<gsz((long)B * Out), 256, 0, st>>>((const bf16*)x,
+                                                   (const bf16*)w, b, y, B,
+                                                   In, Out);
+}
+
+void launch_linear_bwd(const float* dy, const void* x, const void* w,
+                       void* dx, float* dw, float* db, int B, int In, int Out,
+                       hipStream_t st) {
+  if (dx)
+    k_linear_bwd_dx<<<gsz((long)B * In), 256, 0, st>>>(dy, (const bf16*)w,
+                                                       (bf16*)dx, B, In, Out);
+  k_linear_bwd_dw<<<gsz((long)Out * In), 256, 0, st>>>(dy, (const bf16*)x, dw,
+                                                       db, B, In, Out);
+}
+
+void launch_ce_fwd_bwd(const float* logits, const long* target, float* loss,
+                       float* dlogits, int B, int NC, hipStream_t st) {
+  k_ce_fwd_bwd<<<gsz(B), 256, 0, st>>>(logits, target, loss, dlogits, B, NC);
+}
+
+void launch_adam_step(float* master, float* grad, float* m, float* v,
+                      void* shadow, const float* step_t, long n, float lr,
+                      float b1, float b2, float eps, float wd, int zero_grad,
+                      hipStream_t st) {
+  k_inc_step<<<1, 1, 0, st>>>((float*)step_t);
+  k_adam_step<<<gsz(n), 256, 0, st>>>(master, grad, m, v, (bf16*)shadow,
+                                      step_t, n, lr, b1, b2, eps, wd,
+                                      zero_grad);
+}
+
+void launch_sgd_step(float* master, float* grad, float* mom, void* shadow,
+                     long n, float lr, float mu, float wd, int zero_grad,
+                     hipStream_t st) {
+  k_sgd_step<<<gsz(n), 256, 0, st>>>(master, grad, mom, (bf16*)shadow, n, lr,
+                                     mu, wd, zero_grad);
+}
+
+void launch_permute_krsc_rsck(const void* src, void* dst, const int* meta,
+                              int nconv, int max_elem, hipStream_t st) {
+  dim3 grid(gsz(max_elem), nconv);
+  k_permute_krsc_rsck<<<grid, 256, 0, st>>>((const bf16*)src, (bf16*)dst,
+                                            meta, nconv);
+}
+
+void launch_grad_divergence(const float* g, float* prev, float* sumsq,
+                            float* out, long n, int skip_first,
+                            hipStream_t st) {
+  k_gdiv_partial<<<gsz(n), 256, 0, st>>>(g, prev, sumsq, n);
+  k_gdiv_finalize<<<1, 1, 0, st>>>(sumsq, out, skip_first);
+}
+
+}  // extern "C"

@@ -1,0 +1,86 @@
+"""Multi-process CPU/gloo plumbing tests (BASELINE.json config #1)."""
+import os
+
+import pandas as pd
+import pytest
+import torch
+import torch.multiprocessing as mp
+import torch.nn as nn
+
+from horizonml_amd.parallel import BucketedDataParallel
+from horizonml_amd.runtime.distributed import (setup_distributed,
+                                               teardown_distributed)
+from horizonml_amd.utils.ports import find_free_port
+
+
+def _ddp_parity_worker(rank, world, port, result_q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    ctx = setup_distributed(rank, world, port, backend="gloo")
+    torch.manual_seed(0)
+    model = nn.Sequential(nn.Linear(8, 16), nn.ReLU(), nn.Linear(16, 4))
+    ddp = BucketedDataParallel(model, bucket_cap_mb=0.0001)  # many buckets
+    opt = torch.optim.SGD(model.parameters(), lr=0.1)
+    g = torch.Generator().manual_seed(42)
+    X = torch.randn(8, 8, generator=g)
+    Y = torch.randn(8, 4, generator=g)
+    for _ in range(3):
+        opt.zero_grad()
+        shard_x = X[rank * 4:(rank + 1) * 4]
+        shard_y = Y[rank * 4:(rank + 1) * 4]
+        loss = ((ddp(shard_x) - shard_y) ** 2).mean()
+        loss.backward()
+        ddp.finalize_backward()
+        opt.step()
+    flat = torch.cat([p.detach().flatten() for p in model.parameters()])
+    result_q.put((rank, flat.tolist()))  # by-value: survives process exit
+    teardown_distributed(ctx)
+
+
+def test_ddp_matches_single_process():
+    """2-rank DP on half batches must equal single-process full-batch SGD."""
+    mp_ctx = mp.get_context("spawn")
+    q = mp_ctx.Queue()
+    port = find_free_port()
+    procs = [mp_ctx.Process(target=_ddp_parity_worker, args=(r, 2, port, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    results = {r: torch.tensor(v) for r, v in
+               (q.get(timeout=120) for _ in range(2))}
+    for p in procs:
+        p.join(timeout=60)
+
+    # single-process reference: mean loss over the full batch
+    torch.manual_seed(0)
+    model = nn.Sequential(nn.Linear(8, 16), nn.ReLU(), nn.Linear(16, 4))
+    opt = torch.optim.SGD(model.parameters(), lr=0.1)
+    g = torch.Generator().manual_seed(42)
+    X = torch.randn(8, 8, generator=g)
+    Y = torch.randn(8, 4, generator=g)
+    for _ in range(3):
+        opt.zero_grad()
+        ((model(X) - Y) ** 2).mean().backward()
+        opt.step()
+    ref = torch.cat([p.detach().flatten() for p in model.parameters()])
+
+    assert torch.allclose(results[0], results[1], atol=1e-6), \
+        "ranks diverged"
+    assert torch.allclose(results[0], ref, atol=1e-4), \
+        f"DP != single-process (max diff {(results[0] - ref).abs().max()})"
+
+
+@pytest.mark.timeout(300)
+def test_dp_entrypoint_end_to_end(tmp_path):
+    from data_parallel_train import run_data_parallel
+    df = run_data_parallel(world_size=2, epochs=1, sample_size=64,
+                           logs_dir=str(tmp_path), batch_size=32,
+                           backend="gloo", synthetic=True)
+    assert df is not None
+    assert set(["epoch", "loss", "accuracy", "epoch_time", "avg_step_time",
+                "compute_time", "comm_time", "idle_time", "avg_cpu",
+                "avg_memory", "grad_divergence", "worker",
+                "total_training_time"]).issubset(df.columns)
+    assert len(df) == 2  # one epoch row per worker
+    combined = os.path.join(str(tmp_path), "combined_results_64.csv")
+    assert os.path.isfile(combined)
+    assert pd.read_csv(combined).shape[0] == 2
