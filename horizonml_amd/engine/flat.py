@@ -1,0 +1,117 @@
+"""Flat parameter/gradient management for the MI355X fast path.
+
+MI355X-first memory design (288 GB HBM3E: keep everything resident, few big
+buffers, single-kernel updates):
+
+* ONE flat f32 master buffer holds every trainable parameter (modules' params
+  are re-pointed to views), ONE flat f32 gradient buffer (``.grad`` views are
+  pre-assigned so autograd accumulates in place — stable addresses under
+  hipGraph capture), ONE flat bf16 shadow the conv/linear kernels consume.
+* The fused Adam/SGD kernel updates master + emits the bf16
+  shadow + zeroes the gradient buffer in a single pass over the flat range.
+* A second flat bf16 buffer holds the RSCK (dgrad) weight images, refreshed
+  by one batched permute kernel per step.
+* DP gradient sync becomes ONE bf16 all-reduce of the flat buffer.
+"""
+from __future__ import annotations
+
+from typing import List, Optional
+
+import torch
+import torch.nn as nn
+
+from .. import ops as _ops
+from ..models.layers import ConvBNAct, Linear
+
+
+class FlatParamManager:
+    def __init__(self, model: nn.Module, device: torch.device):
+        params = [p for p in model.parameters() if p.requires_grad]
+        total = sum(p.numel() for p in params)
+        self.params = params
+        self.numel = total
+        self.master = torch.zeros(total, device=device)
+        self.grad = torch.zeros(total, device=device)
+        self.shadow = torch.zeros(total, device=device, dtype=torch.bfloat16)
+        self.slices = {}
+        off = 0
+        for p in params:
+            n = p.numel()
+            with torch.no_grad():
+                self.master[off:off + n].copy_(p.data.flatten())
+            p.data = self.master[off:off + n].view(p.shape)
+            p.grad = self.grad[off:off + n].view(p.shape)
+            self.slices[id(p)] = (off, n)
+            off += n
+        with torch.no_grad():
+            self.shadow.copy_(self.master)
+
+        # wire bf16 shadow views into the modules; collect conv RSCK metadata
+        convs = []
+        for m in model.modules():
+            if isinstance(m, (ConvBNAct, Linear)):
+                woff, wn = self.slices[id(m.weight)]
+                m.weight_bf16 = self.shadow[woff:woff + wn].view(
+                    m.weight.shape)
+                m._managed = True
+                if isinstance(m, ConvBNAct):
+                    convs.append((m, woff, wn))
+        rsck_total = sum(wn for _, _, wn in convs)
+        self.rsck = torch.zeros(rsck_total, device=device,
+                                dtype=torch.bfloat16)
+        meta: List[List[int]] = []
+        doff = 0
+        self.max_elem = 1
+        for m, woff, wn in convs:
+            K, R, S, C = m.weight.shape
+            meta.append([woff, doff, wn, (K << 16) | C])
+            m._w_rsck = self.rsck[doff:doff + wn].view(R, S, C, K)
+            doff += wn
+            self.max_elem = max(self.max_elem, wn)
+        self.meta = (torch.tensor(meta, dtype=torch.int32, device=device)
+                     if meta else None)
+        self.refresh_rsck()
+
+    def refresh_rsck(self):
+        if self.meta is not None:
+            _ops.extension().permute_krsc_rsck(self.shadow, self.rsck,
+                                               self.meta, self.max_elem)
+
+    def zero_grad(self):
+        self.grad.zero_()
+
+
+class HorizonAdam:
+    """Fused flat-buffer Adam (K9): 1 elementwise kernel + step increment +
+    batched RSCK permute per step; zero_grad folded in."""
+
+    def __init__(self, mgr: FlatParamManager, lr: float = 1e-3,
+                 betas=(0.9, 0.999), eps: float = 1e-8,
+                 weight_decay: float = 0.0):
+        self.mgr = mgr
+        self.lr, self.betas, self.eps, self.wd = lr, betas, eps, weight_decay
+        dev = mgr.master.device
+        self.m = torch.zeros_like(mgr.master)
+        self.v = torch.zeros_like(mgr.master)
+        self.step_t = torch.zeros(1, device=dev)
+
+    def step(self, zero_grad: bool = True):
+        _ops.extension().adam_step(self.mgr.master, self.mgr.grad, self.m,
+                                   self.v, self.mgr.shadow, self.step_t,
+                                   self.lr, self.betas[0], self.betas[1],
+                                   self.eps, self.wd, zero_grad)
+        self.mgr.refresh_rsck()
+
+
+class HorizonSGD:
+    def __init__(self, mgr: FlatParamManager, lr: float = 0.1,
+                 momentum: float = 0.9, weight_decay: float = 0.0):
+        self.mgr = mgr
+        self.lr, self.mu, self.wd = lr, momentum, weight_decay
+        self.mom = (torch.zeros_like(mgr.master) if momentum > 0 else None)
+
+    def step(self, zero_grad: bool = True):
+        _ops.extension().sgd_step(self.mgr.master, self.mgr.grad, self.mom,
+                                  self.mgr.shadow, self.lr, self.mu, self.wd,
+                                  zero_grad)
+        self.mgr.refresh_rsck()

@@ -1,0 +1,131 @@
+"""Tensor-parallel training engine (strategy 3).
+
+Reference loop parity (``tensor_parallel_train.py:155-296``): identical data
+on every rank, per-epoch CSV on all ranks with ``avg_bandwidth``.  The
+semantics are corrected per SURVEY.md Q3: shard outputs are all-gathered
+through autograd (backward slices/reduces), only *replicated* parameters get
+gradient averaging, and shard parameters keep shard-local optimizer state.
+"""
+from __future__ import annotations
+
+import time
+from typing import Optional
+
+import torch
+import torch.nn.functional as F
+
+from ..data import get_dataloader
+from ..engine.common import GradDivergenceProbe, Meters, build_optimizer
+from ..parallel import BucketedDataParallel
+from ..parallel.tensor_parallel import replicated_parameters
+from ..parallel.tp_models import build_tp_resnet18
+from ..profiling.metrics import (EpochMetrics, MetricsWriter,
+                                 sample_gpu_resources, sample_host_resources)
+from ..profiling.timers import StepProfiler
+from ..runtime.distributed import (DistContext, barrier, setup_distributed,
+                                   teardown_distributed)
+from ..utils.seed import seed_everything
+
+
+def train_tp(ctx: DistContext, epochs: int, sample_size: int, logs_dir: str,
+             batch_size: int = 64, lr: float = 1e-3,
+             optimizer_name: str = "adam", synthetic: Optional[bool] = None,
+             data_dir: str = "./data", tp_mode: str = "fc",
+             probe_divergence: bool = True, log_progress: bool = True):
+    rank, world = ctx.rank, ctx.world_size
+    seed_everything(rank=0)   # replicated params identical across ranks
+    torch.manual_seed(1234 + rank)  # shard params differ per rank by design
+    loader, _ = get_dataloader(rank, world, batch_size, sample_size,
+                               strategy="tp", data_dir=data_dir,
+                               synthetic=synthetic)
+
+    model = build_tp_resnet18(world, rank, num_classes=10, mode=tp_mode)
+    if ctx.is_gpu:
+        model = model.to(ctx.device)
+    prof = StepProfiler(ctx.device if ctx.is_gpu else None)
+    rep_params = replicated_parameters(model)
+    ddp = BucketedDataParallel(model, profiler=prof, parameters=rep_params) \
+        if rep_params else None
+    optimizer = build_optimizer(model.parameters(), optimizer_name, lr=lr)
+    probe = (GradDivergenceProbe(model.parameters())
+             if probe_divergence else None)
+
+    writer = MetricsWriter(logs_dir, rank, sample_size, with_bandwidth=True,
+                           with_gpu=ctx.is_gpu)
+    meters = Meters(ctx.device if ctx.is_gpu else None)
+    import psutil
+    proc = psutil.Process()
+    proc.cpu_percent(interval=None)
+
+    for epoch in range(epochs):
+        with prof.idle():
+            barrier(ctx)
+        epoch_start = time.time()
+        cpu_samples, mem_samples = [], []
+        for x, y in loader:
+            prof.step_begin()
+            cpu, mem = sample_host_resources(proc)
+            cpu_samples.append(cpu)
+            mem_samples.append(mem)
+            if ctx.is_gpu:
+                x = x.to(ctx.device, non_blocking=True).to(
+                    memory_format=torch.channels_last).to(torch.bfloat16)
+                y = y.to(ctx.device, non_blocking=True)
+            with prof.compute():
+                optimizer.zero_grad(set_to_none=False)
+                logits = model(x)
+                if logits.is_cuda:
+                    from ..models._functional_gpu import cross_entropy
+                    loss = cross_entropy(logits, y)
+                else:
+                    loss = F.cross_entropy(logits.float(), y)
+                loss.backward()
+            if ddp is not None:
+                with prof.comm():
+                    ddp.finalize_backward()
+            with prof.compute():
+                optimizer.step()
+            meters.update(loss, logits, y)
+            if probe is not None:
+                probe.step()
+            prof.step_end()
+        if ctx.is_gpu:
+            from ..models import refresh_all_shadows
+            refresh_all_shadows(model)
+        epoch_time = time.time() - epoch_start
+        loss_v, acc_v = meters.epoch_values()
+        t = prof.epoch_end()
+        gmem, gutil = sample_gpu_resources(ctx.device if ctx.is_gpu else None)
+        m = EpochMetrics(
+            epoch=epoch + 1, loss=loss_v, accuracy=acc_v,
+            epoch_time=epoch_time, avg_step_time=t["avg_step_time"],
+            compute_time=t["compute_time"], comm_time=t["comm_time"],
+            idle_time=t["idle_time"],
+            avg_cpu=sum(cpu_samples) / max(1, len(cpu_samples)),
+            avg_memory=sum(mem_samples) / max(1, len(mem_samples)),
+            grad_divergence=probe.epoch_value() if probe is not None else 0.0,
+            avg_bandwidth=t["avg_bandwidth"], gpu_memory_mb=gmem,
+            gpu_util=gutil)
+        writer.append(m)
+        if log_progress and rank == 0:
+            print(f"[tp rank0] epoch {epoch + 1}/{epochs} loss={loss_v:.4f} "
+                  f"acc={acc_v:.2f}% time={epoch_time:.2f}s", flush=True)
+        barrier(ctx)
+    return writer.path
+
+
+def tp_worker(rank: int, world_size: int, epochs: int, sample_size: int,
+              port: int, logs_dir: str, batch_size: int = 64,
+              backend: Optional[str] = None,
+              synthetic: Optional[bool] = None, lr: float = 1e-3,
+              optimizer_name: str = "adam", tp_mode: str = "fc"):
+    ctx = setup_distributed(rank, world_size, port, backend=backend)
+    try:
+        if ctx.is_gpu:
+            from .. import ops as _ops
+            _ops.extension()
+        train_tp(ctx, epochs, sample_size, logs_dir, batch_size=batch_size,
+                 synthetic=synthetic, lr=lr, optimizer_name=optimizer_name,
+                 tp_mode=tp_mode)
+    finally:
+        teardown_distributed(ctx)

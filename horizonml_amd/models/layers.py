@@ -78,11 +78,17 @@ class ConvBNAct(nn.Module):
 
     # -- GPU shadow management -------------------------------------------
     def refresh_shadow(self):
-        """(Re)materialize the bf16 KRSC weight shadow on the weight's device."""
+        """(Re)materialize the bf16 KRSC weight shadow on the weight's device.
+        When a FlatParamManager owns the shadow (``_managed``) the fused
+        optimizer kernel keeps it current — nothing to do here."""
+        if getattr(self, "_managed", False):
+            return
         with torch.no_grad():
             self.weight_bf16 = self.weight.detach().to(torch.bfloat16).contiguous()
 
     def _shadow(self) -> torch.Tensor:
+        if getattr(self, "_managed", False):
+            return self.weight_bf16
         if (self.weight_bf16.numel() != self.weight.numel()
                 or self.weight_bf16.device != self.weight.device):
             self.refresh_shadow()
@@ -148,10 +154,14 @@ class Linear(nn.Module):
                 self.bias.uniform_(-bound, bound)
 
     def refresh_shadow(self):
+        if getattr(self, "_managed", False):
+            return
         with torch.no_grad():
             self.weight_bf16 = self.weight.detach().to(torch.bfloat16).contiguous()
 
     def _shadow(self) -> torch.Tensor:
+        if getattr(self, "_managed", False):
+            return self.weight_bf16
         if (self.weight_bf16.numel() != self.weight.numel()
                 or self.weight_bf16.device != self.weight.device):
             self.refresh_shadow()

@@ -53,14 +53,15 @@ class BucketedDataParallel(nn.Module):
     def __init__(self, module: nn.Module, bucket_cap_mb: float = 25.0,
                  comm_dtype: Optional[torch.dtype] = None,
                  process_group=None, profiler=None,
-                 broadcast_params: bool = True):
+                 broadcast_params: bool = True, parameters=None):
         super().__init__()
         self.module = module
         self.group = process_group
         self.profiler = profiler
         self.world_size = (dist.get_world_size(process_group)
                            if dist.is_initialized() else 1)
-        params = [p for p in module.parameters() if p.requires_grad]
+        params = (list(parameters) if parameters is not None
+                  else [p for p in module.parameters() if p.requires_grad])
         if not params:
             raise ValueError("module has no trainable parameters")
         self.device = params[0].device

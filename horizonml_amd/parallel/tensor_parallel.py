@@ -1,0 +1,203 @@
+"""Tensor parallelism: autograd-correct sharded layers over RCCL.
+
+The reference's ``TensorParallelLinear`` assembles shard outputs with a
+non-autograd broadcast loop and then all-reduce-averages *every* parameter
+including the per-rank-different shards (``tensor_parallel_train.py:27-64,
+215-218`` — SURVEY.md Q3: it degenerates to replicated DP and is flaky).
+This module implements it for real (north star: "all-gather + reduce-scatter
+for tensor-parallel conv/linear shards"):
+
+* ``ColumnParallelLinear`` — weight split along out-features; forward
+  all-gathers the shard outputs; backward slices the upstream grad and
+  all-reduces the input grad.
+* ``RowParallelLinear`` — weight split along in-features; forward
+  all-reduces partial outputs; backward is communication-free for dx
+  (the "reduce-scatter" pairing when composed column→row).
+* ``ShardedConvBNAct`` — out-channel-sharded fused conv+BN block (each rank
+  owns its channels' filters *and* BN params); forward all-gathers along C.
+* shard-local parameters carry ``.tensor_parallel = True`` so the trainer
+  excludes them from replica gradient averaging (shard-local Adam state).
+"""
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+import torch.distributed as dist
+import torch.nn as nn
+
+from ..models.layers import ConvBNAct, Linear
+
+
+def _world(group):
+    return dist.get_world_size(group) if dist.is_initialized() else 1
+
+
+class _CopyToParallel(torch.autograd.Function):
+    """Identity forward; all-reduce (sum) backward — the entry point of a
+    column-parallel region whose input is replicated."""
+
+    @staticmethod
+    def forward(ctx, x, group):
+        ctx.group = group
+        return x
+
+    @staticmethod
+    def backward(ctx, gx):
+        if _world(ctx.group) > 1:
+            gx = gx.contiguous()
+            dist.all_reduce(gx, group=ctx.group)
+        return gx, None
+
+
+class _GatherFromParallel(torch.autograd.Function):
+    """All-gather shard outputs along `dim`; backward takes the local slice."""
+
+    @staticmethod
+    def forward(ctx, x, dim, group):
+        ctx.dim = dim
+        ctx.group = group
+        ws = _world(group)
+        if ws == 1:
+            return x
+        x = x.contiguous()
+        parts = [torch.empty_like(x) for _ in range(ws)]
+        dist.all_gather(parts, x, group=group)
+        ctx.rank = dist.get_rank(group)
+        ctx.shard = x.shape[dim]
+        return torch.cat(parts, dim=dim)
+
+    @staticmethod
+    def backward(ctx, gy):
+        if _world(ctx.group) == 1:
+            return gy, None, None
+        start = ctx.rank * ctx.shard
+        gx = gy.narrow(ctx.dim, start, ctx.shard).contiguous()
+        return gx, None, None
+
+
+class _ReduceFromParallel(torch.autograd.Function):
+    """All-reduce (sum) forward; identity backward — row-parallel output."""
+
+    @staticmethod
+    def forward(ctx, x, group):
+        if _world(group) > 1:
+            x = x.contiguous()
+            dist.all_reduce(x, group=group)
+        return x
+
+    @staticmethod
+    def backward(ctx, gy):
+        return gy, None
+
+
+def copy_to_parallel(x, group=None):
+    return _CopyToParallel.apply(x, group)
+
+
+def gather_from_parallel(x, dim=-1, group=None):
+    return _GatherFromParallel.apply(x, dim, group)
+
+
+def reduce_from_parallel(x, group=None):
+    return _ReduceFromParallel.apply(x, group)
+
+
+def _mark_tp(module: nn.Module):
+    for p in module.parameters():
+        p.tensor_parallel = True
+    return module
+
+
+class ColumnParallelLinear(nn.Module):
+    def __init__(self, in_features: int, out_features: int, world_size: int,
+                 rank: int, bias: bool = True, group=None,
+                 gather_output: bool = True):
+        super().__init__()
+        if out_features % world_size != 0:
+            raise ValueError(f"out_features={out_features} not divisible by "
+                             f"world_size={world_size}")
+        self.group = group
+        self.world_size, self.rank = world_size, rank
+        self.gather_output = gather_output
+        self.shard = out_features // world_size
+        self.local = _mark_tp(Linear(in_features, self.shard, bias=bias))
+
+    def forward(self, x):
+        x = copy_to_parallel(x, self.group)
+        y = self.local(x)
+        if self.gather_output:
+            y = gather_from_parallel(y, dim=-1, group=self.group)
+        return y
+
+
+class RowParallelLinear(nn.Module):
+    def __init__(self, in_features: int, out_features: int, world_size: int,
+                 rank: int, bias: bool = True, group=None,
+                 input_is_parallel: bool = True):
+        super().__init__()
+        if in_features % world_size != 0:
+            raise ValueError("in_features not divisible by world_size")
+        self.group = group
+        self.world_size, self.rank = world_size, rank
+        self.input_is_parallel = input_is_parallel
+        self.shard = in_features // world_size
+        # bias added once (after the reduce), kept on the local module
+        self.local = _mark_tp(Linear(self.shard, out_features, bias=False))
+        self.bias = nn.Parameter(torch.zeros(out_features)) if bias else None
+
+    def forward(self, x):
+        if not self.input_is_parallel:
+            start = self.rank * self.shard
+            x = x.narrow(-1, start, self.shard)
+        y = self.local(x)
+        y = reduce_from_parallel(y, self.group)
+        if self.bias is not None:
+            y = y + self.bias
+        return y
+
+
+class ShardedConvBNAct(nn.Module):
+    """Out-channel-sharded fused conv+BN(+ReLU): each rank computes its
+    K/world_size filters on the full input, then the channel dim is
+    all-gathered (NCHW dim 1).  BN batch statistics are per-channel, so the
+    shard owns them exactly (no cross-rank BN sync needed)."""
+
+    def __init__(self, in_ch: int, out_ch: int, kernel_size: int = 3,
+                 stride: int = 1, padding: Optional[int] = None,
+                 act: bool = True, world_size: int = 1, rank: int = 0,
+                 group=None):
+        super().__init__()
+        if out_ch % world_size != 0:
+            raise ValueError("out_ch not divisible by world_size")
+        self.group = group
+        self.world_size = world_size
+        self.local = _mark_tp(ConvBNAct(in_ch, out_ch // world_size,
+                                        kernel_size, stride, padding, act))
+
+    def forward(self, x, residual=None):
+        x = copy_to_parallel(x, self.group)
+        if residual is not None:
+            # residual is full-width; take this shard's channel slice
+            shard = self.local.out_ch
+            rank = dist.get_rank(self.group) if dist.is_initialized() else 0
+            residual = residual.narrow(1, rank * shard, shard)
+            if residual.dim() == 4:
+                residual = residual.contiguous(
+                    memory_format=torch.channels_last)
+        y = self.local(x, residual=residual)
+        y = gather_from_parallel(y, dim=1, group=self.group)
+        if y.dim() == 4 and y.is_cuda:
+            y = y.contiguous(memory_format=torch.channels_last)
+        return y
+
+
+def replicated_parameters(model: nn.Module):
+    """Parameters NOT marked tensor-parallel (need replica grad averaging)."""
+    return [p for p in model.parameters()
+            if not getattr(p, "tensor_parallel", False)]
+
+
+def shard_parameters(model: nn.Module):
+    return [p for p in model.parameters()
+            if getattr(p, "tensor_parallel", False)]

@@ -31,6 +31,9 @@ class StepProfiler:
     def __init__(self, device: Optional[torch.device] = None):
         self.use_cuda = device is not None and device.type == "cuda"
         self.device = device
+        # set True when comm() sections are nested inside compute() (the
+        # pipeline relay): epoch_end then reports compute − comm.
+        self.subtract_comm_from_compute = False
         self.reset_epoch()
         self.bytes_sent = 0.0
         self.step_times: List[float] = []
@@ -90,6 +93,8 @@ class StepProfiler:
             torch.cuda.synchronize(self.device)
             for cat, s, e in self._event_pairs:
                 totals[cat] += s.elapsed_time(e) / 1000.0  # ms -> s
+        if self.subtract_comm_from_compute:
+            totals["compute"] = max(0.0, totals["compute"] - totals["comm"])
         n_steps = max(1, len(self.step_times))
         out = {
             "compute_time": totals["compute"],

@@ -1,0 +1,60 @@
+#!/usr/bin/env python3
+"""Tensor-parallel training entrypoint (strategy 3).
+
+CLI/CSV parity with the reference's ``tensor_parallel_train.py`` (flags
+``--world_size --epochs --sample_size``; per-worker CSVs with
+``avg_bandwidth``), corrected semantics: autograd-aware
+all-gather of shard outputs, replicated-only gradient averaging, shard-local
+optimizer state.  ``--tp_mode full`` shards every conv's output channels
+(BASELINE.json config #4); the default ``fc`` mode shards the classifier
+like the reference.
+"""
+from __future__ import annotations
+
+import argparse
+
+from horizonml_amd.engine.tp import tp_worker
+from horizonml_amd.runtime.launcher import run_workers
+
+
+def run_tensor_parallel(world_size: int, epochs: int, sample_size: int,
+                        logs_dir: str = "tensor_parallel_logs",
+                        batch_size: int = 64, backend=None, synthetic=None,
+                        lr: float = 1e-3, optimizer_name: str = "adam",
+                        tp_mode: str = "fc"):
+    """Launcher parity with reference ``run_tensor_parallel``
+    (``tensor_parallel_train.py:327-385``; TP keeps the 400 s timeout base)."""
+    return run_workers(tp_worker, world_size, epochs, sample_size, logs_dir,
+                       timeout_base=400,
+                       extra_args=(batch_size, backend, synthetic, lr,
+                                   optimizer_name, tp_mode))
+
+
+def main():
+    ap = argparse.ArgumentParser(description="Tensor-parallel training")
+    ap.add_argument("--world_size", type=int, default=5)
+    ap.add_argument("--epochs", type=int, default=5)
+    ap.add_argument("--sample_size", type=int, default=1000)
+    ap.add_argument("--logs_dir", type=str, default="tensor_parallel_logs")
+    ap.add_argument("--batch_size", type=int, default=64)
+    ap.add_argument("--backend", type=str, default=None, nargs="?",
+                    choices=[None, "nccl", "gloo"])
+    ap.add_argument("--synthetic", action="store_true", default=None)
+    ap.add_argument("--lr", type=float, default=1e-3)
+    ap.add_argument("--optimizer", type=str, default="adam",
+                    choices=["adam", "sgd"])
+    ap.add_argument("--tp_mode", type=str, default="fc",
+                    choices=["fc", "full"],
+                    help="fc: reference-parity classifier shard; "
+                         "full: sharded conv2d + all-gather")
+    args = ap.parse_args()
+    df = run_tensor_parallel(args.world_size, args.epochs, args.sample_size,
+                             args.logs_dir, args.batch_size, args.backend,
+                             args.synthetic, args.lr, args.optimizer,
+                             args.tp_mode)
+    if df is not None:
+        print(df.tail(args.world_size).to_string(index=False))
+
+
+if __name__ == "__main__":
+    main()
