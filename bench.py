@@ -109,13 +109,23 @@ def main():
         opt.step()                            # fused adam + zero_grad + rsck
         return loss
 
-    # ---- warmup (eager) + graph capture ---------------------------------
+    # ---- warmup + graph capture -----------------------------------------
+    # hipGraph capture recipe (torch "whole-network capture"): warm up on a
+    # SIDE stream so AccumulateGrad nodes are not bound to the default
+    # stream, drop every reference to the warmup autograd graph, then
+    # capture one full step.
     mode = "eager" if args.no_graph else "graph"
-    for i in range(3):
-        x_static.copy_(pool_x[i % pool_n])
-        y_static.copy_(pool_y[i % pool_n])
-        loss = train_step()
+    side = torch.cuda.Stream()
+    side.wait_stream(torch.cuda.current_stream())
+    with torch.cuda.stream(side):
+        for i in range(3):
+            x_static.copy_(pool_x[i % pool_n])
+            y_static.copy_(pool_y[i % pool_n])
+            loss = train_step()
+    torch.cuda.current_stream().wait_stream(side)
     torch.cuda.synchronize()
+    del loss
+    loss_static = None
 
     graph = None
     if mode == "graph":
@@ -161,8 +171,10 @@ def main():
     ms_per_step = elapsed_s / args.steps * 1000.0
     global_batch = bs * world
     ips = global_batch * args.steps / elapsed_s
-    final_loss = float((loss_static if graph is not None else loss)
-                       .detach().float().cpu())
+    if graph is not None:
+        final_loss = float(loss_static.detach().float().cpu())
+    else:
+        final_loss = float(train_step().detach().float().cpu())
 
     if rank == 0:
         out = {

@@ -92,6 +92,7 @@ def test_conv_bn_eval_mode():
     (64, 128, 3, 2, 8),
     (128, 128, 1, 1, 4),
     (256, 512, 3, 2, 2),
+    (512, 512, 3, 1, 1),  # layer4: 1x1 spatial, wgrad M=batch only
     (3, 64, 7, 2, 32),    # stem: scalar-gather wgrad path
 ])
 def test_conv_bn_relu_bwd(cfg):
@@ -270,7 +271,7 @@ def test_resnet18_step_matches_cpu():
         return torch.dot(a, b) / (a.norm() * b.norm()).clamp_min(1e-12)
 
     checks = [("tail.fc.weight", 0.08, None),
-              ("layer4.1.conv2.weight", 0.15, None),
+              ("layer4.1.conv2.weight", None, 0.97),
               ("layer1.0.conv1.weight", None, 0.95),
               ("stem.conv.weight", None, 0.90)]
     for name, rtol, ctol in checks:
