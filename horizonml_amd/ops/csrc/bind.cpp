@@ -14,15 +14,26 @@ struct ConvP {
 };
 
 extern "C" {
+int conv_fwd_splitk(ConvP);
+int conv_dgrad_splitk(ConvP);
 void launch_conv_fwd(const void*, const void*, void*, float*, ConvP,
                      hipStream_t);
-void launch_conv_dgrad(const void*, const void*, void*, ConvP, hipStream_t);
+void launch_conv_fwd_splitk(const void*, const void*, float*, ConvP, int,
+                            hipStream_t);
+void launch_conv_dgrad(const void*, const void*, void*, float*, int, ConvP,
+                       hipStream_t);
 void launch_gemm_bf16(const void*, const void*, void*, int, int, int,
                       hipStream_t);
 void launch_wgrad(const void*, const void*, float*, ConvP, hipStream_t);
 void launch_bn_apply(const void*, const void*, void*, const float*,
                      const float*, const float*, float*, float*, float*,
                      float*, long, int, float, float, int, int, hipStream_t);
+void launch_bn_apply_f32(const float*, const void*, void*, void*,
+                         const float*, const float*, const float*, float*,
+                         float*, float*, float*, long, int, float, float,
+                         int, int, hipStream_t);
+void launch_stats_reduce(const float*, float*, long, int, hipStream_t);
+void launch_cast_f32_bf16(const float*, void*, long, hipStream_t);
 void launch_bnact_bwd_reduce(const void*, const void*, const void*,
                              const float*, const float*, float*, float*, long,
                              int, int, hipStream_t);
@@ -135,16 +146,35 @@ std::vector<Tensor> conv_bn_act_fwd(
     res_ptr = residual->data_ptr();
   }
   auto st = cur_stream();
-  launch_conv_fwd(x.data_ptr(), w.data_ptr(), convout.data_ptr(), stats_ptr,
-                  p, st);
-  launch_bn_apply(convout.data_ptr(), res_ptr, y.data_ptr(), stats_ptr,
-                  gamma.data_ptr<float>(), beta.data_ptr<float>(),
-                  running_mean.data_ptr<float>(),
-                  running_var.data_ptr<float>(),
-                  training ? smean.data_ptr<float>() : nullptr,
-                  training ? sinvstd.data_ptr<float>() : nullptr,
-                  (long)p.M, K, (float)momentum, (float)eps, training ? 1 : 0,
-                  act ? 1 : 0, st);
+  int splitk = conv_fwd_splitk(p);
+  if (splitk > 1) {
+    // latency path: f32 atomic partials -> stats reduce -> BN apply (+cast)
+    Tensor ws = at::zeros({(long)p.M * K}, fopt);
+    launch_conv_fwd_splitk(x.data_ptr(), w.data_ptr(), ws.data_ptr<float>(),
+                           p, splitk, st);
+    if (training)
+      launch_stats_reduce(ws.data_ptr<float>(), stats_ptr, (long)p.M, K, st);
+    launch_bn_apply_f32(ws.data_ptr<float>(), res_ptr, y.data_ptr(),
+                        convout.data_ptr(), stats_ptr,
+                        gamma.data_ptr<float>(), beta.data_ptr<float>(),
+                        running_mean.data_ptr<float>(),
+                        running_var.data_ptr<float>(),
+                        training ? smean.data_ptr<float>() : nullptr,
+                        training ? sinvstd.data_ptr<float>() : nullptr,
+                        (long)p.M, K, (float)momentum, (float)eps,
+                        training ? 1 : 0, act ? 1 : 0, st);
+  } else {
+    launch_conv_fwd(x.data_ptr(), w.data_ptr(), convout.data_ptr(),
+                    stats_ptr, p, st);
+    launch_bn_apply(convout.data_ptr(), res_ptr, y.data_ptr(), stats_ptr,
+                    gamma.data_ptr<float>(), beta.data_ptr<float>(),
+                    running_mean.data_ptr<float>(),
+                    running_var.data_ptr<float>(),
+                    training ? smean.data_ptr<float>() : nullptr,
+                    training ? sinvstd.data_ptr<float>() : nullptr,
+                    (long)p.M, K, (float)momentum, (float)eps,
+                    training ? 1 : 0, act ? 1 : 0, st);
+  }
   return {y, convout, smean, sinvstd};
 }
 
@@ -195,8 +225,17 @@ std::vector<Tensor> conv_bn_act_bwd(
     ConvP pd = p;
     pd.M = p.Nb * p.H * p.W;
     pd.Kd = R * S * K;
-    launch_conv_dgrad(dconv.data_ptr(), w_rsck.data_ptr(), dx.data_ptr(), pd,
-                      st);
+    int splitk = conv_dgrad_splitk(p);
+    if (splitk > 1) {
+      Tensor wsd = at::zeros({(long)pd.M * p.C}, fopt);
+      launch_conv_dgrad(dconv.data_ptr(), w_rsck.data_ptr(), nullptr,
+                        wsd.data_ptr<float>(), splitk, pd, st);
+      launch_cast_f32_bf16(wsd.data_ptr<float>(), dx.data_ptr(),
+                           (long)pd.M * p.C, st);
+    } else {
+      launch_conv_dgrad(dconv.data_ptr(), w_rsck.data_ptr(), dx.data_ptr(),
+                        nullptr, 1, pd, st);
+    }
   }
   // dgamma = Σ dz·xhat, dbeta = Σ dz
   return {dx, dw, sum_dzx, sum_dz, dres};

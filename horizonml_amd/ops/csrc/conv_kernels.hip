@@ -7,20 +7,29 @@
 //
 // Geometry: 64×64 block tile, BK=32 K-steps, 256 threads = 4 waves in 2×2,
 // each wave computes a 32×32 sub-tile as 2×2 mfma_f32_16x16x32_bf16
-// fragments accumulating in f32.  A/B tiles are LDS-staged with +8-element
-// row padding (row stride 80 B → 16-lane ds_read_b128 groups land on
-// distinct banks).  The forward kernel's epilogue also accumulates the
-// per-channel BN batch statistics (Σy, Σy²) with wave-level shuffle
-// reduction + one atomicAdd per 16-lane group, so a training conv+BN block
-// is 2 kernels total (this + bn_apply) instead of torch's ~5
-// (north-star fused conv+BN+ReLU requirement).
+// fragments.  A/B tiles are LDS-staged with +8-element row padding.
+//
+// Latency design (these conv shapes are tiny — batch 64 CIFAR gives
+// M=64..16k, so kernels are latency-bound, not throughput-bound):
+//  * register-staged prefetch: tile t+1's global loads are issued right
+//    after the barrier so their latency hides under tile t's ds_read+MFMA
+//    (guide §5.5 T14 minimum form);
+//  * split-K: when the tile grid would underfill the 256 CUs, the K
+//    reduction is partitioned over blockIdx.z and f32 partials are
+//    atomically combined in an [M][N] f32 workspace; BN statistics +
+//    bf16 conversion then happen in the bn_apply pass (bind.cpp decides).
+//
+// The non-split forward epilogue also accumulates per-channel BN batch
+// statistics (Σy, Σy²) via wave shuffle reduction + one atomicAdd per
+// 16-lane group, so a training conv+BN block is 2 kernels total
+// (north-star fused conv+BN+ReLU).
 #include "common.h"
 
 struct ConvP {
   int Nb, H, W, C, K;      // batch, input spatial, in/out channels
   int Ho, Wo, R, S;        // output spatial, filter
   int str, pad;
-  int M, Kd;               // GEMM rows (=Nb*Ho*Wo fwd, Nb*H*W dgrad), reduction size
+  int M, Kd;               // GEMM rows, reduction size
 };
 
 // ---------------------------------------------------------------- staging --
@@ -62,15 +71,12 @@ DEV int b_addr(const ConvP& p, int n, int k, int Ntot, bool& valid) {
     valid = (n < Ntot) & (k < p.Kd);
     return (rs * p.C + n) * p.K + ko;
   }
-  // MODE 0/1: row-major [Ntot][Kd] (KRSC weights are exactly this for fwd)
   valid = (n < Ntot) & (k < p.Kd);
   return n * p.Kd + k;
 }
 
-// Load 8 contiguous-k elements (vector when layout allows, scalar otherwise).
 template <int MODE, bool VEC>
-DEV void stage8_a(bf16* dst, const bf16* __restrict__ src, const ConvP& p,
-                  int m, int k) {
+DEV V8 load8_a(const bf16* __restrict__ src, const ConvP& p, int m, int k) {
   V8 v;
   if (VEC) {
     bool ok;
@@ -84,12 +90,12 @@ DEV void stage8_a(bf16* dst, const bf16* __restrict__ src, const ConvP& p,
       v.e[e] = ok ? src[a] : (bf16)0.f;
     }
   }
-  *(V8*)dst = v;
+  return v;
 }
 
 template <int MODE, bool VEC>
-DEV void stage8_b(bf16* dst, const bf16* __restrict__ src, const ConvP& p,
-                  int n, int k, int Ntot) {
+DEV V8 load8_b(const bf16* __restrict__ src, const ConvP& p, int n, int k,
+               int Ntot) {
   V8 v;
   if (VEC) {
     bool ok;
@@ -103,34 +109,46 @@ DEV void stage8_b(bf16* dst, const bf16* __restrict__ src, const ConvP& p,
       v.e[e] = ok ? src[a] : (bf16)0.f;
     }
   }
-  *(V8*)dst = v;
+  return v;
 }
 
 // ------------------------------------------------------------- main tile --
-// LDS row stride 40 elems (32 + 8 pad) = 80 B.
-#define LDA 40
+#define LDA 40  // 32 + 8 pad; row stride 80 B
 
-template <int MODE, bool VECA, bool VECB, bool STATS>
+// SPLIT: accumulate f32 partials into `ws_out` ([M][Ntot] f32) with atomics;
+// otherwise write bf16 into Y (+ optional stats).
+template <int MODE, bool VECA, bool VECB, bool STATS, bool SPLIT>
 __global__ __launch_bounds__(256) void k_conv_mfma(
     const bf16* __restrict__ A, const bf16* __restrict__ Bw,
-    bf16* __restrict__ Y, float* __restrict__ stats, ConvP p, int Ntot) {
+    bf16* __restrict__ Y, float* __restrict__ ws_out,
+    float* __restrict__ stats, ConvP p, int Ntot, int kchunk) {
   __shared__ bf16 As[64 * LDA];
   __shared__ bf16 Bs[64 * LDA];
 
   const int m0 = blockIdx.y * 64, n0 = blockIdx.x * 64;
+  const int kbeg = SPLIT ? blockIdx.z * kchunk : 0;
+  const int kend = SPLIT ? min(p.Kd, kbeg + kchunk) : p.Kd;
   const int tid = threadIdx.x;
-  const int srow = tid >> 2, scol = (tid & 3) * 8;  // staging: 1 vec8 each
+  const int srow = tid >> 2, scol = (tid & 3) * 8;  // 1 vec8 per thread
   const int lane = tid & 63, wave = tid >> 6;
   const int wr = wave >> 1, wc = wave & 1;
   const int fr = lane & 15, fk = lane >> 4;
 
   f32x4 acc[2][2] = {};
 
-  for (int k0 = 0; k0 < p.Kd; k0 += 32) {
-    stage8_a<MODE, VECA>(&As[srow * LDA + scol], A, p, m0 + srow, k0 + scol);
-    stage8_b<MODE, VECB>(&Bs[srow * LDA + scol], Bw, p, n0 + srow, k0 + scol,
-                         Ntot);
+  // prologue: prefetch the first K-tile into registers
+  V8 a_nx = load8_a<MODE, VECA>(A, p, m0 + srow, kbeg + scol);
+  V8 b_nx = load8_b<MODE, VECB>(Bw, p, n0 + srow, kbeg + scol, Ntot);
+
+  for (int k0 = kbeg; k0 < kend; k0 += 32) {
+    *(V8*)&As[srow * LDA + scol] = a_nx;
+    *(V8*)&Bs[srow * LDA + scol] = b_nx;
     __syncthreads();
+    // issue next tile's loads now: latency hides under ds_read + MFMA
+    if (k0 + 32 < kend) {
+      a_nx = load8_a<MODE, VECA>(A, p, m0 + srow, k0 + 32 + scol);
+      b_nx = load8_b<MODE, VECB>(Bw, p, n0 + srow, k0 + 32 + scol, Ntot);
+    }
     bf16x8 af[2], bf[2];
 #pragma unroll
     for (int mi = 0; mi < 2; mi++)
@@ -159,17 +177,20 @@ __global__ __launch_bounds__(256) void k_conv_mfma(
         int gm = m0 + wr * 32 + mi * 16 + fk * 4 + q;
         if (gm < p.M && gn < Ntot) {
           float v = acc[mi][ni][q];
-          Y[(long)gm * Ntot + gn] = f2b(v);
-          if (STATS) {
-            ssum[ni] += v;
-            ssq[ni] += v * v;
+          if (SPLIT) {
+            atomicAdd(&ws_out[(long)gm * Ntot + gn], v);
+          } else {
+            Y[(long)gm * Ntot + gn] = f2b(v);
+            if (STATS) {
+              ssum[ni] += v;
+              ssq[ni] += v * v;
+            }
           }
         }
       }
     }
   }
-  if (STATS && stats != nullptr) {
-    // reduce the 4 lanes sharing a column (l, l+16, l+32, l+48)
+  if (STATS && !SPLIT && stats != nullptr) {
 #pragma unroll
     for (int ni = 0; ni < 2; ni++) {
       float s = ssum[ni] + __shfl_xor(ssum[ni], 16, 64);
@@ -186,12 +207,10 @@ __global__ __launch_bounds__(256) void k_conv_mfma(
 }
 
 // ------------------------------------------------------------------ wgrad --
-// dW[ko][k3] += Σ_m patch(m,k3) · dz(m,ko);  dW is KRSC flat [Ntot][Kd] f32
-// (identical layout to the f32 master parameter → autograd-direct).
-// VALU outer-product v1: 64×64 tile per block, 4×4 f32 accumulators per
-// thread, M reduced in 32-deep LDS-staged chunks, msplit-way M parallelism
-// with f32 atomics.  (MFMA tr-read upgrade is a follow-up; wgrad ≈ 1/3 of
-// backward FLOPs at these shapes.)
+// dW[ko][k3] += Σ_m patch(m,k3) · dz(m,ko);  dW is KRSC flat [Ntot][Kd] f32.
+// VALU outer-product: 64×64 tile, 4×4 f32 acc/thread, M in 32-deep LDS
+// chunks, msplit-way M parallelism with f32 atomics.  LDS reads are b64
+// (4×bf16) per operand — 4 LDS ops per 16 FMA.
 #define LDW 72
 
 template <bool VECA>
@@ -204,29 +223,40 @@ __global__ __launch_bounds__(256) void k_wgrad(
   const int mbeg = blockIdx.z * mchunk;
   const int mend = min(p.M, mbeg + mchunk);
   const int tid = threadIdx.x;
-  const int srow = tid >> 3, scol = (tid & 7) * 8;  // 32×64 tile: 1 vec8 each
+  const int srow = tid >> 3, scol = (tid & 7) * 8;
   const int trow = (tid >> 4) * 4, tcol = (tid & 15) * 4;
 
+  typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4;
   float acc[4][4] = {};
+  // register prefetch across chunks
+  V8 a_nx = load8_a<1, VECA>(X, p, mbeg + srow, k3_0 + scol);
+  V8 d_nx;
+  {
+    int m = mbeg + srow, n = n0 + scol;
+    if (m < p.M && n < Ntot) d_nx.u = *(const uint4*)(Dz + (long)m * Ntot + n);
+    else d_nx.u = uint4{0, 0, 0, 0};
+  }
   for (int m0 = mbeg; m0 < mend; m0 += 32) {
-    stage8_a<1, VECA>(&As[srow * LDW + scol], X, p, m0 + srow, k3_0 + scol);
-    {
-      V8 v;
-      int m = m0 + srow, n = n0 + scol;
-      if (m < p.M && n < Ntot)
-        v.u = *(const uint4*)(Dz + (long)m * Ntot + n);
-      else
-        v.u = uint4{0, 0, 0, 0};
-      *(V8*)&Ds[srow * LDW + scol] = v;
-    }
+    *(V8*)&As[srow * LDW + scol] = a_nx;
+    *(V8*)&Ds[srow * LDW + scol] = d_nx;
     __syncthreads();
+    if (m0 + 32 < mend) {
+      a_nx = load8_a<1, VECA>(X, p, m0 + 32 + srow, k3_0 + scol);
+      int m = m0 + 32 + srow, n = n0 + scol;
+      if (m < p.M && n < Ntot)
+        d_nx.u = *(const uint4*)(Dz + (long)m * Ntot + n);
+      else
+        d_nx.u = uint4{0, 0, 0, 0};
+    }
 #pragma unroll 4
     for (int mm = 0; mm < 32; mm++) {
+      bf16x4 av = *(const bf16x4*)&As[mm * LDW + trow];
+      bf16x4 dv = *(const bf16x4*)&Ds[mm * LDW + tcol];
       float a[4], d[4];
 #pragma unroll
-      for (int i = 0; i < 4; i++) a[i] = b2f(As[mm * LDW + trow + i]);
+      for (int i = 0; i < 4; i++) a[i] = (float)av[i];
 #pragma unroll
-      for (int j = 0; j < 4; j++) d[j] = b2f(Ds[mm * LDW + tcol + j]);
+      for (int j = 0; j < 4; j++) d[j] = (float)dv[j];
 #pragma unroll
       for (int i = 0; i < 4; i++)
 #pragma unroll
@@ -247,38 +277,92 @@ __global__ __launch_bounds__(256) void k_wgrad(
 // ------------------------------------------------------------- launchers --
 static inline int cdiv_h(int a, int b) { return (a + b - 1) / b; }
 
+// Split-K heuristic: fill ≥256 workgroups when the tile grid alone cannot.
+static inline int pick_splitk(int tiles, int Kd) {
+  int target = 256;
+  int sk = tiles >= target ? 1 : cdiv_h(target, tiles);
+  int maxsk = cdiv_h(Kd, 32);
+  if (sk > maxsk) sk = maxsk;
+  if (sk > 32) sk = 32;
+  return sk;
+}
+
+extern "C" int conv_fwd_splitk(ConvP p) {
+  int tiles = cdiv_h(p.K, 64) * cdiv_h(p.M, 64);
+  return pick_splitk(tiles, p.Kd);
+}
+
+extern "C" int conv_dgrad_splitk(ConvP p) {
+  // dgrad geometry: M=Nb*H*W rows, Kd=R*S*K, N=C
+  int tiles = cdiv_h(p.C, 64) * cdiv_h(p.Nb * p.H * p.W, 64);
+  return pick_splitk(tiles, p.R * p.S * p.K);
+}
+
+// Non-split forward (bf16 out + fused stats).
 extern "C" void launch_conv_fwd(const void* x, const void* w, void* y,
                                 float* stats, ConvP p, hipStream_t st) {
   dim3 grid(cdiv_h(p.K, 64), cdiv_h(p.M, 64));
-  bool veca = (p.C % 8) == 0;
-  bool vecb = (p.Kd % 8) == 0;
+  bool vec = (p.C % 8) == 0 && (p.Kd % 8) == 0;
   bool s = stats != nullptr;
   auto A = (const bf16*)x;
   auto B = (const bf16*)w;
   auto Y = (bf16*)y;
-#define CASE(VA, VB, ST)                                                   \
-  k_conv_mfma<1, VA, VB, ST><<<grid, 256, 0, st>>>(A, B, Y, stats, p, p.K)
-  if (veca && vecb && s) CASE(true, true, true);
-  else if (veca && vecb) CASE(true, true, false);
+#define CASE(VA, VB, ST)                                     \
+  k_conv_mfma<1, VA, VB, ST, false><<<grid, 256, 0, st>>>(   \
+      A, B, Y, nullptr, stats, p, p.K, 0)
+  if (vec && s) CASE(true, true, true);
+  else if (vec) CASE(true, true, false);
   else if (s) CASE(false, false, true);
   else CASE(false, false, false);
 #undef CASE
 }
 
-extern "C" void launch_conv_dgrad(const void* dz, const void* w_rsck, void* dx,
-                                  ConvP p, hipStream_t st) {
+// Split-K forward: f32 atomic partials into ws (pre-zeroed [M][K]).
+extern "C" void launch_conv_fwd_splitk(const void* x, const void* w,
+                                       float* ws, ConvP p, int splitk,
+                                       hipStream_t st) {
+  int kchunk = cdiv_h(cdiv_h(p.Kd, splitk), 32) * 32;
+  splitk = cdiv_h(p.Kd, kchunk);
+  dim3 grid(cdiv_h(p.K, 64), cdiv_h(p.M, 64), splitk);
+  bool vec = (p.C % 8) == 0 && (p.Kd % 8) == 0;
+  auto A = (const bf16*)x;
+  auto B = (const bf16*)w;
+  if (vec)
+    k_conv_mfma<1, true, true, false, true>
+        <<<grid, 256, 0, st>>>(A, B, nullptr, ws, nullptr, p, p.K, kchunk);
+  else
+    k_conv_mfma<1, false, false, false, true>
+        <<<grid, 256, 0, st>>>(A, B, nullptr, ws, nullptr, p, p.K, kchunk);
+}
+
+extern "C" void launch_conv_dgrad(const void* dz, const void* w_rsck,
+                                  void* dx, float* ws, int splitk, ConvP p,
+                                  hipStream_t st) {
   // p here: M = Nb*H*W, Kd = R*S*K, output channels = C
-  dim3 grid(cdiv_h(p.C, 64), cdiv_h(p.M, 64));
   bool vec = (p.K % 8) == 0;
   auto A = (const bf16*)dz;
   auto B = (const bf16*)w_rsck;
-  auto Y = (bf16*)dx;
-  if (vec)
-    k_conv_mfma<2, true, true, false>
-        <<<grid, 256, 0, st>>>(A, B, Y, nullptr, p, p.C);
-  else
-    k_conv_mfma<2, false, false, false>
-        <<<grid, 256, 0, st>>>(A, B, Y, nullptr, p, p.C);
+  if (splitk > 1) {
+    int kchunk = cdiv_h(cdiv_h(p.Kd, splitk), 32) * 32;
+    splitk = cdiv_h(p.Kd, kchunk);
+    dim3 grid(cdiv_h(p.C, 64), cdiv_h(p.M, 64), splitk);
+    if (vec)
+      k_conv_mfma<2, true, true, false, true>
+          <<<grid, 256, 0, st>>>(A, B, nullptr, ws, nullptr, p, p.C, kchunk);
+    else
+      k_conv_mfma<2, false, false, false, true>
+          <<<grid, 256, 0, st>>>(A, B, nullptr, ws, nullptr, p, p.C, kchunk);
+  } else {
+    dim3 grid(cdiv_h(p.C, 64), cdiv_h(p.M, 64));
+    if (vec)
+      k_conv_mfma<2, true, true, false, false>
+          <<<grid, 256, 0, st>>>(A, B, (bf16*)dx, nullptr, nullptr, p, p.C,
+                                 0);
+    else
+      k_conv_mfma<2, false, false, false, false>
+          <<<grid, 256, 0, st>>>(A, B, (bf16*)dx, nullptr, nullptr, p, p.C,
+                                 0);
+  }
 }
 
 extern "C" void launch_gemm_bf16(const void* a, const void* b, void* c, int M,
@@ -289,13 +373,13 @@ extern "C" void launch_gemm_bf16(const void* a, const void* b, void* c, int M,
   dim3 grid(cdiv_h(N, 64), cdiv_h(M, 64));
   bool vec = (K % 8) == 0;
   if (vec)
-    k_conv_mfma<0, true, true, false>
+    k_conv_mfma<0, true, true, false, false>
         <<<grid, 256, 0, st>>>((const bf16*)a, (const bf16*)b, (bf16*)c,
-                               nullptr, p, N);
+                               nullptr, nullptr, p, N, 0);
   else
-    k_conv_mfma<0, false, false, false>
+    k_conv_mfma<0, false, false, false, false>
         <<<grid, 256, 0, st>>>((const bf16*)a, (const bf16*)b, (bf16*)c,
-                               nullptr, p, N);
+                               nullptr, nullptr, p, N, 0);
 }
 
 extern "C" void launch_wgrad(const void* x, const void* dz, float* dw, ConvP p,
