@@ -58,10 +58,21 @@ class ConvBNActFn(torch.autograd.Function):
         if need_dx and (w_rsck is None or w_rsck.numel() != w_bf16.numel()):
             w_rsck = w_bf16.permute(1, 2, 3, 0).contiguous()
         dy = _cl(dy.to(torch.bfloat16) if dy.dtype != torch.bfloat16 else dy)
+        # Direct-grad mode (FlatParamManager): kernels accumulate straight
+        # into the pre-zeroed flat .grad views and we return None so autograd
+        # skips its per-parameter accumulate kernels (~60 launches/step).
+        direct = (getattr(mod, "_managed", False)
+                  and mod.weight.grad is not None)
+        dw_out = mod.weight.grad if direct else None
+        dg_out = mod.bn_weight.grad if direct else None
+        db_out = mod.bn_bias.grad if direct else None
         dx, dw, dgamma, dbeta, dres = _C().conv_bn_act_bwd(
             dy, y, x, w_bf16, w_rsck if need_dx else w_bf16, convout, gamma,
             smean, sinvstd, mod.stride, mod.padding, mod.act, need_dx,
-            ctx.has_res)
+            ctx.has_res, dw_out, dg_out, db_out)
+        if direct:
+            return (dx if need_dx else None, None, None, None,
+                    dres if ctx.has_res else None, None)
         return (dx if need_dx else None, dw, dgamma, dbeta,
                 dres if ctx.has_res else None, None)
 
@@ -116,14 +127,24 @@ class LinearFn(torch.autograd.Function):
         y = _C().linear_fwd(x, w_bf16, bias)
         ctx.save_for_backward(x, w_bf16)
         ctx.need_db = bias is not None
+        ctx.mod = mod
         return y
 
     @staticmethod
     def backward(ctx, dy):
         x, w_bf16 = ctx.saved_tensors
+        mod = ctx.mod
         dy = dy.float().contiguous()
         need_dx = ctx.needs_input_grad[0]
-        dx, dw, db = _C().linear_bwd(dy, x, w_bf16, need_dx, ctx.need_db)
+        direct = (getattr(mod, "_managed", False)
+                  and mod.weight.grad is not None)
+        dw_out = mod.weight.grad if direct else None
+        db_out = (mod.bias.grad if (direct and mod.bias is not None)
+                  else None)
+        dx, dw, db = _C().linear_bwd(dy, x, w_bf16, need_dx, ctx.need_db,
+                                     dw_out, db_out)
+        if direct:
+            return (dx if need_dx else None, None, None, None)
         return (dx if need_dx else None, dw,
                 db if ctx.need_db else None, None)
 

@@ -31,9 +31,9 @@ void launch_bn_apply(const void*, const void*, void*, const float*,
 void launch_bn_apply_f32(const float*, const void*, void*, void*,
                          const float*, const float*, const float*, float*,
                          float*, float*, float*, long, int, float, float,
-                         int, int, hipStream_t);
-void launch_stats_reduce(const float*, float*, long, int, hipStream_t);
-void launch_cast_f32_bf16(const float*, void*, long, hipStream_t);
+                         int, int, int, hipStream_t);
+void launch_stats_reduce(const float*, float*, long, int, int, hipStream_t);
+void launch_cast_f32_bf16(const float*, void*, long, int, hipStream_t);
 void launch_bnact_bwd_reduce(const void*, const void*, const void*,
                              const float*, const float*, float*, float*, long,
                              int, int, hipStream_t);
@@ -50,7 +50,7 @@ void launch_avgpool_bwd(const void*, void*, int, int, int, hipStream_t);
 void launch_linear_fwd(const void*, const void*, const float*, float*, int,
                        int, int, hipStream_t);
 void launch_linear_bwd(const float*, const void*, const void*, void*, float*,
-                       float*, int, int, int, hipStream_t);
+                       float*, int, int, int, int, hipStream_t);
 void launch_ce_fwd_bwd(const float*, const long*, float*, float*, int, int,
                        hipStream_t);
 void launch_adam_step(float*, float*, float*, float*, void*, const float*,
@@ -148,12 +148,16 @@ std::vector<Tensor> conv_bn_act_fwd(
   auto st = cur_stream();
   int splitk = conv_fwd_splitk(p);
   if (splitk > 1) {
-    // latency path: f32 atomic partials -> stats reduce -> BN apply (+cast)
-    Tensor ws = at::zeros({(long)p.M * K}, fopt);
+    int kchunk = ((p.Kd + splitk - 1) / splitk + 31) / 32 * 32;
+    splitk = (p.Kd + kchunk - 1) / kchunk;
+    // latency path: per-split f32 slabs -> stats reduce -> BN apply (+cast);
+    // slabs are fully written by the conv grid, so no zeroing needed
+    Tensor ws = at::empty({(long)splitk * p.M * K}, fopt);
     launch_conv_fwd_splitk(x.data_ptr(), w.data_ptr(), ws.data_ptr<float>(),
                            p, splitk, st);
     if (training)
-      launch_stats_reduce(ws.data_ptr<float>(), stats_ptr, (long)p.M, K, st);
+      launch_stats_reduce(ws.data_ptr<float>(), stats_ptr, (long)p.M, K,
+                          splitk, st);
     launch_bn_apply_f32(ws.data_ptr<float>(), res_ptr, y.data_ptr(),
                         convout.data_ptr(), stats_ptr,
                         gamma.data_ptr<float>(), beta.data_ptr<float>(),
@@ -162,7 +166,7 @@ std::vector<Tensor> conv_bn_act_fwd(
                         training ? smean.data_ptr<float>() : nullptr,
                         training ? sinvstd.data_ptr<float>() : nullptr,
                         (long)p.M, K, (float)momentum, (float)eps,
-                        training ? 1 : 0, act ? 1 : 0, st);
+                        training ? 1 : 0, act ? 1 : 0, splitk, st);
   } else {
     launch_conv_fwd(x.data_ptr(), w.data_ptr(), convout.data_ptr(),
                     stats_ptr, p, st);
@@ -181,7 +185,12 @@ std::vector<Tensor> conv_bn_act_fwd(
 std::vector<Tensor> conv_bn_act_bwd(
     Tensor dy, Tensor y, Tensor x, Tensor w, Tensor w_rsck, Tensor convout,
     Tensor gamma, Tensor save_mean, Tensor save_invstd, int64_t stride,
-    int64_t pad, bool act, bool need_dx, bool has_res) {
+    int64_t pad, bool act, bool need_dx, bool has_res,
+    c10::optional<Tensor> dw_out, c10::optional<Tensor> dgamma_out,
+    c10::optional<Tensor> dbeta_out) {
+  // Direct-grad mode: when dw_out/dgamma_out/dbeta_out are given they are
+  // PRE-ZEROED flat .grad views — the kernels accumulate straight into
+  // them, skipping autograd's per-parameter accumulate pass.
   check_cl(dy, "dy");
   check_cl(y, "y");
   check_cl(x, "x");
@@ -191,8 +200,9 @@ std::vector<Tensor> conv_bn_act_bwd(
   auto fopt = x.options().dtype(torch::kFloat32);
   auto st = cur_stream();
 
-  Tensor sum_dz = at::empty({K}, fopt);
-  Tensor sum_dzx = at::empty({K}, fopt);
+  bool direct = dw_out.has_value();
+  Tensor sum_dz = direct ? *dbeta_out : at::zeros({K}, fopt);
+  Tensor sum_dzx = direct ? *dgamma_out : at::zeros({K}, fopt);
   launch_bnact_bwd_reduce(dy.data_ptr(), y.data_ptr(), convout.data_ptr(),
                           save_mean.data_ptr<float>(),
                           save_invstd.data_ptr<float>(),
@@ -213,7 +223,8 @@ std::vector<Tensor> conv_bn_act_bwd(
                       dconv.data_ptr(), dres_ptr, (long)p.M, K, act ? 1 : 0,
                       st);
 
-  Tensor dw = at::zeros({(int64_t)K, R, S, (int64_t)p.C}, fopt);
+  Tensor dw = direct ? *dw_out
+                     : at::zeros({(int64_t)K, R, S, (int64_t)p.C}, fopt);
   launch_wgrad(x.data_ptr(), dconv.data_ptr(), dw.data_ptr<float>(), p, st);
 
   Tensor dx;
@@ -227,11 +238,13 @@ std::vector<Tensor> conv_bn_act_bwd(
     pd.Kd = R * S * K;
     int splitk = conv_dgrad_splitk(p);
     if (splitk > 1) {
-      Tensor wsd = at::zeros({(long)pd.M * p.C}, fopt);
+      int kchunk = ((pd.Kd + splitk - 1) / splitk + 31) / 32 * 32;
+      splitk = (pd.Kd + kchunk - 1) / kchunk;
+      Tensor wsd = at::empty({(long)splitk * pd.M * p.C}, fopt);
       launch_conv_dgrad(dconv.data_ptr(), w_rsck.data_ptr(), nullptr,
                         wsd.data_ptr<float>(), splitk, pd, st);
       launch_cast_f32_bf16(wsd.data_ptr<float>(), dx.data_ptr(),
-                           (long)pd.M * p.C, st);
+                           (long)pd.M * p.C, splitk, st);
     } else {
       launch_conv_dgrad(dconv.data_ptr(), w_rsck.data_ptr(), dx.data_ptr(),
                         nullptr, 1, pd, st);
@@ -302,12 +315,15 @@ Tensor linear_fwd(Tensor x, Tensor w, c10::optional<Tensor> bias) {
 }
 
 std::vector<Tensor> linear_bwd(Tensor dy, Tensor x, Tensor w, bool need_dx,
-                               bool need_db) {
+                               bool need_db, c10::optional<Tensor> dw_out,
+                               c10::optional<Tensor> db_out) {
   check_f32(dy, "dy");
   int B = (int)x.size(0), In = (int)x.size(1), Out = (int)w.size(0);
   auto fopt = x.options().dtype(torch::kFloat32);
-  Tensor dx, dw = at::empty({Out, In}, fopt);
-  Tensor db = need_db ? at::empty({Out}, fopt) : Tensor();
+  bool direct = dw_out.has_value();
+  Tensor dx, dw = direct ? *dw_out : at::empty({Out, In}, fopt);
+  Tensor db = direct ? (db_out.has_value() ? *db_out : Tensor())
+                     : (need_db ? at::empty({Out}, fopt) : Tensor());
   void* dx_ptr = nullptr;
   if (need_dx) {
     dx = at::empty({B, In}, x.options());
@@ -315,8 +331,8 @@ std::vector<Tensor> linear_bwd(Tensor dy, Tensor x, Tensor w, bool need_dx,
   }
   launch_linear_bwd(dy.data_ptr<float>(), x.data_ptr(), w.data_ptr(), dx_ptr,
                     dw.data_ptr<float>(),
-                    need_db ? db.data_ptr<float>() : nullptr, B, In, Out,
-                    cur_stream());
+                    db.defined() ? db.data_ptr<float>() : nullptr, B, In, Out,
+                    direct ? 1 : 0, cur_stream());
   return {dx, dw, db};
 }
 

@@ -115,8 +115,9 @@ DEV V8 load8_b(const bf16* __restrict__ src, const ConvP& p, int n, int k,
 // ------------------------------------------------------------- main tile --
 #define LDA 40  // 32 + 8 pad; row stride 80 B
 
-// SPLIT: accumulate f32 partials into `ws_out` ([M][Ntot] f32) with atomics;
-// otherwise write bf16 into Y (+ optional stats).
+// SPLIT: each blockIdx.z writes its f32 partial into its own slab
+// ws_out[z][M][Ntot] (no atomics, no pre-zeroing — the consumer pass sums
+// the slabs); otherwise write bf16 into Y (+ optional stats).
 template <int MODE, bool VECA, bool VECB, bool STATS, bool SPLIT>
 __global__ __launch_bounds__(256) void k_conv_mfma(
     const bf16* __restrict__ A, const bf16* __restrict__ Bw,
@@ -178,7 +179,7 @@ __global__ __launch_bounds__(256) void k_conv_mfma(
         if (gm < p.M && gn < Ntot) {
           float v = acc[mi][ni][q];
           if (SPLIT) {
-            atomicAdd(&ws_out[(long)gm * Ntot + gn], v);
+            ws_out[(long)blockIdx.z * p.M * Ntot + (long)gm * Ntot + gn] = v;
           } else {
             Y[(long)gm * Ntot + gn] = f2b(v);
             if (STATS) {

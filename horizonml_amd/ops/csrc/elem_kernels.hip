@@ -18,7 +18,9 @@ __global__ __launch_bounds__(256) void k_bn_apply(
     const float* __restrict__ gamma, const float* __restrict__ beta,
     float* __restrict__ running_mean, float* __restrict__ running_var,
     float* __restrict__ save_mean, float* __restrict__ save_invstd,
-    long M, int C, float momentum, float eps, int training, int act) {
+    long M, int C, float momentum, float eps, int training, int act,
+    int nsplit) {
+  const long slab = M * (long)C;
   const bf16* xb = (const bf16*)xv;
   const float* xf = (const float*)xv;
   const float invM = 1.f / (float)M;
@@ -47,7 +49,13 @@ __global__ __launch_bounds__(256) void k_bn_apply(
       mean = running_mean[c];
       invstd = rsqrtf(running_var[c] + eps);
     }
-    float xi = F32SRC ? xf[i] : b2f(xb[i]);
+    float xi;
+    if (F32SRC) {
+      xi = xf[i];
+      for (int z = 1; z < nsplit; z++) xi += xf[z * slab + i];
+    } else {
+      xi = b2f(xb[i]);
+    }
     if (F32SRC && convout != nullptr) convout[i] = f2b(xi);
     float v = (xi - mean) * invstd * gamma[c] + beta[c];
     if (res != nullptr) v += b2f(res[i]);
@@ -56,21 +64,24 @@ __global__ __launch_bounds__(256) void k_bn_apply(
   }
 }
 
-// Per-channel (Σx, Σx²) over an f32 [M][C] workspace (split-K conv path).
-// grid: (cdiv(C,64), msplit); stats must be pre-zeroed.
+// Per-channel (Σx, Σx²) over `nsplit` stacked f32 [M][C] slabs (split-K
+// conv path — slabs are summed here).  grid: (cdiv(C,64), msplit);
+// stats must be pre-zeroed.
 __global__ __launch_bounds__(256) void k_stats_reduce(
     const float* __restrict__ x, float* __restrict__ stats, long M, int C,
-    long mchunk) {
+    long mchunk, int nsplit) {
   __shared__ float s1[4][64];
   __shared__ float s2[4][64];
   const int c = blockIdx.x * 64 + (threadIdx.x & 63);
   const int mlane = threadIdx.x >> 6;
   const long mbeg = (long)blockIdx.y * mchunk;
   const long mend = min((long)M, mbeg + mchunk);
+  const long slab = (long)M * C;
   float a1 = 0.f, a2 = 0.f;
   if (c < C) {
     for (long m = mbeg + mlane; m < mend; m += 4) {
-      float v = x[m * C + c];
+      float v = 0.f;
+      for (int z = 0; z < nsplit; z++) v += x[z * slab + m * C + c];
       a1 += v;
       a2 += v * v;
     }
@@ -88,12 +99,16 @@ __global__ __launch_bounds__(256) void k_stats_reduce(
   }
 }
 
-// f32 -> bf16 elementwise (split-K dgrad output).
+// Σ over nsplit f32 slabs -> bf16 elementwise (split-K dgrad output).
 __global__ __launch_bounds__(256) void k_cast_f32_bf16(
-    const float* __restrict__ src, bf16* __restrict__ dst, long n) {
+    const float* __restrict__ src, bf16* __restrict__ dst, long n,
+    int nsplit) {
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += (long)gridDim.x * blockDim.x)
-    dst[i] = f2b(src[i]);
+       i += (long)gridDim.x * blockDim.x) {
+    float v = src[i];
+    for (int z = 1; z < nsplit; z++) v += src[z * n + i];
+    dst[i] = f2b(v);
+  }
 }
 
 // ------------------------------------------------------ BN+act backward ----
@@ -130,13 +145,10 @@ __global__ __launch_bounds__(256) void k_bnact_bwd_reduce(
                sdz[2][threadIdx.x] + sdz[3][threadIdx.x];
     float t2 = sdzx[0][threadIdx.x] + sdzx[1][threadIdx.x] +
                sdzx[2][threadIdx.x] + sdzx[3][threadIdx.x];
-    if (gridDim.y == 1) {
-      sum_dz[c] = t1;
-      sum_dzx[c] = t2;
-    } else {
-      atomicAdd(&sum_dz[c], t1);
-      atomicAdd(&sum_dzx[c], t2);
-    }
+    // always accumulate: the output buffers double as dbeta/dgamma and in
+    // direct-grad mode they are the pre-zeroed flat .grad views
+    atomicAdd(&sum_dz[c], t1);
+    atomicAdd(&sum_dzx[c], t2);
   }
 }
 
@@ -317,10 +329,13 @@ __global__ __launch_bounds__(256) void k_linear_bwd_dx(
   }
 }
 
-// dw[Out,In] f32 += Σ_b dy[b,j]·x[b,t];  db[Out] f32 = Σ_b dy[b,j]
+// dw[Out,In] f32 (+)= Σ_b dy[b,j]·x[b,t];  db[Out] f32 (+)= Σ_b dy[b,j]
+// accum=1 adds into the existing buffer (direct-grad mode: dw/db are the
+// pre-zeroed flat .grad views).
 __global__ __launch_bounds__(256) void k_linear_bwd_dw(
     const float* __restrict__ dy, const bf16* __restrict__ x,
-    float* __restrict__ dw, float* __restrict__ db, int B, int In, int Out) {
+    float* __restrict__ dw, float* __restrict__ db, int B, int In, int Out,
+    int accum) {
   long total = (long)Out * In;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += (long)gridDim.x * blockDim.x) {
@@ -328,11 +343,11 @@ __global__ __launch_bounds__(256) void k_linear_bwd_dw(
     float s = 0.f;
     for (int bi = 0; bi < B; bi++)
       s = fmaf(dy[(long)bi * Out + j], b2f(x[(long)bi * In + t]), s);
-    dw[i] = s;
+    dw[i] = accum ? dw[i] + s : s;
     if (t == 0 && db != nullptr) {
       float sb = 0.f;
       for (int bi = 0; bi < B; bi++) sb += dy[(long)bi * Out + j];
-      db[j] = sb;
+      db[j] = accum ? db[j] + sb : sb;
     }
   }
 }
@@ -495,7 +510,7 @@ void launch_bn_apply(const void* x, const void* res, void* y,
                      hipStream_t st) {
   k_bn_apply<false><<<gsz(M * (long)C), 256, 0, st>>>(
       x, (const bf16*)res, (bf16*)y, nullptr, stats, gamma, beta, rmean,
-      rvar, smean, sinvstd, M, C, momentum, eps, training, act);
+      rvar, smean, sinvstd, M, C, momentum, eps, training, act, 1);
 }
 
 void launch_bn_apply_f32(const float* ws, const void* res, void* y,
@@ -503,39 +518,37 @@ void launch_bn_apply_f32(const float* ws, const void* res, void* y,
                          const float* gamma, const float* beta, float* rmean,
                          float* rvar, float* smean, float* sinvstd, long M,
                          int C, float momentum, float eps, int training,
-                         int act, hipStream_t st) {
+                         int act, int nsplit, hipStream_t st) {
   k_bn_apply<true><<<gsz(M * (long)C), 256, 0, st>>>(
       ws, (const bf16*)res, (bf16*)y, (bf16*)convout, stats, gamma, beta,
-      rmean, rvar, smean, sinvstd, M, C, momentum, eps, training, act);
+      rmean, rvar, smean, sinvstd, M, C, momentum, eps, training, act,
+      nsplit);
 }
 
 void launch_stats_reduce(const float* ws, float* stats, long M, int C,
-                         hipStream_t st) {
+                         int nsplit, hipStream_t st) {
   int cblocks = (C + 63) / 64;
   int msplit = (int)min((long)64, max((long)1, (long)(256 / cblocks)));
   long mchunk = (M + msplit - 1) / msplit;
   msplit = (int)((M + mchunk - 1) / mchunk);
   dim3 grid(cblocks, msplit);
-  k_stats_reduce<<<grid, 256, 0, st>>>(ws, stats, M, C, mchunk);
+  k_stats_reduce<<<grid, 256, 0, st>>>(ws, stats, M, C, mchunk, nsplit);
 }
 
-void launch_cast_f32_bf16(const float* src, void* dst, long n,
+void launch_cast_f32_bf16(const float* src, void* dst, long n, int nsplit,
                           hipStream_t st) {
-  k_cast_f32_bf16<<<gsz(n), 256, 0, st>>>(src, (bf16*)dst, n);
+  k_cast_f32_bf16<<<gsz(n), 256, 0, st>>>(src, (bf16*)dst, n, nsplit);
 }
 
 void launch_bnact_bwd_reduce(const void* dy, const void* yout, const void* x,
                              const float* smean, const float* sinvstd,
                              float* sum_dz, float* sum_dzx, long M, int C,
                              int act, hipStream_t st) {
+  // sum_dz/sum_dzx must be pre-zeroed (they are accumulated atomically)
   int cblocks = (C + 63) / 64;
   int msplit = (int)min((long)64, max((long)1, (long)(256 / cblocks)));
   long mchunk = (M + msplit - 1) / msplit;
   msplit = (int)((M + mchunk - 1) / mchunk);
-  if (msplit > 1) {
-    hipMemsetAsync(sum_dz, 0, C * sizeof(float), st);
-    hipMemsetAsync(sum_dzx, 0, C * sizeof(float), st);
-  }
   dim3 grid(cblocks, msplit);
   k_bnact_bwd_reduce<<<grid, 256, 0, st>>>(
       (const bf16*)dy, (const bf16*)yout, (const bf16*)x, smean, sinvstd,
@@ -585,12 +598,12 @@ void launch_linear_fwd(const void* x, const void* w, const float* b, float* y,
 
 void launch_linear_bwd(const float* dy, const void* x, const void* w,
                        void* dx, float* dw, float* db, int B, int In, int Out,
-                       hipStream_t st) {
+                       int accum, hipStream_t st) {
   if (dx)
     k_linear_bwd_dx<<<gsz((long)B * In), 256, 0, st>>>(dy, (const bf16*)w,
                                                        (bf16*)dx, B, In, Out);
   k_linear_bwd_dw<<<gsz((long)Out * In), 256, 0, st>>>(dy, (const bf16*)x, dw,
-                                                       db, B, In, Out);
+                                                       db, B, In, Out, accum);
 }
 
 void launch_ce_fwd_bwd(const float* logits, const long* target, float* loss,
