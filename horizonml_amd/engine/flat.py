@@ -56,6 +56,16 @@ class FlatParamManager:
                 m._managed = True
                 if isinstance(m, ConvBNAct):
                     convs.append((m, woff, wn))
+        # BN-stats arena: one pre-zeroed f32 slab per conv ([2K] each),
+        # re-zeroed by the fused optimizer kernel each step (no per-conv
+        # fill kernels in the forward).
+        stats_total = sum(2 * m.out_ch for m, _, _ in convs)
+        self.stats_arena = torch.zeros(max(1, stats_total), device=device)
+        soff = 0
+        for m, _, _ in convs:
+            m._stats_buf = self.stats_arena[soff:soff + 2 * m.out_ch]
+            soff += 2 * m.out_ch
+
         rsck_total = sum(wn for _, _, wn in convs)
         self.rsck = torch.zeros(rsck_total, device=device,
                                 dtype=torch.bfloat16)
@@ -99,7 +109,8 @@ class HorizonAdam:
         _ops.extension().adam_step(self.mgr.master, self.mgr.grad, self.m,
                                    self.v, self.mgr.shadow, self.step_t,
                                    self.lr, self.betas[0], self.betas[1],
-                                   self.eps, self.wd, zero_grad)
+                                   self.eps, self.wd, zero_grad,
+                                   self.mgr.stats_arena)
         self.mgr.refresh_rsck()
 
 
@@ -114,4 +125,5 @@ class HorizonSGD:
         _ops.extension().sgd_step(self.mgr.master, self.mgr.grad, self.mom,
                                   self.mgr.shadow, self.lr, self.mu, self.wd,
                                   zero_grad)
+        self.mgr.stats_arena.zero_()
         self.mgr.refresh_rsck()

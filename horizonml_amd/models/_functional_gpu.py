@@ -36,10 +36,12 @@ class ConvBNActFn(torch.autograd.Function):
         w_bf16 = mod._shadow()
         if residual is not None:
             residual = _cl(residual)
+        stats_buf = (getattr(mod, "_stats_buf", None)
+                     if mod.training else None)
         y, convout, smean, sinvstd = _C().conv_bn_act_fwd(
             x, w_bf16, bn_weight, bn_bias, mod.running_mean, mod.running_var,
             mod.stride, mod.padding, mod.momentum, mod.eps, mod.training,
-            mod.act, residual)
+            mod.act, residual, stats_buf)
         ctx.save_for_backward(x, y, convout, w_bf16, bn_weight, smean,
                               sinvstd)
         ctx.mod = mod

@@ -54,8 +54,8 @@ void launch_linear_bwd(const float*, const void*, const void*, void*, float*,
 void launch_ce_fwd_bwd(const float*, const long*, float*, float*, int, int,
                        hipStream_t);
 void launch_adam_step(float*, float*, float*, float*, void*, const float*,
-                      long, float, float, float, float, float, int,
-                      hipStream_t);
+                      long, float, float, float, float, float, int, float*,
+                      long, hipStream_t);
 void launch_sgd_step(float*, float*, float*, void*, long, float, float, float,
                      int, hipStream_t);
 void launch_permute_krsc_rsck(const void*, void*, const int*, int, int,
@@ -115,7 +115,7 @@ std::vector<Tensor> conv_bn_act_fwd(
     Tensor x, Tensor w, Tensor gamma, Tensor beta, Tensor running_mean,
     Tensor running_var, int64_t stride, int64_t pad, double momentum,
     double eps, bool training, bool act,
-    c10::optional<Tensor> residual) {
+    c10::optional<Tensor> residual, c10::optional<Tensor> stats_buf) {
   check_cl(x, "x");
   TORCH_CHECK(w.is_cuda() && w.scalar_type() == torch::kBFloat16 &&
                   w.is_contiguous() && w.dim() == 4,
@@ -132,7 +132,8 @@ std::vector<Tensor> conv_bn_act_fwd(
   Tensor stats, smean, sinvstd;
   float* stats_ptr = nullptr;
   if (training) {
-    stats = at::zeros({2L * K}, fopt);
+    // stats_buf (manager arena) arrives pre-zeroed by the fused optimizer
+    stats = stats_buf.has_value() ? *stats_buf : at::zeros({2L * K}, fopt);
     smean = at::empty({K}, fopt);
     sinvstd = at::empty({K}, fopt);
     stats_ptr = stats.data_ptr<float>();
@@ -365,14 +366,19 @@ Tensor gemm_bf16(Tensor a, Tensor b) {
 // ------------------------------------------------------------- optimizers --
 void adam_step(Tensor master, Tensor grad, Tensor m, Tensor v,
                c10::optional<Tensor> shadow, Tensor step_t, double lr,
-               double b1, double b2, double eps, double wd, bool zero_grad) {
+               double b1, double b2, double eps, double wd, bool zero_grad,
+               c10::optional<Tensor> extra_zero) {
   check_f32(master, "master");
   launch_adam_step(master.data_ptr<float>(), grad.data_ptr<float>(),
                    m.data_ptr<float>(), v.data_ptr<float>(),
                    shadow.has_value() ? shadow->data_ptr() : nullptr,
                    step_t.data_ptr<float>(), master.numel(), (float)lr,
                    (float)b1, (float)b2, (float)eps, (float)wd,
-                   zero_grad ? 1 : 0, cur_stream());
+                   zero_grad ? 1 : 0,
+                   extra_zero.has_value() ? extra_zero->data_ptr<float>()
+                                          : nullptr,
+                   extra_zero.has_value() ? extra_zero->numel() : 0,
+                   cur_stream());
 }
 
 void sgd_step(Tensor master, Tensor grad, c10::optional<Tensor> mom,

@@ -209,69 +209,80 @@ __global__ __launch_bounds__(256) void k_conv_mfma(
 
 // ------------------------------------------------------------------ wgrad --
 // dW[ko][k3] += Σ_m patch(m,k3) · dz(m,ko);  dW is KRSC flat [Ntot][Kd] f32.
-// VALU outer-product: 64×64 tile, 4×4 f32 acc/thread, M in 32-deep LDS
-// chunks, msplit-way M parallelism with f32 atomics.  LDS reads are b64
-// (4×bf16) per operand — 4 LDS ops per 16 FMA.
-#define LDW 72
+// MFMA formulation: the reduction axis is m, so both operands are staged
+// TRANSPOSED into LDS ([k3][m] and [ko][m]) during the gather (coalesced
+// global vec8 reads, 8 strided b16 LDS writes per thread), which makes the
+// MFMA fragment reads contiguous ds_read_b128.  Output tile 64(k3)×64(ko),
+// m reduced 32-deep per MFMA step, msplit-way M parallelism with f32
+// atomics into the KRSC gradient (in direct-grad mode that IS .grad).
+#define LDW 40  // 32 m + 8 pad
 
 template <bool VECA>
 __global__ __launch_bounds__(256) void k_wgrad(
     const bf16* __restrict__ X, const bf16* __restrict__ Dz,
     float* __restrict__ dW, ConvP p, int Ntot, int mchunk) {
-  __shared__ bf16 As[32 * LDW];   // [m within chunk][k3]
-  __shared__ bf16 Ds[32 * LDW];   // [m within chunk][ko]
+  __shared__ bf16 At[64 * LDW];   // [k3][m]
+  __shared__ bf16 Dt[64 * LDW];   // [ko][m]
   const int k3_0 = blockIdx.x * 64, n0 = blockIdx.y * 64;
   const int mbeg = blockIdx.z * mchunk;
   const int mend = min(p.M, mbeg + mchunk);
   const int tid = threadIdx.x;
-  const int srow = tid >> 3, scol = (tid & 7) * 8;
-  const int trow = (tid >> 4) * 4, tcol = (tid & 15) * 4;
+  const int sm = tid >> 3, sv = (tid & 7) * 8;  // m-lane, 8-wide k3/ko chunk
+  const int lane = tid & 63, wave = tid >> 6;
+  const int wr = wave >> 1, wc = wave & 1;
+  const int fr = lane & 15, fk = lane >> 4;
 
-  typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4;
-  float acc[4][4] = {};
-  // register prefetch across chunks
-  V8 a_nx = load8_a<1, VECA>(X, p, mbeg + srow, k3_0 + scol);
+  f32x4 acc[2][2] = {};
+
+  V8 a_nx = load8_a<1, VECA>(X, p, mbeg + sm, k3_0 + sv);
   V8 d_nx;
   {
-    int m = mbeg + srow, n = n0 + scol;
+    int m = mbeg + sm, n = n0 + sv;
     if (m < p.M && n < Ntot) d_nx.u = *(const uint4*)(Dz + (long)m * Ntot + n);
     else d_nx.u = uint4{0, 0, 0, 0};
   }
   for (int m0 = mbeg; m0 < mend; m0 += 32) {
-    *(V8*)&As[srow * LDW + scol] = a_nx;
-    *(V8*)&Ds[srow * LDW + scol] = d_nx;
+    // transposed scatter into LDS: element (m=sm, k3=sv+e) -> At[sv+e][sm]
+#pragma unroll
+    for (int e = 0; e < 8; e++) At[(sv + e) * LDW + sm] = a_nx.e[e];
+#pragma unroll
+    for (int e = 0; e < 8; e++) Dt[(sv + e) * LDW + sm] = d_nx.e[e];
     __syncthreads();
     if (m0 + 32 < mend) {
-      a_nx = load8_a<1, VECA>(X, p, m0 + 32 + srow, k3_0 + scol);
-      int m = m0 + 32 + srow, n = n0 + scol;
+      a_nx = load8_a<1, VECA>(X, p, m0 + 32 + sm, k3_0 + sv);
+      int m = m0 + 32 + sm, n = n0 + sv;
       if (m < p.M && n < Ntot)
         d_nx.u = *(const uint4*)(Dz + (long)m * Ntot + n);
       else
         d_nx.u = uint4{0, 0, 0, 0};
     }
-#pragma unroll 4
-    for (int mm = 0; mm < 32; mm++) {
-      bf16x4 av = *(const bf16x4*)&As[mm * LDW + trow];
-      bf16x4 dv = *(const bf16x4*)&Ds[mm * LDW + tcol];
-      float a[4], d[4];
+    bf16x8 af[2], bf[2];
 #pragma unroll
-      for (int i = 0; i < 4; i++) a[i] = (float)av[i];
+    for (int mi = 0; mi < 2; mi++)
+      af[mi] = *(const bf16x8*)&At[(wr * 32 + mi * 16 + fr) * LDW + fk * 8];
 #pragma unroll
-      for (int j = 0; j < 4; j++) d[j] = (float)dv[j];
+    for (int ni = 0; ni < 2; ni++)
+      bf[ni] = *(const bf16x8*)&Dt[(wc * 32 + ni * 16 + fr) * LDW + fk * 8];
 #pragma unroll
-      for (int i = 0; i < 4; i++)
+    for (int mi = 0; mi < 2; mi++)
 #pragma unroll
-        for (int j = 0; j < 4; j++) acc[i][j] = fmaf(a[i], d[j], acc[i][j]);
-    }
+      for (int ni = 0; ni < 2; ni++)
+        acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
     __syncthreads();
   }
+  // D: col (=ko) = fr, row (=k3) = fk*4+q
 #pragma unroll
-  for (int i = 0; i < 4; i++)
+  for (int mi = 0; mi < 2; mi++)
 #pragma unroll
-    for (int j = 0; j < 4; j++) {
-      int k3 = k3_0 + trow + i, n = n0 + tcol + j;
-      if (k3 < p.Kd && n < Ntot)
-        atomicAdd(&dW[(long)n * p.Kd + k3], acc[i][j]);
+    for (int ni = 0; ni < 2; ni++) {
+      int ko = n0 + wc * 32 + ni * 16 + fr;
+#pragma unroll
+      for (int q = 0; q < 4; q++) {
+        int k3 = k3_0 + wr * 32 + mi * 16 + fk * 4 + q;
+        if (k3 < p.Kd && ko < Ntot)
+          atomicAdd(&dW[(long)ko * p.Kd + k3], acc[mi][ni][q]);
+      }
     }
 }
 

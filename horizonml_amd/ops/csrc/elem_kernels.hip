@@ -384,7 +384,11 @@ __global__ __launch_bounds__(256) void k_adam_step(
     float* __restrict__ master, float* __restrict__ grad,
     float* __restrict__ m, float* __restrict__ v, bf16* __restrict__ shadow,
     const float* __restrict__ step_t, long n, float lr, float b1, float b2,
-    float eps, float wd, int zero_grad) {
+    float eps, float wd, int zero_grad, float* __restrict__ extra_zero,
+    long n_extra) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n_extra;
+       i += (long)gridDim.x * blockDim.x)
+    extra_zero[i] = 0.f;
   float t = step_t[0];
   float bc1 = 1.f - __powf(b1, t), bc2 = 1.f - __powf(b2, t);
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
@@ -528,7 +532,7 @@ void launch_bn_apply_f32(const float* ws, const void* res, void* y,
 void launch_stats_reduce(const float* ws, float* stats, long M, int C,
                          int nsplit, hipStream_t st) {
   int cblocks = (C + 63) / 64;
-  int msplit = (int)min((long)64, max((long)1, (long)(256 / cblocks)));
+  int msplit = (int)min((long)128, max((long)1, (long)(512 / cblocks)));
   long mchunk = (M + msplit - 1) / msplit;
   msplit = (int)((M + mchunk - 1) / mchunk);
   dim3 grid(cblocks, msplit);
@@ -546,7 +550,7 @@ void launch_bnact_bwd_reduce(const void* dy, const void* yout, const void* x,
                              int act, hipStream_t st) {
   // sum_dz/sum_dzx must be pre-zeroed (they are accumulated atomically)
   int cblocks = (C + 63) / 64;
-  int msplit = (int)min((long)64, max((long)1, (long)(256 / cblocks)));
+  int msplit = (int)min((long)128, max((long)1, (long)(512 / cblocks)));
   long mchunk = (M + msplit - 1) / msplit;
   msplit = (int)((M + mchunk - 1) / mchunk);
   dim3 grid(cblocks, msplit);
@@ -614,11 +618,11 @@ void launch_ce_fwd_bwd(const float* logits, const long* target, float* loss,
 void launch_adam_step(float* master, float* grad, float* m, float* v,
                       void* shadow, const float* step_t, long n, float lr,
                       float b1, float b2, float eps, float wd, int zero_grad,
-                      hipStream_t st) {
+                      float* extra_zero, long n_extra, hipStream_t st) {
   k_inc_step<<<1, 1, 0, st>>>((float*)step_t);
   k_adam_step<<<gsz(n), 256, 0, st>>>(master, grad, m, v, (bf16*)shadow,
                                       step_t, n, lr, b1, b2, eps, wd,
-                                      zero_grad);
+                                      zero_grad, extra_zero, n_extra);
 }
 
 void launch_sgd_step(float* master, float* grad, float* mom, void* shadow,

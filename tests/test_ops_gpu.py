@@ -197,7 +197,7 @@ def test_fused_adam_matches_torch():
     step_t = torch.zeros(1, device="cuda")
     for _ in range(3):
         _C().adam_step(mg, gg, m, v, shadow, step_t, 1e-3, 0.9, 0.999, 1e-8,
-                       0.0, False)
+                       0.0, False, None)
     assert rel(mg, p.detach()) < 1e-5
     assert rel(shadow, p.detach()) < 1e-2
 
