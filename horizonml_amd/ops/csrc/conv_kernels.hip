@@ -241,12 +241,22 @@ __global__ __launch_bounds__(256) void k_wgrad(
     if (m < p.M && n < Ntot) d_nx.u = *(const uint4*)(Dz + (long)m * Ntot + n);
     else d_nx.u = uint4{0, 0, 0, 0};
   }
+  const int smlo = sm & 7, smhi = sm >> 3;
   for (int m0 = mbeg; m0 < mend; m0 += 32) {
-    // transposed scatter into LDS: element (m=sm, k3=sv+e) -> At[sv+e][sm]
+    // transposed scatter into LDS: element (m=sm, k3=sv+e) -> At[sv+e][sm']
+    // with the m 8-chunk XOR-swizzled by row&3 (8-way -> 2-way write banks)
 #pragma unroll
-    for (int e = 0; e < 8; e++) At[(sv + e) * LDW + sm] = a_nx.e[e];
+    for (int e = 0; e < 8; e++) {
+      int row = sv + e;
+      int smx = smlo | (((smhi ^ row) & 3) << 3);
+      At[row * LDW + smx] = a_nx.e[e];
+    }
 #pragma unroll
-    for (int e = 0; e < 8; e++) Dt[(sv + e) * LDW + sm] = d_nx.e[e];
+    for (int e = 0; e < 8; e++) {
+      int row = sv + e;
+      int smx = smlo | (((smhi ^ row) & 3) << 3);
+      Dt[row * LDW + smx] = d_nx.e[e];
+    }
     __syncthreads();
     if (m0 + 32 < mend) {
       a_nx = load8_a<1, VECA>(X, p, m0 + 32 + sm, k3_0 + sv);
@@ -258,11 +268,15 @@ __global__ __launch_bounds__(256) void k_wgrad(
     }
     bf16x8 af[2], bf[2];
 #pragma unroll
-    for (int mi = 0; mi < 2; mi++)
-      af[mi] = *(const bf16x8*)&At[(wr * 32 + mi * 16 + fr) * LDW + fk * 8];
+    for (int mi = 0; mi < 2; mi++) {
+      int row = wr * 32 + mi * 16 + fr;
+      af[mi] = *(const bf16x8*)&At[row * LDW + ((fk ^ row) & 3) * 8];
+    }
 #pragma unroll
-    for (int ni = 0; ni < 2; ni++)
-      bf[ni] = *(const bf16x8*)&Dt[(wc * 32 + ni * 16 + fr) * LDW + fk * 8];
+    for (int ni = 0; ni < 2; ni++) {
+      int row = wc * 32 + ni * 16 + fr;
+      bf[ni] = *(const bf16x8*)&Dt[row * LDW + ((fk ^ row) & 3) * 8];
+    }
 #pragma unroll
     for (int mi = 0; mi < 2; mi++)
 #pragma unroll
@@ -397,7 +411,7 @@ extern "C" void launch_gemm_bf16(const void* a, const void* b, void* c, int M,
 extern "C" void launch_wgrad(const void* x, const void* dz, float* dw, ConvP p,
                              hipStream_t st) {
   int tiles = cdiv_h(p.Kd, 64) * cdiv_h(p.K, 64);
-  int msplit = max(1, min(cdiv_h(p.M, 32), 256 / max(1, tiles)));
+  int msplit = max(1, min(cdiv_h(p.M, 32), 512 / max(1, tiles)));
   int mchunk = cdiv_h(cdiv_h(p.M, msplit), 32) * 32;
   msplit = cdiv_h(p.M, mchunk);
   dim3 grid(cdiv_h(p.Kd, 64), cdiv_h(p.K, 64), msplit);
