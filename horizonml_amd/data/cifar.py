@@ -31,12 +31,15 @@ _STD = 0.5
 
 
 class SyntheticCIFAR10(Dataset):
-    """Deterministic random CIFAR-shaped dataset (uint8 in memory)."""
+    """Deterministic random CIFAR-shaped dataset (uint8 in memory).
+
+    ``image_size=224`` gives the ImageNet-shaped variant used by the hybrid
+    DP×PP ResNet50 configuration (BASELINE.json config #5)."""
 
     def __init__(self, n: int = 50000, num_classes: int = 10,
-                 seed: int = DEFAULT_SEED):
+                 seed: int = DEFAULT_SEED, image_size: int = 32):
         g = torch.Generator().manual_seed(seed)
-        self.images = torch.randint(0, 256, (n, 3, 32, 32),
+        self.images = torch.randint(0, 256, (n, 3, image_size, image_size),
                                     dtype=torch.uint8, generator=g)
         self.labels = torch.randint(0, num_classes, (n,),
                                     dtype=torch.long, generator=g)
@@ -78,24 +81,27 @@ class CIFAR10Local(Dataset):
 
 
 def build_dataset(data_dir: str = "./data", synthetic: Optional[bool] = None,
-                  n: int = 50000, seed: int = DEFAULT_SEED) -> Dataset:
+                  n: int = 50000, seed: int = DEFAULT_SEED,
+                  image_size: int = 32, num_classes: int = 10) -> Dataset:
     """Real CIFAR-10 if present under ``data_dir`` (unless ``synthetic=True``),
     else synthetic."""
-    if synthetic is not True:
+    if synthetic is not True and image_size == 32 and num_classes == 10:
         for cand in (data_dir, os.path.join(data_dir, "cifar-10-batches-py")):
             if os.path.isfile(os.path.join(cand, "data_batch_1")):
                 return CIFAR10Local(data_dir, train=True)
         if synthetic is False:
             raise FileNotFoundError(
                 f"no CIFAR-10 batches under {data_dir!r} and synthetic=False")
-    return SyntheticCIFAR10(n=n, seed=seed)
+    return SyntheticCIFAR10(n=n, seed=seed, image_size=image_size,
+                            num_classes=num_classes)
 
 
 def get_dataloader(rank: int, world_size: int, batch_size: int = 64,
                    sample_size: int = 1000, strategy: str = "dp",
                    data_dir: str = "./data", synthetic: Optional[bool] = None,
-                   seed: int = DEFAULT_SEED,
-                   drop_last: bool = False) -> Tuple[DataLoader, Optional[DistributedSampler]]:
+                   seed: int = DEFAULT_SEED, drop_last: bool = False,
+                   image_size: int = 32, num_classes: int = 10
+                   ) -> Tuple[DataLoader, Optional[DistributedSampler]]:
     """Reference-parity dataloader.
 
     * ``dp``: shared random subset + DistributedSampler shard per rank
@@ -103,14 +109,20 @@ def get_dataloader(rank: int, world_size: int, batch_size: int = 64,
     * ``mp``/``tp``: every rank iterates the *same* full subset in the same
       order (the reference intends this but breaks it — Q1; fixed here),
       ``shuffle=False`` like ``layer_model_parallel_train.py:103-131``.
+    * ``hybrid``: DP×PP — ``rank``/``world_size`` here are the *DP replica*
+      coordinates (dp_rank, dp_size); the shard is deterministic
+      (``shuffle=False``) so every pipeline stage of one DP chain iterates
+      identical batches without extra synchronization.
     """
     ds = build_dataset(data_dir, synthetic,
-                       n=max(50000, sample_size), seed=seed)
+                       n=max(50000, sample_size), seed=seed,
+                       image_size=image_size, num_classes=num_classes)
     idx = shared_subset_indices(len(ds), sample_size, seed=seed)
     subset = Subset(ds, idx.tolist())
-    if strategy == "dp":
+    if strategy in ("dp", "hybrid"):
         sampler = DistributedSampler(subset, num_replicas=world_size,
-                                     rank=rank, shuffle=True, seed=seed,
+                                     rank=rank,
+                                     shuffle=(strategy == "dp"), seed=seed,
                                      drop_last=drop_last)
         loader = DataLoader(subset, batch_size=batch_size, sampler=sampler,
                             num_workers=0, pin_memory=torch.cuda.is_available(),

@@ -33,16 +33,28 @@ def train_pp(ctx: DistContext, epochs: int, sample_size: int, logs_dir: str,
              synthetic: Optional[bool] = None, data_dir: str = "./data",
              microbatches: int = 1, group=None, n_stages: Optional[int] = None,
              stage_idx: Optional[int] = None, dp_group=None,
+             data_rank: Optional[int] = None, data_world: Optional[int] = None,
+             num_classes: int = 10, image_size: int = 32,
              log_progress: bool = True, probe_divergence: bool = True):
     rank, world = ctx.rank, ctx.world_size
     n_stages = n_stages or world
     stage_idx = stage_idx if stage_idx is not None else rank
     seed_everything(rank=0)  # identical model init on every rank
-    loader, _ = get_dataloader(rank, world, batch_size, sample_size,
-                               strategy="mp", data_dir=data_dir,
-                               synthetic=synthetic)
+    if data_rank is not None:
+        # hybrid DP×PP: each DP chain iterates its own deterministic shard
+        loader, _ = get_dataloader(data_rank, data_world, batch_size,
+                                   sample_size, strategy="hybrid",
+                                   data_dir=data_dir, synthetic=synthetic,
+                                   image_size=image_size,
+                                   num_classes=num_classes)
+    else:
+        loader, _ = get_dataloader(rank, world, batch_size, sample_size,
+                                   strategy="mp", data_dir=data_dir,
+                                   synthetic=synthetic,
+                                   image_size=image_size,
+                                   num_classes=num_classes)
 
-    model = build_model(model_name, num_classes=10)
+    model = build_model(model_name, num_classes=num_classes)
     segments = partition_model(model, n_stages)
     seg = segments[stage_idx]
     if ctx.is_gpu:
@@ -140,7 +152,38 @@ def train_pp(ctx: DistContext, epochs: int, sample_size: int, logs_dir: str,
             print(f"[pp stage{stage_idx}] epoch {epoch + 1}/{epochs} "
                   f"loss={loss_v:.4f} time={epoch_time:.2f}s", flush=True)
         barrier(ctx)
+    train_pp.last_segment = seg  # exposed for tests (DP-sync verification)
     return writer.path
+
+
+def hybrid_worker(rank: int, world_size: int, epochs: int, sample_size: int,
+                  port: int, logs_dir: str, batch_size: int = 64,
+                  model_name: str = "resnet50",
+                  backend: Optional[str] = None,
+                  synthetic: Optional[bool] = None, lr: float = 1e-3,
+                  optimizer_name: str = "adam", microbatches: int = 1,
+                  dp_size: int = 2, pp_size: int = 4,
+                  num_classes: int = 10, image_size: int = 32):
+    """Hybrid DP×PP worker (BASELINE.json config #5: ResNet50, 2×4 on 8
+    GPUs).  Rank layout ``rank = dp_rank * pp_size + pp_stage`` keeps each
+    pipeline chain on contiguous (xGMI-adjacent) GPUs; the DP replica sync
+    is a bucketed all-reduce over the per-stage sub-communicator."""
+    from ..runtime.distributed import make_hybrid_groups
+    ctx = setup_distributed(rank, world_size, port, backend=backend)
+    try:
+        if ctx.is_gpu and model_name.startswith("resnet"):
+            from .. import ops as _ops
+            _ops.extension()
+        dp_group, pp_group, dp_rank, pp_stage = make_hybrid_groups(
+            ctx, dp_size, pp_size)
+        train_pp(ctx, epochs, sample_size, logs_dir, batch_size=batch_size,
+                 model_name=model_name, synthetic=synthetic, lr=lr,
+                 optimizer_name=optimizer_name, microbatches=microbatches,
+                 group=pp_group, n_stages=pp_size, stage_idx=pp_stage,
+                 dp_group=dp_group, data_rank=dp_rank, data_world=dp_size,
+                 num_classes=num_classes, image_size=image_size)
+    finally:
+        teardown_distributed(ctx)
 
 
 def pp_worker(rank: int, world_size: int, epochs: int, sample_size: int,
