@@ -103,6 +103,7 @@ def main():
         loss = cross_entropy(logits, y_static)
         loss.backward()
         if world > 1:
+            _ops.extension().flush_wgrad()    # wgrads complete before sync
             comm_buf.copy_(mgr.grad)          # pack f32 -> bf16 (half bytes)
             dist.all_reduce(comm_buf)         # RCCL over xGMI
             mgr.grad.copy_(comm_buf).mul_(inv_world)

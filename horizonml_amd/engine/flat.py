@@ -26,6 +26,9 @@ from ..models.layers import ConvBNAct, Linear
 
 class FlatParamManager:
     def __init__(self, model: nn.Module, device: torch.device):
+        # managed convs defer their weight-grad GEMMs: flush_wgrad() (called
+        # by the fused optimizer step) runs them as one batched kernel
+        _ops.extension().set_wgrad_defer(True)
         params = [p for p in model.parameters() if p.requires_grad]
         total = sum(p.numel() for p in params)
         self.params = params
@@ -106,6 +109,7 @@ class HorizonAdam:
         self.step_t = torch.zeros(1, device=dev)
 
     def step(self, zero_grad: bool = True):
+        _ops.extension().flush_wgrad()  # batched deferred weight grads
         _ops.extension().adam_step(self.mgr.master, self.mgr.grad, self.m,
                                    self.v, self.mgr.shadow, self.step_t,
                                    self.lr, self.betas[0], self.betas[1],
@@ -122,6 +126,7 @@ class HorizonSGD:
         self.mom = (torch.zeros_like(mgr.master) if momentum > 0 else None)
 
     def step(self, zero_grad: bool = True):
+        _ops.extension().flush_wgrad()  # batched deferred weight grads
         _ops.extension().sgd_step(self.mgr.master, self.mgr.grad, self.mom,
                                   self.mgr.shadow, self.lr, self.mu, self.wd,
                                   zero_grad)

@@ -3,7 +3,9 @@
 // linked as extra objects (no hipify, no CUDA compat).
 #include <torch/extension.h>
 #include <c10/hip/HIPStream.h>
+#include <c10/hip/HIPCachingAllocator.h>
 
+#include <memory>
 #include <vector>
 
 struct ConvP {
@@ -24,7 +26,11 @@ void launch_conv_dgrad(const void*, const void*, void*, float*, int, ConvP,
                        hipStream_t);
 void launch_gemm_bf16(const void*, const void*, void*, int, int, int,
                       hipStream_t);
-void launch_wgrad(const void*, const void*, float*, ConvP, hipStream_t);
+int wgrad_msplit(ConvP);
+void launch_wgrad(const void*, const void*, float*, float*, ConvP,
+                  hipStream_t);
+void launch_wgrad_batched(const void*, int, hipStream_t);
+void launch_wgrad_reduce_batched(const void*, int, hipStream_t);
 void launch_bn_apply(const void*, const void*, void*, const float*,
                      const float*, const float*, float*, float*, float*,
                      float*, long, int, float, float, int, int, hipStream_t);
@@ -69,6 +75,162 @@ namespace {
 using torch::Tensor;
 
 hipStream_t cur_stream() { return c10::hip::getCurrentHIPStream().stream(); }
+
+// ---- deferred (batched) wgrad ---------------------------------------------
+// The weight-gradient GEMMs are off the backward's critical chain (nothing
+// reads dW until the optimizer step), and each conv's own wgrad underfills
+// the 256 CUs at CIFAR shapes.  In direct-grad mode with defer enabled
+// (FlatParamManager), conv backward only records the task; ``flush_wgrad``
+// (called by the fused optimizer / before the DP all-reduce) then runs ALL
+// pending weight gradients as one batched kernel + one batched slab-reduce —
+// 2 launches instead of ~40, full chip occupancy, no atomics.  Task tables
+// travel as kernel arguments by value, so the path is hipGraph-capture-safe
+// (no host staging buffers whose contents could change between replays).
+constexpr int WG_MAX_TASKS = 14;
+
+struct WgradTask {
+  const void* X;
+  const void* Dz;
+  float* out;
+  ConvP p;
+  int Ntot, mchunk, msplit, tx, ty;
+  int base;
+  int vec;
+};
+
+struct WgradBatchArgs {
+  int n;
+  WgradTask t[WG_MAX_TASKS];
+};
+
+struct WredTask {
+  const float* ws;
+  float* dW;
+  long n;
+  int msplit;
+  int base;
+};
+
+struct WredBatchArgs {
+  int n;
+  WredTask t[WG_MAX_TASKS];
+};
+
+struct PendingWgrad {
+  torch::Tensor x, dconv, dw;
+  ConvP p;
+};
+
+bool g_wgrad_defer = false;
+std::vector<PendingWgrad> g_pending;
+torch::Tensor g_wgrad_ws;  // persistent slab workspace (grows on demand)
+
+void set_wgrad_defer(bool on) { g_wgrad_defer = on; }
+bool wgrad_defer_enabled() { return g_wgrad_defer; }
+int64_t wgrad_pending() { return (int64_t)g_pending.size(); }
+
+static inline int cdiv_i(long a, long b) { return (int)((a + b - 1) / b); }
+
+void flush_one_group(int lo, int hi);
+
+// Tasks writing the SAME grad tensor (gradient accumulation: several
+// backwards before one flush) must not RMW concurrently — partition the
+// pending list into maximal runs of unique dW targets and flush each run
+// as its own batched launch pair (runs are stream-ordered).
+void flush_wgrad() {
+  if (g_pending.empty()) return;
+  const int total = (int)g_pending.size();
+  int lo = 0;
+  while (lo < total) {
+    std::vector<const void*> seen;
+    int hi = lo;
+    for (; hi < total; hi++) {
+      const void* ptr = g_pending[hi].dw.data_ptr();
+      bool dup = false;
+      for (const void* q : seen)
+        if (q == ptr) { dup = true; break; }
+      if (dup) break;
+      seen.push_back(ptr);
+    }
+    flush_one_group(lo, hi);
+    lo = hi;
+  }
+  g_pending.clear();
+}
+
+void flush_one_group(int group_lo, int group_hi) {
+  auto st = cur_stream();
+  const int n = group_hi - group_lo;
+  const PendingWgrad* pend = g_pending.data() + group_lo;
+  // split choice: mchunk ≈ 512 rows keeps per-block m-loops short while the
+  // batched grid (all tasks together) fills the chip
+  std::vector<int> msplit(n), mchunk(n);
+  std::vector<long> ws_off(n, 0);
+  long ws_total = 0;
+  for (int i = 0; i < n; i++) {
+    const ConvP& p = pend[i].p;
+    int ms = std::max(1, cdiv_i(p.M, 512));
+    int mc = cdiv_i(cdiv_i(p.M, ms), 32) * 32;
+    ms = cdiv_i(p.M, mc);
+    msplit[i] = ms;
+    mchunk[i] = mc;
+    if (ms > 1) {
+      ws_off[i] = ws_total;
+      ws_total += (long)ms * p.K * p.Kd;
+    }
+  }
+  if (ws_total > 0 &&
+      (!g_wgrad_ws.defined() || g_wgrad_ws.numel() < ws_total))
+    g_wgrad_ws = at::empty({ws_total},
+                           pend[0].x.options().dtype(torch::kFloat32));
+  float* ws_base = ws_total > 0 ? g_wgrad_ws.data_ptr<float>() : nullptr;
+
+  for (int lo = 0; lo < n; lo += WG_MAX_TASKS) {
+    WgradBatchArgs a{};
+    a.n = std::min(WG_MAX_TASKS, n - lo);
+    int blocks = 0;
+    for (int j = 0; j < a.n; j++) {
+      const PendingWgrad& pw = pend[lo + j];
+      WgradTask& t = a.t[j];
+      t.X = pw.x.data_ptr();
+      t.Dz = pw.dconv.data_ptr();
+      t.out = msplit[lo + j] > 1 ? ws_base + ws_off[lo + j]
+                                 : pw.dw.data_ptr<float>();
+      t.p = pw.p;
+      t.Ntot = pw.p.K;
+      t.mchunk = mchunk[lo + j];
+      t.msplit = msplit[lo + j];
+      t.tx = cdiv_i(pw.p.Kd, 64);
+      t.ty = cdiv_i(pw.p.K, 64);
+      t.base = blocks;
+      t.vec = (pw.p.C % 8) == 0;
+      blocks += t.tx * t.ty * t.msplit;
+    }
+    launch_wgrad_batched(&a, blocks, st);
+  }
+
+  // batched slab reduce for the split tasks
+  WredBatchArgs r{};
+  r.n = 0;
+  int rblocks = 0;
+  auto flush_red = [&]() {
+    if (r.n > 0) launch_wgrad_reduce_batched(&r, rblocks, st);
+    r.n = 0;
+    rblocks = 0;
+  };
+  for (int i = 0; i < n; i++) {
+    if (msplit[i] <= 1) continue;
+    if (r.n == WG_MAX_TASKS) flush_red();
+    WredTask& t = r.t[r.n++];
+    t.ws = ws_base + ws_off[i];
+    t.dW = pend[i].dw.data_ptr<float>();
+    t.n = (long)pend[i].p.K * pend[i].p.Kd;
+    t.msplit = msplit[i];
+    t.base = rblocks;
+    rblocks += cdiv_i(t.n, 1024);
+  }
+  flush_red();
+}
 
 void check_cl(const Tensor& t, const char* name) {
   TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
@@ -226,7 +388,19 @@ std::vector<Tensor> conv_bn_act_bwd(
 
   Tensor dw = direct ? *dw_out
                      : at::zeros({(int64_t)K, R, S, (int64_t)p.C}, fopt);
-  launch_wgrad(x.data_ptr(), dconv.data_ptr(), dw.data_ptr<float>(), p, st);
+  if (direct && g_wgrad_defer) {
+    g_pending.push_back({x, dconv, dw, p});  // runs batched at flush_wgrad
+  } else {
+    int msplit = wgrad_msplit(p);
+    Tensor ws;
+    float* ws_ptr = nullptr;
+    if (msplit > 1) {
+      ws = at::empty({(long)msplit * K * p.Kd}, fopt);
+      ws_ptr = ws.data_ptr<float>();
+    }
+    launch_wgrad(x.data_ptr(), dconv.data_ptr(), dw.data_ptr<float>(), ws_ptr,
+                 p, st);
+  }
 
   Tensor dx;
   if (need_dx) {
@@ -363,6 +537,34 @@ Tensor gemm_bf16(Tensor a, Tensor b) {
   return c;
 }
 
+// Debug/test entry: standalone weight-gradient (immediate or deferred).
+Tensor wgrad_only(Tensor x, Tensor dz, int64_t K, int64_t R, int64_t S,
+                  int64_t stride, int64_t pad, bool defer,
+                  c10::optional<Tensor> dw_out) {
+  check_cl(x, "x");
+  check_cl(dz, "dz");
+  ConvP p = make_convp(x, (int)K, (int)R, (int)S, (int)stride, (int)pad);
+  Tensor dw = dw_out.has_value()
+                  ? *dw_out
+                  : at::zeros({K, R, S, (int64_t)p.C},
+                              x.options().dtype(torch::kFloat32));
+  if (defer) {
+    g_pending.push_back({x, dz, dw, p});
+  } else {
+    int msplit = wgrad_msplit(p);
+    Tensor ws;
+    float* ws_ptr = nullptr;
+    if (msplit > 1) {
+      ws = at::empty({(long)msplit * p.K * p.Kd},
+                     x.options().dtype(torch::kFloat32));
+      ws_ptr = ws.data_ptr<float>();
+    }
+    launch_wgrad(x.data_ptr(), dz.data_ptr(), dw.data_ptr<float>(), ws_ptr,
+                 p, cur_stream());
+  }
+  return dw;
+}
+
 // ------------------------------------------------------------- optimizers --
 void adam_step(Tensor master, Tensor grad, Tensor m, Tensor v,
                c10::optional<Tensor> shadow, Tensor step_t, double lr,
@@ -424,4 +626,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sgd_step", &sgd_step);
   m.def("permute_krsc_rsck", &permute_krsc_rsck);
   m.def("grad_divergence", &grad_divergence);
+  m.def("set_wgrad_defer", &set_wgrad_defer);
+  m.def("wgrad_defer_enabled", &wgrad_defer_enabled);
+  m.def("wgrad_pending", &wgrad_pending);
+  m.def("flush_wgrad", &flush_wgrad);
+  m.def("wgrad_only", &wgrad_only);
 }

@@ -213,18 +213,40 @@ __global__ __launch_bounds__(256) void k_conv_mfma(
 // TRANSPOSED into LDS ([k3][m] and [ko][m]) during the gather (coalesced
 // global vec8 reads, 8 strided b16 LDS writes per thread), which makes the
 // MFMA fragment reads contiguous ds_read_b128.  Output tile 64(k3)×64(ko),
-// m reduced 32-deep per MFMA step, msplit-way M parallelism with f32
-// atomics into the KRSC gradient (in direct-grad mode that IS .grad).
+// m reduced 32-deep per MFMA step, msplit-way M parallelism.
+//
+// Epilogue is ATOMIC-FREE (measured: the former per-element f32 atomicAdd
+// into dW cost ~12.5 µs per million atomics and dominated wgrad — see
+// profiles/bench_r01_opt1_graph_stats.txt): the accumulator tile is
+// transposed through LDS into [ko][k3] row order and written with plain
+// coalesced float4 accesses — read-modify-write into the pre-zeroed
+// gradient when msplit == 1 (single writer per element; += preserves
+// multi-backward accumulation), else pure stores into a per-split slab
+// ws[z][Ntot][Kd] that the reduce kernel sums into the gradient.
+//
+// Deferred/batched mode (the flat fast path): all convs' wgrads of one
+// backward run as ONE kernel — every task's (tile, z) blocks are in flight
+// together, so the chip fills without oversplitting M, and 2 launches
+// replace ~40 (bind.cpp::flush_wgrad).
 #define LDW 40  // 32 m + 8 pad
 
+union WgradSmem {
+  struct {
+    bf16 At[64 * LDW];  // [k3][m]
+    bf16 Dt[64 * LDW];  // [ko][m]
+  } s;
+  float out[64][65];  // transposed epilogue staging (65: conflict-free)
+};
+
 template <bool VECA>
-__global__ __launch_bounds__(256) void k_wgrad(
-    const bf16* __restrict__ X, const bf16* __restrict__ Dz,
-    float* __restrict__ dW, ConvP p, int Ntot, int mchunk) {
-  __shared__ bf16 At[64 * LDW];   // [k3][m]
-  __shared__ bf16 Dt[64 * LDW];   // [ko][m]
-  const int k3_0 = blockIdx.x * 64, n0 = blockIdx.y * 64;
-  const int mbeg = blockIdx.z * mchunk;
+DEV void wgrad_tile(const bf16* __restrict__ X, const bf16* __restrict__ Dz,
+                    float* __restrict__ out, const ConvP& p, int Ntot,
+                    int mchunk, int tx, int ty, int z, bool split,
+                    WgradSmem& smem) {
+  bf16* At = smem.s.At;   // [k3][m]
+  bf16* Dt = smem.s.Dt;   // [ko][m]
+  const int k3_0 = tx * 64, n0 = ty * 64;
+  const int mbeg = z * mchunk;
   const int mend = min(p.M, mbeg + mchunk);
   const int tid = threadIdx.x;
   const int sm = tid >> 3, sv = (tid & 7) * 8;  // m-lane, 8-wide k3/ko chunk
@@ -285,19 +307,153 @@ __global__ __launch_bounds__(256) void k_wgrad(
             af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
     __syncthreads();
   }
-  // D: col (=ko) = fr, row (=k3) = fk*4+q
+  // D: col (=ko) = fr, row (=k3) = fk*4+q.  Transpose through LDS.
+  __syncthreads();  // LDS union: staging buffers are done
 #pragma unroll
   for (int mi = 0; mi < 2; mi++)
 #pragma unroll
     for (int ni = 0; ni < 2; ni++) {
-      int ko = n0 + wc * 32 + ni * 16 + fr;
+      int ko_l = wc * 32 + ni * 16 + fr;
 #pragma unroll
-      for (int q = 0; q < 4; q++) {
-        int k3 = k3_0 + wr * 32 + mi * 16 + fk * 4 + q;
-        if (k3 < p.Kd && ko < Ntot)
-          atomicAdd(&dW[(long)ko * p.Kd + k3], acc[mi][ni][q]);
-      }
+      for (int q = 0; q < 4; q++)
+        smem.out[ko_l][wr * 32 + mi * 16 + fk * 4 + q] = acc[mi][ni][q];
     }
+  __syncthreads();
+  float* dst = out + (split ? (long)z * Ntot * (long)p.Kd : 0L);
+  const int ko_r = tid >> 2, cch = (tid & 3) * 16;
+  const int gko = n0 + ko_r;
+  if (gko < Ntot) {
+    long base = (long)gko * p.Kd + k3_0 + cch;
+    if (k3_0 + cch + 16 <= p.Kd) {
+#pragma unroll
+      for (int e = 0; e < 16; e += 4) {
+        float4 v = *(const float4*)&smem.out[ko_r][cch + e];
+        if (!split) {  // accumulate into .grad (single writer, RMW is safe)
+          float4 d = *(const float4*)&dst[base + e];
+          v.x += d.x; v.y += d.y; v.z += d.z; v.w += d.w;
+        }
+        *(float4*)&dst[base + e] = v;
+      }
+    } else {
+      for (int e = 0; e < 16 && k3_0 + cch + e < p.Kd; e++)
+        dst[base + e] = smem.out[ko_r][cch + e] + (split ? 0.f : dst[base + e]);
+    }
+  }
+}
+
+template <bool VECA>
+__global__ __launch_bounds__(256) void k_wgrad(
+    const bf16* __restrict__ X, const bf16* __restrict__ Dz,
+    float* __restrict__ dW, ConvP p, int Ntot, int mchunk) {
+  __shared__ WgradSmem smem;
+  wgrad_tile<VECA>(X, Dz, dW, p, Ntot, mchunk, blockIdx.x, blockIdx.y,
+                   blockIdx.z, gridDim.z > 1, smem);
+}
+
+// Σ over msplit wgrad slabs [z][n] -> dW[n] (+=; dW pre-zeroed or fresh).
+__global__ __launch_bounds__(256) void k_wgrad_reduce(
+    const float* __restrict__ ws, float* __restrict__ dW, long n,
+    int msplit) {
+  long i4 = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 4;
+  long stride = (long)gridDim.x * blockDim.x * 4;
+  for (; i4 + 4 <= n; i4 += stride) {
+    float4 v = *(const float4*)&ws[i4];
+    for (int z = 1; z < msplit; z++) {
+      float4 w = *(const float4*)&ws[(long)z * n + i4];
+      v.x += w.x; v.y += w.y; v.z += w.z; v.w += w.w;
+    }
+    float4 d = *(float4*)&dW[i4];
+    d.x += v.x; d.y += v.y; d.z += v.z; d.w += v.w;
+    *(float4*)&dW[i4] = d;
+  }
+  // tail (Kd*K not divisible by 4)
+  if (blockIdx.x == 0 && threadIdx.x < 4) {
+    long tail = n & ~3L;
+    for (long i = tail + threadIdx.x; i < n; i += 4) {
+      float v = ws[i];
+      for (int z = 1; z < msplit; z++) v += ws[(long)z * n + i];
+      dW[i] += v;
+    }
+  }
+}
+
+// ------------------------------------------------- batched (deferred) wgrad --
+// One launch computes every pending conv's weight gradient: all tasks'
+// (tile, z) blocks are in flight together so msplit can stay small
+// (mchunk ≈ 512 rows) and the grid still fills 256 CUs without atomics.
+// Task tables travel as kernel arguments by value — no staging buffers,
+// hipGraph-capture-safe.
+#define WG_MAX_TASKS 14
+
+struct WgradTask {
+  const bf16* X;
+  const bf16* Dz;
+  float* out;      // dW when msplit==1, else slab base [msplit][Ntot][Kd]
+  ConvP p;
+  int Ntot, mchunk, msplit, tx, ty;  // tile counts
+  int base;        // first block id of this task
+  int vec;
+};
+
+struct WgradBatchArgs {
+  int n;
+  WgradTask t[WG_MAX_TASKS];
+};
+
+__global__ __launch_bounds__(256) void k_wgrad_batched(WgradBatchArgs a) {
+  int bid = blockIdx.x;
+  int i = 0;
+  while (i + 1 < a.n && bid >= a.t[i + 1].base) i++;
+  const WgradTask& t = a.t[i];
+  int local = bid - t.base;
+  int nt = t.tx * t.ty;
+  int z = local / nt, rem = local - z * nt;
+  int ty = rem / t.tx, tx = rem - ty * t.tx;
+  __shared__ WgradSmem smem;
+  if (t.vec)
+    wgrad_tile<true>(t.X, t.Dz, t.out, t.p, t.Ntot, t.mchunk, tx, ty, z,
+                     t.msplit > 1, smem);
+  else
+    wgrad_tile<false>(t.X, t.Dz, t.out, t.p, t.Ntot, t.mchunk, tx, ty, z,
+                      t.msplit > 1, smem);
+}
+
+struct WredTask {
+  const float* ws;
+  float* dW;
+  long n;
+  int msplit;
+  int base;  // first block id; each block covers 1024 floats
+};
+
+struct WredBatchArgs {
+  int n;
+  WredTask t[WG_MAX_TASKS];
+};
+
+__global__ __launch_bounds__(256) void k_wgrad_reduce_batched(
+    WredBatchArgs a) {
+  int bid = blockIdx.x;
+  int i = 0;
+  while (i + 1 < a.n && bid >= a.t[i + 1].base) i++;
+  const WredTask& t = a.t[i];
+  long i4 = ((long)(bid - t.base) * 256 + threadIdx.x) * 4;
+  if (i4 + 4 <= t.n) {
+    float4 v = *(const float4*)&t.ws[i4];
+    for (int z = 1; z < t.msplit; z++) {
+      float4 w = *(const float4*)&t.ws[(long)z * t.n + i4];
+      v.x += w.x; v.y += w.y; v.z += w.z; v.w += w.w;
+    }
+    float4 d = *(float4*)&t.dW[i4];
+    d.x += v.x; d.y += v.y; d.z += v.z; d.w += v.w;
+    *(float4*)&t.dW[i4] = d;
+  } else {
+    for (long j = i4; j < t.n; j++) {
+      float v = t.ws[j];
+      for (int z = 1; z < t.msplit; z++) v += t.ws[(long)z * t.n + j];
+      t.dW[j] += v;
+    }
+  }
 }
 
 // ------------------------------------------------------------- launchers --
@@ -408,18 +564,45 @@ extern "C" void launch_gemm_bf16(const void* a, const void* b, void* c, int M,
                                nullptr, nullptr, p, N, 0);
 }
 
-extern "C" void launch_wgrad(const void* x, const void* dz, float* dw, ConvP p,
-                             hipStream_t st) {
+extern "C" void launch_wgrad_batched(const void* args, int blocks,
+                                     hipStream_t st) {
+  if (blocks > 0)
+    k_wgrad_batched<<<blocks, 256, 0, st>>>(
+        *(const WgradBatchArgs*)args);
+}
+
+extern "C" void launch_wgrad_reduce_batched(const void* args, int blocks,
+                                            hipStream_t st) {
+  if (blocks > 0)
+    k_wgrad_reduce_batched<<<blocks, 256, 0, st>>>(
+        *(const WredBatchArgs*)args);
+}
+
+extern "C" int wgrad_msplit(ConvP p) {
   int tiles = cdiv_h(p.Kd, 64) * cdiv_h(p.K, 64);
   int msplit = max(1, min(cdiv_h(p.M, 32), 512 / max(1, tiles)));
   int mchunk = cdiv_h(cdiv_h(p.M, msplit), 32) * 32;
-  msplit = cdiv_h(p.M, mchunk);
+  return cdiv_h(p.M, mchunk);
+}
+
+// ws: per-split slab workspace [msplit][K][Kd] f32 (only read/written when
+// msplit > 1); dw must be pre-zeroed (direct-grad .grad view or fresh).
+extern "C" void launch_wgrad(const void* x, const void* dz, float* dw,
+                             float* ws, ConvP p, hipStream_t st) {
+  int msplit = wgrad_msplit(p);
+  int mchunk = cdiv_h(cdiv_h(p.M, msplit), 32) * 32;
   dim3 grid(cdiv_h(p.Kd, 64), cdiv_h(p.K, 64), msplit);
   bool vec = (p.C % 8) == 0;
+  float* out = msplit > 1 ? ws : dw;
   if (vec)
-    k_wgrad<true><<<grid, 256, 0, st>>>((const bf16*)x, (const bf16*)dz, dw, p,
-                                        p.K, mchunk);
+    k_wgrad<true><<<grid, 256, 0, st>>>((const bf16*)x, (const bf16*)dz, out,
+                                        p, p.K, mchunk);
   else
-    k_wgrad<false><<<grid, 256, 0, st>>>((const bf16*)x, (const bf16*)dz, dw,
+    k_wgrad<false><<<grid, 256, 0, st>>>((const bf16*)x, (const bf16*)dz, out,
                                          p, p.K, mchunk);
+  if (msplit > 1) {
+    long n = (long)p.K * p.Kd;
+    int blocks = (int)min((n / 4 + 255) / 256, (long)1024);
+    k_wgrad_reduce<<<blocks, 256, 0, st>>>(ws, dw, n, msplit);
+  }
 }

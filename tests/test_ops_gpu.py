@@ -331,5 +331,73 @@ def test_flat_manager_grad_views_accumulate():
         memory_format=torch.channels_last).to(torch.bfloat16)
     y = torch.randint(0, 10, (8,), device=dev)
     cross_entropy(model(x), y).backward()
+    _C().flush_wgrad()
     assert [p.grad.data_ptr() for p in model.parameters()] == ptrs
     assert mgr.grad.abs().sum() > 0
+
+
+def test_wgrad_defer_matches_immediate():
+    """Batched/deferred wgrad vs per-conv immediate wgrad on 16 shapes
+    (crosses the WG_MAX_TASKS chunk boundary), identical inputs — the only
+    allowed difference is f32 summation order (msplit differs)."""
+    torch.manual_seed(0)
+    shapes = [
+        # (cin, cout, k, stride, hw, bs)
+        (3, 64, 7, 2, 32, 16),     # stem: scalar gather, Kd=147
+        (64, 64, 3, 1, 8, 64),
+        (64, 128, 3, 2, 8, 64),
+        (128, 128, 3, 1, 4, 64),
+        (64, 128, 1, 2, 8, 64),
+        (128, 256, 3, 2, 4, 64),
+        (256, 256, 3, 1, 2, 64),
+        (128, 256, 1, 2, 4, 64),
+        (256, 512, 3, 2, 2, 64),
+        (512, 512, 3, 1, 1, 64),
+        (256, 512, 1, 2, 2, 64),
+        (64, 64, 3, 1, 8, 32),
+        (64, 64, 3, 1, 8, 16),
+        (128, 128, 3, 1, 4, 32),
+        (256, 256, 3, 1, 2, 32),
+        (512, 512, 3, 1, 1, 32),
+    ]
+    cases = []
+    for cin, cout, k, s, hw, bs in shapes:
+        x = torch.randn(bs, cin, hw, hw)
+        ho = (hw + 2 * (k // 2) - k) // s + 1
+        dz = torch.randn(bs, cout, ho, ho)
+        cases.append((to_gpu_cl(x), to_gpu_cl(dz), cout, k, s))
+    imm = [_C().wgrad_only(x, dz, co, k, k, s, k // 2, False, None)
+           for x, dz, co, k, s in cases]
+    dfr = [_C().wgrad_only(x, dz, co, k, k, s, k // 2, True, None)
+           for x, dz, co, k, s in cases]
+    assert _C().wgrad_pending() == len(cases)
+    _C().flush_wgrad()
+    torch.cuda.synchronize()
+    for i, (a, b) in enumerate(zip(imm, dfr)):
+        r = ((a - b).norm() / a.norm().clamp_min(1e-12)).item()
+        assert r < 1e-4, f"shape {shapes[i]}: rel={r}"
+
+
+def test_wgrad_accumulates_across_backwards():
+    """Repeated wgrads into the SAME grad tensor must sum (PP microbatch /
+    grad-accumulation semantics) — guards the atomic-free epilogue's
+    read-modify-write and the batched reduce's +=, in both paths."""
+    torch.manual_seed(3)
+    x1 = to_gpu_cl(torch.randn(32, 64, 8, 8))
+    x2 = to_gpu_cl(torch.randn(32, 64, 8, 8))
+    dz = to_gpu_cl(torch.randn(32, 64, 8, 8))
+    g1 = _C().wgrad_only(x1, dz, 64, 3, 3, 1, 1, False, None)
+    g2 = _C().wgrad_only(x2, dz, 64, 3, 3, 1, 1, False, None)
+    ref = g1 + g2
+    for defer in (False, True):
+        acc = torch.zeros_like(g1)
+        _C().set_wgrad_defer(defer)
+        try:
+            _C().wgrad_only(x1, dz, 64, 3, 3, 1, 1, defer, acc)
+            _C().wgrad_only(x2, dz, 64, 3, 3, 1, 1, defer, acc)
+            _C().flush_wgrad()
+        finally:
+            _C().set_wgrad_defer(False)
+        torch.cuda.synchronize()
+        r = ((acc - ref).norm() / ref.norm()).item()
+        assert r < 1e-4, f"defer={defer}: accumulation broken, rel {r}"
