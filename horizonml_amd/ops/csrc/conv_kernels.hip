@@ -459,10 +459,25 @@ __global__ __launch_bounds__(256) void k_wgrad_reduce_batched(
 // ------------------------------------------------------------- launchers --
 static inline int cdiv_h(int a, int b) { return (a + b - 1) / b; }
 
-// Split-K heuristic: fill ≥256 workgroups when the tile grid alone cannot.
+// Split-K heuristic.  Split-K costs a whole consumer chain (f32 slabs +
+// stats_reduce/cast + slab re-reads), so it only pays when a non-split
+// launch would serialize a long K loop on an underfilled grid: measured on
+// MI355X, non-split wins for Kd ≤ ~1152 even at 32–64 workgroups (the
+// K-loop is ~18–36 latency-hidden iterations), while layer3/4 shapes
+// (Kd ≥ 2304, ≤16 tiles) still want the split.
 static inline int pick_splitk(int tiles, int Kd) {
-  int target = 256;
-  int sk = tiles >= target ? 1 : cdiv_h(target, tiles);
+  // tunables (read once): grid-fill target and minimum Kd that justifies
+  // the split-K consumer chain
+  static int tile_min = [] {
+    const char* e = getenv("HZ_SK_TILES");
+    return e ? atoi(e) : 96;
+  }();
+  static int kd_min = [] {
+    const char* e = getenv("HZ_SK_KD");
+    return e ? atoi(e) : 1024;
+  }();
+  if (tiles >= tile_min || Kd < kd_min) return 1;
+  int sk = cdiv_h(256, tiles);
   int maxsk = cdiv_h(Kd, 32);
   if (sk > maxsk) sk = maxsk;
   if (sk > 32) sk = 32;
