@@ -332,41 +332,52 @@ __global__ __launch_bounds__(256) void k_linear_bwd_dx(
 // dw[Out,In] f32 (+)= Σ_b dy[b,j]·x[b,t];  db[Out] f32 (+)= Σ_b dy[b,j]
 // accum=1 adds into the existing buffer (direct-grad mode: dw/db are the
 // pre-zeroed flat .grad views).
-// dW[j][t] = Σ_b dy[b][j]·x[b][t].  Classifier shape (Out ≤ 32): one
-// thread per input column t holding all Out partials in registers; dy is
-// LDS-staged once per block (x column reads coalesce across threads).
+// dW[j][t] = Σ_b dy[b][j]·x[b][t].  Classifier shape (Out ≤ 16): one
+// thread per input column t holding OUTN=16 partials in REGISTERS (all
+// loops compile-time — dynamic bounds would spill the accumulator array to
+// scratch); dy is LDS-staged zero-padded to 16 columns, x column reads
+// coalesce across threads.
+__global__ __launch_bounds__(256) void k_linear_bwd_dw_small(
+    const float* __restrict__ dy, const bf16* __restrict__ x,
+    float* __restrict__ dw, float* __restrict__ db, int B, int In, int Out,
+    int accum) {
+  constexpr int OUTN = 16;
+  __shared__ float dys[128 * OUTN];  // [B][16], B ≤ 128
+  const int tid = threadIdx.x;
+  for (int i = tid; i < B * OUTN; i += blockDim.x) {
+    int j = i & (OUTN - 1);
+    dys[i] = j < Out ? dy[(i >> 4) * Out + j] : 0.f;
+  }
+  __syncthreads();
+  float s[OUTN];
+#pragma unroll
+  for (int j = 0; j < OUTN; j++) s[j] = 0.f;
+  int t = blockIdx.x * blockDim.x + tid;
+  if (t < In) {
+    for (int bi = 0; bi < B; bi++) {
+      float xv = b2f(x[(long)bi * In + t]);
+#pragma unroll
+      for (int j = 0; j < OUTN; j++)
+        s[j] = fmaf(dys[bi * OUTN + j], xv, s[j]);
+    }
+    for (int j = 0; j < Out; j++) {
+      long i = (long)j * In + t;
+      dw[i] = accum ? dw[i] + s[j] : s[j];
+    }
+  }
+  if (blockIdx.x == 0 && tid < Out && db != nullptr) {
+    float sb = 0.f;
+    for (int bi = 0; bi < B; bi++) sb += dys[bi * OUTN + tid];
+    db[tid] = accum ? db[tid] + sb : sb;
+  }
+}
+
 __global__ __launch_bounds__(256) void k_linear_bwd_dw(
     const float* __restrict__ dy, const bf16* __restrict__ x,
     float* __restrict__ dw, float* __restrict__ db, int B, int In, int Out,
     int accum) {
-  __shared__ float dys[2048];  // [B][Out], B·Out ≤ 2048 fast path
-  const int tid = threadIdx.x;
-  if (Out <= 32 && B * Out <= 2048) {
-    for (int i = tid; i < B * Out; i += blockDim.x) dys[i] = dy[i];
-    __syncthreads();
-    float s[32];
-    int t = blockIdx.x * blockDim.x + tid;
-    if (t < In) {
-      for (int j = 0; j < Out; j++) s[j] = 0.f;
-      for (int bi = 0; bi < B; bi++) {
-        float xv = b2f(x[(long)bi * In + t]);
-        for (int j = 0; j < Out; j++)
-          s[j] = fmaf(dys[bi * Out + j], xv, s[j]);
-      }
-      for (int j = 0; j < Out; j++) {
-        long i = (long)j * In + t;
-        dw[i] = accum ? dw[i] + s[j] : s[j];
-      }
-    }
-    if (blockIdx.x == 0 && tid < Out && db != nullptr) {
-      float sb = 0.f;
-      for (int bi = 0; bi < B; bi++) sb += dys[bi * Out + tid];
-      db[tid] = accum ? db[tid] + sb : sb;
-    }
-    return;
-  }
   long total = (long)Out * In;
-  for (long i = (long)blockIdx.x * blockDim.x + tid; i < total;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += (long)gridDim.x * blockDim.x) {
     int t = (int)(i % In), j = (int)(i / In);
     float s = 0.f;
@@ -635,9 +646,12 @@ void launch_linear_bwd(const float* dy, const void* x, const void* w,
   if (dx)
     k_linear_bwd_dx<<<gsz((long)B * In), 256, 0, st>>>(dy, (const bf16*)w,
                                                        (bf16*)dx, B, In, Out);
-  long dw_work = (Out <= 32 && B * Out <= 2048) ? In : (long)Out * In;
-  k_linear_bwd_dw<<<gsz(dw_work), 256, 0, st>>>(dy, (const bf16*)x, dw,
-                                                db, B, In, Out, accum);
+  if (Out <= 16 && B <= 128)
+    k_linear_bwd_dw_small<<<gsz((long)In), 256, 0, st>>>(
+        dy, (const bf16*)x, dw, db, B, In, Out, accum);
+  else
+    k_linear_bwd_dw<<<gsz((long)Out * In), 256, 0, st>>>(
+        dy, (const bf16*)x, dw, db, B, In, Out, accum);
 }
 
 void launch_ce_fwd_bwd(const float* logits, const long* target, float* loss,
