@@ -134,14 +134,25 @@ def main():
             graph = torch.cuda.CUDAGraph()
             with torch.cuda.graph(graph):
                 loss_static = train_step()
-            graph.replay()
-            torch.cuda.synchronize()
         except Exception as e:  # noqa: BLE001
             if rank == 0:
                 print(f"[bench] graph capture failed ({e!r}); eager fallback",
                       file=sys.stderr)
             graph = None
             mode = "eager"
+        if world > 1:
+            # all ranks must agree on the execution mode BEFORE any replay:
+            # a rank-divergent capture failure would otherwise deadlock the
+            # captured collectives (replay on one side only)
+            ok = torch.tensor([1.0 if graph is not None else 0.0],
+                              device=dev)
+            dist.all_reduce(ok, op=dist.ReduceOp.MIN)
+            if float(ok) < 1.0:
+                graph = None
+                mode = "eager"
+        if graph is not None:
+            graph.replay()
+            torch.cuda.synchronize()
 
     def run_step(i):
         x_static.copy_(pool_x[i % pool_n])

@@ -71,7 +71,7 @@ class ConvBNActFn(torch.autograd.Function):
         dx, dw, dgamma, dbeta, dres = _C().conv_bn_act_bwd(
             dy, y, x, w_bf16, w_rsck if need_dx else w_bf16, convout, gamma,
             smean, sinvstd, mod.stride, mod.padding, mod.act, need_dx,
-            ctx.has_res, dw_out, dg_out, db_out)
+            ctx.has_res, dw_out, dg_out, db_out, None)
         if direct:
             return (dx if need_dx else None, None, None, None,
                     dres if ctx.has_res else None, None)
@@ -174,3 +174,120 @@ def cross_entropy(logits: torch.Tensor, target: torch.Tensor) -> torch.Tensor:
     if logits.is_cuda:
         return CrossEntropyFn.apply(logits, target)
     return torch.nn.functional.cross_entropy(logits.float(), target)
+
+
+def _rsck(mod, w_bf16):
+    w = getattr(mod, "_w_rsck", None)
+    if w is None or w.numel() != w_bf16.numel():
+        w = w_bf16.permute(1, 2, 3, 0).contiguous()
+    return w
+
+
+class ResBlockFn(torch.autograd.Function):
+    """Whole residual block (BasicBlock / Bottleneck) as one autograd node.
+
+    Besides saving Python/autograd overhead, this fuses the residual
+    gradient junction: the reference graph produces dx(main path) + d(res
+    path) as two tensors summed by an autograd add kernel per block — here
+    the last backward dgrad ACCUMULATES into the other path's buffer
+    (``dx_accum`` in ``conv_bn_act_bwd``), eliminating one elementwise
+    kernel + one tensor round-trip per block (8 per ResNet18 step).
+    """
+
+    @staticmethod
+    def forward(ctx, x, mods, ds_mod, *params):
+        x = _cl(x)
+        training = mods[0].training
+        if ds_mod is not None:
+            y_ds, convout_ds, smean_ds, sinvstd_ds = _C().conv_bn_act_fwd(
+                x, ds_mod._shadow(), ds_mod.bn_weight, ds_mod.bn_bias,
+                ds_mod.running_mean, ds_mod.running_var, ds_mod.stride,
+                ds_mod.padding, ds_mod.momentum, ds_mod.eps, training,
+                False, None,
+                getattr(ds_mod, "_stats_buf", None) if training else None)
+            identity = y_ds
+        else:
+            identity = x
+        h = x
+        saved = [x]
+        n = len(mods)
+        for i, m in enumerate(mods):
+            res = identity if i == n - 1 else None
+            y, convout, smean, sinvstd = _C().conv_bn_act_fwd(
+                h, m._shadow(), m.bn_weight, m.bn_bias, m.running_mean,
+                m.running_var, m.stride, m.padding, m.momentum, m.eps,
+                training, m.act, res,
+                getattr(m, "_stats_buf", None) if training else None)
+            saved += [y, convout, smean, sinvstd, m._shadow(), m.bn_weight]
+            h = y
+        if ds_mod is not None:
+            saved += [y_ds, convout_ds, smean_ds, sinvstd_ds,
+                      ds_mod._shadow(), ds_mod.bn_weight]
+        ctx.save_for_backward(*saved)
+        ctx.mods = mods
+        ctx.ds_mod = ds_mod
+        return h
+
+    @staticmethod
+    def backward(ctx, dy):
+        saved = ctx.saved_tensors
+        mods, ds_mod = ctx.mods, ctx.ds_mod
+        n = len(mods)
+        x = saved[0]
+        per = [saved[1 + 6 * i: 1 + 6 * (i + 1)] for i in range(n)]
+        ds = (saved[1 + 6 * n: 1 + 6 * (n + 1)]
+              if ds_mod is not None else None)
+
+        dy = _cl(dy.to(torch.bfloat16) if dy.dtype != torch.bfloat16 else dy)
+        cur = dy
+        dres = None
+        grads = [None] * (3 * n + (3 if ds_mod is not None else 0))
+
+        for i in range(n - 1, -1, -1):
+            m = mods[i]
+            y, convout, smean, sinvstd, w_bf16, gamma = per[i]
+            inp = per[i - 1][0] if i > 0 else x
+            direct = (getattr(m, "_managed", False)
+                      and m.weight.grad is not None)
+            has_res = i == n - 1
+            # fuse the residual junction: for identity blocks the first
+            # conv's dgrad accumulates into dres; for downsample blocks the
+            # downsample dgrad accumulates into the main-path dx below
+            dx_accum = dres if (i == 0 and ds_mod is None) else None
+            dx, dw, dgamma, dbeta, dres_i = _C().conv_bn_act_bwd(
+                cur, y, inp, w_bf16, _rsck(m, w_bf16), convout, gamma,
+                smean, sinvstd, m.stride, m.padding, m.act, True, has_res,
+                m.weight.grad if direct else None,
+                m.bn_weight.grad if direct else None,
+                m.bn_bias.grad if direct else None, dx_accum)
+            if has_res:
+                dres = dres_i
+            if not direct:
+                grads[3 * i: 3 * i + 3] = [dw, dgamma, dbeta]
+            cur = dx
+
+        if ds_mod is not None:
+            y_ds, convout_ds, smean_ds, sinvstd_ds, w_ds, gamma_ds = ds
+            direct = (getattr(ds_mod, "_managed", False)
+                      and ds_mod.weight.grad is not None)
+            dxds, dw, dgamma, dbeta, _ = _C().conv_bn_act_bwd(
+                dres, y_ds, x, w_ds, _rsck(ds_mod, w_ds), convout_ds,
+                gamma_ds, smean_ds, sinvstd_ds, ds_mod.stride,
+                ds_mod.padding, False, True, False,
+                ds_mod.weight.grad if direct else None,
+                ds_mod.bn_weight.grad if direct else None,
+                ds_mod.bn_bias.grad if direct else None, cur)
+            if not direct:
+                grads[3 * n: 3 * n + 3] = [dw, dgamma, dbeta]
+            cur = dxds  # == cur buffer, accumulated
+
+        return (cur, None, None, *grads)
+
+
+def res_block(x, mods, ds_mod):
+    params = []
+    for m in mods:
+        params += [m.weight, m.bn_weight, m.bn_bias]
+    if ds_mod is not None:
+        params += [ds_mod.weight, ds_mod.bn_weight, ds_mod.bn_bias]
+    return ResBlockFn.apply(x, mods, ds_mod, *params)

@@ -33,7 +33,15 @@ class BasicBlock(nn.Module):
         self.downsample = (ConvBNAct(in_ch, ch, 1, stride=stride, act=False)
                            if (stride != 1 or in_ch != ch) else None)
 
+    def _fusable(self) -> bool:
+        return (type(self.conv1) is ConvBNAct and type(self.conv2) is ConvBNAct
+                and (self.downsample is None
+                     or type(self.downsample) is ConvBNAct))
+
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if x.is_cuda and self._fusable():
+            from ._functional_gpu import res_block
+            return res_block(x, (self.conv1, self.conv2), self.downsample)
         identity = self.downsample(x) if self.downsample is not None else x
         out = self.conv1(x)
         return self.conv2(out, residual=identity)
@@ -51,7 +59,17 @@ class Bottleneck(nn.Module):
         self.downsample = (ConvBNAct(in_ch, out_ch, 1, stride=stride, act=False)
                            if (stride != 1 or in_ch != out_ch) else None)
 
+    def _fusable(self) -> bool:
+        return (type(self.conv1) is ConvBNAct and type(self.conv2) is ConvBNAct
+                and type(self.conv3) is ConvBNAct
+                and (self.downsample is None
+                     or type(self.downsample) is ConvBNAct))
+
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if x.is_cuda and self._fusable():
+            from ._functional_gpu import res_block
+            return res_block(x, (self.conv1, self.conv2, self.conv3),
+                             self.downsample)
         identity = self.downsample(x) if self.downsample is not None else x
         out = self.conv1(x)
         out = self.conv2(out)

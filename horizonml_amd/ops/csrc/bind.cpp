@@ -23,7 +23,7 @@ void launch_conv_fwd(const void*, const void*, void*, float*, ConvP,
 void launch_conv_fwd_splitk(const void*, const void*, float*, ConvP, int,
                             hipStream_t);
 void launch_conv_dgrad(const void*, const void*, void*, float*, int, ConvP,
-                       hipStream_t);
+                       int, hipStream_t);
 void launch_gemm_bf16(const void*, const void*, void*, int, int, int,
                       hipStream_t);
 int wgrad_msplit(ConvP);
@@ -39,7 +39,7 @@ void launch_bn_apply_f32(const float*, const void*, void*, void*,
                          float*, float*, float*, long, int, float, float,
                          int, int, int, hipStream_t);
 void launch_stats_reduce(const float*, float*, long, int, int, hipStream_t);
-void launch_cast_f32_bf16(const float*, void*, long, int, hipStream_t);
+void launch_cast_f32_bf16(const float*, void*, long, int, int, hipStream_t);
 void launch_bnact_bwd_reduce(const void*, const void*, const void*,
                              const float*, const float*, float*, float*, long,
                              int, int, hipStream_t);
@@ -350,7 +350,7 @@ std::vector<Tensor> conv_bn_act_bwd(
     Tensor gamma, Tensor save_mean, Tensor save_invstd, int64_t stride,
     int64_t pad, bool act, bool need_dx, bool has_res,
     c10::optional<Tensor> dw_out, c10::optional<Tensor> dgamma_out,
-    c10::optional<Tensor> dbeta_out) {
+    c10::optional<Tensor> dbeta_out, c10::optional<Tensor> dx_accum) {
   // Direct-grad mode: when dw_out/dgamma_out/dbeta_out are given they are
   // PRE-ZEROED flat .grad views — the kernels accumulate straight into
   // them, skipping autograd's per-parameter accumulate pass.
@@ -407,7 +407,11 @@ std::vector<Tensor> conv_bn_act_bwd(
     TORCH_CHECK(w_rsck.is_contiguous() &&
                     w_rsck.scalar_type() == torch::kBFloat16,
                 "w_rsck must be contiguous bf16");
-    dx = empty_cl_bf16(p.Nb, p.C, p.H, p.W, x);
+    // Fused residual-junction add: with dx_accum given, dgrad ACCUMULATES
+    // into it (dx_total = dx + residual-path grad without a separate
+    // elementwise add kernel).
+    int accum = dx_accum.has_value() ? 1 : 0;
+    dx = accum ? *dx_accum : empty_cl_bf16(p.Nb, p.C, p.H, p.W, x);
     ConvP pd = p;
     pd.M = p.Nb * p.H * p.W;
     pd.Kd = R * S * K;
@@ -417,12 +421,12 @@ std::vector<Tensor> conv_bn_act_bwd(
       splitk = (pd.Kd + kchunk - 1) / kchunk;
       Tensor wsd = at::empty({(long)splitk * pd.M * p.C}, fopt);
       launch_conv_dgrad(dconv.data_ptr(), w_rsck.data_ptr(), nullptr,
-                        wsd.data_ptr<float>(), splitk, pd, st);
+                        wsd.data_ptr<float>(), splitk, pd, 0, st);
       launch_cast_f32_bf16(wsd.data_ptr<float>(), dx.data_ptr(),
-                           (long)pd.M * p.C, splitk, st);
+                           (long)pd.M * p.C, splitk, accum, st);
     } else {
       launch_conv_dgrad(dconv.data_ptr(), w_rsck.data_ptr(), dx.data_ptr(),
-                        nullptr, 1, pd, st);
+                        nullptr, 1, pd, accum, st);
     }
   }
   // dgamma = Σ dz·xhat, dbeta = Σ dz

@@ -122,7 +122,7 @@ template <int MODE, bool VECA, bool VECB, bool STATS, bool SPLIT>
 __global__ __launch_bounds__(256) void k_conv_mfma(
     const bf16* __restrict__ A, const bf16* __restrict__ Bw,
     bf16* __restrict__ Y, float* __restrict__ ws_out,
-    float* __restrict__ stats, ConvP p, int Ntot, int kchunk) {
+    float* __restrict__ stats, ConvP p, int Ntot, int kchunk, int accum) {
   __shared__ bf16 As[64 * LDA];
   __shared__ bf16 Bs[64 * LDA];
 
@@ -181,7 +181,9 @@ __global__ __launch_bounds__(256) void k_conv_mfma(
           if (SPLIT) {
             ws_out[(long)blockIdx.z * p.M * Ntot + (long)gm * Ntot + gn] = v;
           } else {
-            Y[(long)gm * Ntot + gn] = f2b(v);
+            long yi = (long)gm * Ntot + gn;
+            if (accum) v += b2f(Y[yi]);  // fused residual-junction add
+            Y[yi] = f2b(v);
             if (STATS) {
               ssum[ni] += v;
               ssq[ni] += v * v;
@@ -506,7 +508,7 @@ extern "C" void launch_conv_fwd(const void* x, const void* w, void* y,
   auto Y = (bf16*)y;
 #define CASE(VA, VB, ST)                                     \
   k_conv_mfma<1, VA, VB, ST, false><<<grid, 256, 0, st>>>(   \
-      A, B, Y, nullptr, stats, p, p.K, 0)
+      A, B, Y, nullptr, stats, p, p.K, 0, 0)
   if (vec && s) CASE(true, true, true);
   else if (vec) CASE(true, true, false);
   else if (s) CASE(false, false, true);
@@ -526,15 +528,17 @@ extern "C" void launch_conv_fwd_splitk(const void* x, const void* w,
   auto B = (const bf16*)w;
   if (vec)
     k_conv_mfma<1, true, true, false, true>
-        <<<grid, 256, 0, st>>>(A, B, nullptr, ws, nullptr, p, p.K, kchunk);
+        <<<grid, 256, 0, st>>>(A, B, nullptr, ws, nullptr, p, p.K, kchunk,
+                               0);
   else
     k_conv_mfma<1, false, false, false, true>
-        <<<grid, 256, 0, st>>>(A, B, nullptr, ws, nullptr, p, p.K, kchunk);
+        <<<grid, 256, 0, st>>>(A, B, nullptr, ws, nullptr, p, p.K, kchunk,
+                               0);
 }
 
 extern "C" void launch_conv_dgrad(const void* dz, const void* w_rsck,
                                   void* dx, float* ws, int splitk, ConvP p,
-                                  hipStream_t st) {
+                                  int accum, hipStream_t st) {
   // p here: M = Nb*H*W, Kd = R*S*K, output channels = C
   bool vec = (p.K % 8) == 0;
   auto A = (const bf16*)dz;
@@ -545,20 +549,22 @@ extern "C" void launch_conv_dgrad(const void* dz, const void* w_rsck,
     dim3 grid(cdiv_h(p.C, 64), cdiv_h(p.M, 64), splitk);
     if (vec)
       k_conv_mfma<2, true, true, false, true>
-          <<<grid, 256, 0, st>>>(A, B, nullptr, ws, nullptr, p, p.C, kchunk);
+          <<<grid, 256, 0, st>>>(A, B, nullptr, ws, nullptr, p, p.C, kchunk,
+                                 0);
     else
       k_conv_mfma<2, false, false, false, true>
-          <<<grid, 256, 0, st>>>(A, B, nullptr, ws, nullptr, p, p.C, kchunk);
+          <<<grid, 256, 0, st>>>(A, B, nullptr, ws, nullptr, p, p.C, kchunk,
+                                 0);
   } else {
     dim3 grid(cdiv_h(p.C, 64), cdiv_h(p.M, 64));
     if (vec)
       k_conv_mfma<2, true, true, false, false>
           <<<grid, 256, 0, st>>>(A, B, (bf16*)dx, nullptr, nullptr, p, p.C,
-                                 0);
+                                 0, accum);
     else
       k_conv_mfma<2, false, false, false, false>
           <<<grid, 256, 0, st>>>(A, B, (bf16*)dx, nullptr, nullptr, p, p.C,
-                                 0);
+                                 0, accum);
   }
 }
 
@@ -572,11 +578,11 @@ extern "C" void launch_gemm_bf16(const void* a, const void* b, void* c, int M,
   if (vec)
     k_conv_mfma<0, true, true, false, false>
         <<<grid, 256, 0, st>>>((const bf16*)a, (const bf16*)b, (bf16*)c,
-                               nullptr, nullptr, p, N, 0);
+                               nullptr, nullptr, p, N, 0, 0);
   else
     k_conv_mfma<0, false, false, false, false>
         <<<grid, 256, 0, st>>>((const bf16*)a, (const bf16*)b, (bf16*)c,
-                               nullptr, nullptr, p, N, 0);
+                               nullptr, nullptr, p, N, 0, 0);
 }
 
 extern "C" void launch_wgrad_batched(const void* args, int blocks,

@@ -102,11 +102,12 @@ __global__ __launch_bounds__(256) void k_stats_reduce(
 // Σ over nsplit f32 slabs -> bf16 elementwise (split-K dgrad output).
 __global__ __launch_bounds__(256) void k_cast_f32_bf16(
     const float* __restrict__ src, bf16* __restrict__ dst, long n,
-    int nsplit) {
+    int nsplit, int accum) {
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (long)gridDim.x * blockDim.x) {
     float v = src[i];
     for (int z = 1; z < nsplit; z++) v += src[z * n + i];
+    if (accum) v += b2f(dst[i]);  // fused residual-junction add
     dst[i] = f2b(v);
   }
 }
@@ -580,8 +581,8 @@ void launch_stats_reduce(const float* ws, float* stats, long M, int C,
 }
 
 void launch_cast_f32_bf16(const float* src, void* dst, long n, int nsplit,
-                          hipStream_t st) {
-  k_cast_f32_bf16<<<gsz(n), 256, 0, st>>>(src, (bf16*)dst, n, nsplit);
+                          int accum, hipStream_t st) {
+  k_cast_f32_bf16<<<gsz(n), 256, 0, st>>>(src, (bf16*)dst, n, nsplit, accum);
 }
 
 void launch_bnact_bwd_reduce(const void* dy, const void* yout, const void* x,
