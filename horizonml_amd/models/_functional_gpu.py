@@ -42,15 +42,16 @@ class ConvBNActFn(torch.autograd.Function):
             x, w_bf16, bn_weight, bn_bias, mod.running_mean, mod.running_var,
             mod.stride, mod.padding, mod.momentum, mod.eps, mod.training,
             mod.act, residual, stats_buf)
-        ctx.save_for_backward(x, y, convout, w_bf16, bn_weight, smean,
-                              sinvstd)
+        ctx.save_for_backward(x, y, convout, w_bf16, bn_weight, bn_bias,
+                              smean, sinvstd)
         ctx.mod = mod
         ctx.has_res = residual is not None
         return y
 
     @staticmethod
     def backward(ctx, dy):
-        x, y, convout, w_bf16, gamma, smean, sinvstd = ctx.saved_tensors
+        x, y, convout, w_bf16, gamma, beta, smean, sinvstd = \
+            ctx.saved_tensors
         mod = ctx.mod
         need_dx = ctx.needs_input_grad[0]
         # RSCK weight image for the dgrad implicit GEMM (B-fragment wants
@@ -70,7 +71,7 @@ class ConvBNActFn(torch.autograd.Function):
         db_out = mod.bn_bias.grad if direct else None
         dx, dw, dgamma, dbeta, dres = _C().conv_bn_act_bwd(
             dy, y, x, w_bf16, w_rsck if need_dx else w_bf16, convout, gamma,
-            smean, sinvstd, mod.stride, mod.padding, mod.act, need_dx,
+            beta, smean, sinvstd, mod.stride, mod.padding, mod.act, need_dx,
             ctx.has_res, dw_out, dg_out, db_out, None)
         if direct:
             return (dx if need_dx else None, None, None, None,
@@ -218,11 +219,12 @@ class ResBlockFn(torch.autograd.Function):
                 m.running_var, m.stride, m.padding, m.momentum, m.eps,
                 training, m.act, res,
                 getattr(m, "_stats_buf", None) if training else None)
-            saved += [y, convout, smean, sinvstd, m._shadow(), m.bn_weight]
+            saved += [y, convout, smean, sinvstd, m._shadow(),
+                      m.bn_weight, m.bn_bias]
             h = y
         if ds_mod is not None:
             saved += [y_ds, convout_ds, smean_ds, sinvstd_ds,
-                      ds_mod._shadow(), ds_mod.bn_weight]
+                      ds_mod._shadow(), ds_mod.bn_weight, ds_mod.bn_bias]
         ctx.save_for_backward(*saved)
         ctx.mods = mods
         ctx.ds_mod = ds_mod
@@ -234,8 +236,8 @@ class ResBlockFn(torch.autograd.Function):
         mods, ds_mod = ctx.mods, ctx.ds_mod
         n = len(mods)
         x = saved[0]
-        per = [saved[1 + 6 * i: 1 + 6 * (i + 1)] for i in range(n)]
-        ds = (saved[1 + 6 * n: 1 + 6 * (n + 1)]
+        per = [saved[1 + 7 * i: 1 + 7 * (i + 1)] for i in range(n)]
+        ds = (saved[1 + 7 * n: 1 + 7 * (n + 1)]
               if ds_mod is not None else None)
 
         dy = _cl(dy.to(torch.bfloat16) if dy.dtype != torch.bfloat16 else dy)
@@ -245,7 +247,7 @@ class ResBlockFn(torch.autograd.Function):
 
         for i in range(n - 1, -1, -1):
             m = mods[i]
-            y, convout, smean, sinvstd, w_bf16, gamma = per[i]
+            y, convout, smean, sinvstd, w_bf16, gamma, beta = per[i]
             inp = per[i - 1][0] if i > 0 else x
             direct = (getattr(m, "_managed", False)
                       and m.weight.grad is not None)
@@ -255,7 +257,7 @@ class ResBlockFn(torch.autograd.Function):
             # downsample dgrad accumulates into the main-path dx below
             dx_accum = dres if (i == 0 and ds_mod is None) else None
             dx, dw, dgamma, dbeta, dres_i = _C().conv_bn_act_bwd(
-                cur, y, inp, w_bf16, _rsck(m, w_bf16), convout, gamma,
+                cur, y, inp, w_bf16, _rsck(m, w_bf16), convout, gamma, beta,
                 smean, sinvstd, m.stride, m.padding, m.act, True, has_res,
                 m.weight.grad if direct else None,
                 m.bn_weight.grad if direct else None,
@@ -267,12 +269,13 @@ class ResBlockFn(torch.autograd.Function):
             cur = dx
 
         if ds_mod is not None:
-            y_ds, convout_ds, smean_ds, sinvstd_ds, w_ds, gamma_ds = ds
+            (y_ds, convout_ds, smean_ds, sinvstd_ds, w_ds, gamma_ds,
+             beta_ds) = ds
             direct = (getattr(ds_mod, "_managed", False)
                       and ds_mod.weight.grad is not None)
             dxds, dw, dgamma, dbeta, _ = _C().conv_bn_act_bwd(
                 dres, y_ds, x, w_ds, _rsck(ds_mod, w_ds), convout_ds,
-                gamma_ds, smean_ds, sinvstd_ds, ds_mod.stride,
+                gamma_ds, beta_ds, smean_ds, sinvstd_ds, ds_mod.stride,
                 ds_mod.padding, False, True, False,
                 ds_mod.weight.grad if direct else None,
                 ds_mod.bn_weight.grad if direct else None,

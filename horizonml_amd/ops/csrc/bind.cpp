@@ -41,12 +41,13 @@ void launch_bn_apply_f32(const float*, const void*, void*, void*,
 void launch_stats_reduce(const float*, float*, long, int, int, hipStream_t);
 void launch_cast_f32_bf16(const float*, void*, long, int, int, hipStream_t);
 void launch_bnact_bwd_reduce(const void*, const void*, const void*,
-                             const float*, const float*, float*, float*, long,
-                             int, int, hipStream_t);
+                             const float*, const float*, const float*,
+                             const float*, float*, float*, long, int, int,
+                             hipStream_t);
 void launch_bn_bwd_apply(const void*, const void*, const void*, const float*,
                          const float*, const float*, const float*,
-                         const float*, void*, void*, long, int, int,
-                         hipStream_t);
+                         const float*, const float*, void*, void*, long, int,
+                         int, hipStream_t);
 void launch_maxpool_fwd(const void*, void*, unsigned char*, int, int, int,
                         int, int, int, hipStream_t);
 void launch_maxpool_bwd(const void*, const unsigned char*, void*, int, int,
@@ -347,8 +348,8 @@ std::vector<Tensor> conv_bn_act_fwd(
 
 std::vector<Tensor> conv_bn_act_bwd(
     Tensor dy, Tensor y, Tensor x, Tensor w, Tensor w_rsck, Tensor convout,
-    Tensor gamma, Tensor save_mean, Tensor save_invstd, int64_t stride,
-    int64_t pad, bool act, bool need_dx, bool has_res,
+    Tensor gamma, Tensor beta, Tensor save_mean, Tensor save_invstd,
+    int64_t stride, int64_t pad, bool act, bool need_dx, bool has_res,
     c10::optional<Tensor> dw_out, c10::optional<Tensor> dgamma_out,
     c10::optional<Tensor> dbeta_out, c10::optional<Tensor> dx_accum) {
   // Direct-grad mode: when dw_out/dgamma_out/dbeta_out are given they are
@@ -364,13 +365,17 @@ std::vector<Tensor> conv_bn_act_bwd(
   auto st = cur_stream();
 
   bool direct = dw_out.has_value();
+  // ReLU-mask source: residual epilogue needs y; otherwise the mask is
+  // recovered from convout via BN algebra and y is never read
+  int mask_mode = act ? (has_res ? 1 : 2) : 0;
   Tensor sum_dz = direct ? *dbeta_out : at::zeros({K}, fopt);
   Tensor sum_dzx = direct ? *dgamma_out : at::zeros({K}, fopt);
   launch_bnact_bwd_reduce(dy.data_ptr(), y.data_ptr(), convout.data_ptr(),
                           save_mean.data_ptr<float>(),
                           save_invstd.data_ptr<float>(),
+                          gamma.data_ptr<float>(), beta.data_ptr<float>(),
                           sum_dz.data_ptr<float>(), sum_dzx.data_ptr<float>(),
-                          (long)p.M, K, act ? 1 : 0, st);
+                          (long)p.M, K, mask_mode, st);
 
   Tensor dconv = empty_cl_bf16(p.Nb, K, p.Ho, p.Wo, x);
   Tensor dres;
@@ -382,8 +387,9 @@ std::vector<Tensor> conv_bn_act_bwd(
   launch_bn_bwd_apply(dy.data_ptr(), y.data_ptr(), convout.data_ptr(),
                       save_mean.data_ptr<float>(),
                       save_invstd.data_ptr<float>(), gamma.data_ptr<float>(),
+                      beta.data_ptr<float>(),
                       sum_dz.data_ptr<float>(), sum_dzx.data_ptr<float>(),
-                      dconv.data_ptr(), dres_ptr, (long)p.M, K, act ? 1 : 0,
+                      dconv.data_ptr(), dres_ptr, (long)p.M, K, mask_mode,
                       st);
 
   Tensor dw = direct ? *dw_out

@@ -83,6 +83,35 @@ DEV V8 load8_a(const bf16* __restrict__ src, const ConvP& p, int m, int k) {
     int a = a_addr<MODE>(p, m, k, ok);
     if (ok) v.u = *(const uint4*)(src + a);
     else v.u = uint4{0, 0, 0, 0};
+  } else if (MODE == 1) {
+    // scalar x-gather (C not a multiple of 8, e.g. the stem's C=3): one
+    // (r,s,c) decomposition per vec8 and incremental carry across the 8
+    // elements — the naive per-element a_addr costs ~5 integer divides
+    // each and dominated the stem conv
+    int n = m / (p.Ho * p.Wo), hw = m % (p.Ho * p.Wo);
+    int ho = hw / p.Wo, wo = hw % p.Wo;
+    int r = k / (p.S * p.C), rm = k % (p.S * p.C);
+    int s = rm / p.C, c = rm % p.C;
+    int hi = ho * p.str - p.pad + r, wi = wo * p.str - p.pad + s;
+    const bool mok = m < p.M;
+    int kk = k;
+#pragma unroll
+    for (int e = 0; e < 8; e++) {
+      bool ok = mok && kk < p.Kd && hi >= 0 && hi < p.H && wi >= 0 &&
+                wi < p.W;
+      v.e[e] = ok ? src[((n * p.H + hi) * p.W + wi) * p.C + c] : (bf16)0.f;
+      kk++;
+      if (++c == p.C) {
+        c = 0;
+        ++wi;
+        if (++s == p.S) {
+          s = 0;
+          wi -= p.S;
+          ++r;
+          ++hi;
+        }
+      }
+    }
   } else {
     for (int e = 0; e < 8; e++) {
       bool ok;

@@ -115,11 +115,16 @@ __global__ __launch_bounds__(256) void k_cast_f32_bf16(
 // ------------------------------------------------------ BN+act backward ----
 // Pass 1: per-channel Σdz and Σ(dz·xhat) where dz = dy·relu'(y).
 // grid: (cdiv(C,64), msplit); block 256 = 4 m-lanes × 64 channels.
+// mask_mode: 0 = no activation (dz = dy), 1 = ReLU mask from y (residual
+// epilogue: y = relu(bn+res)), 2 = ReLU mask derived from convout —
+// bn(x) > 0 ⇔ a·x+b > 0 with a = γ·invstd, b = β − γ·mean·invstd — so the
+// y tensor is never read (one fewer stream for 12 of 20 ResNet18 convs).
 __global__ __launch_bounds__(256) void k_bnact_bwd_reduce(
     const bf16* __restrict__ dy, const bf16* __restrict__ yout,
     const bf16* __restrict__ x, const float* __restrict__ save_mean,
-    const float* __restrict__ save_invstd, float* __restrict__ sum_dz,
-    float* __restrict__ sum_dzx, long M, int C, int act, long mchunk) {
+    const float* __restrict__ save_invstd, const float* __restrict__ gamma,
+    const float* __restrict__ beta, float* __restrict__ sum_dz,
+    float* __restrict__ sum_dzx, long M, int C, int mask_mode, long mchunk) {
   __shared__ float sdz[4][64];
   __shared__ float sdzx[4][64];
   const int c = blockIdx.x * 64 + (threadIdx.x & 63);
@@ -129,13 +134,19 @@ __global__ __launch_bounds__(256) void k_bnact_bwd_reduce(
   float a_dz = 0.f, a_dzx = 0.f;
   if (c < C) {
     const float mean = save_mean[c], invstd = save_invstd[c];
+    const float ga = gamma[c] * invstd;
+    const float gb = beta[c] - mean * ga;
     for (long m = mbeg + mlane; m < mend; m += 4) {
       long i = m * C + c;
       float g = b2f(dy[i]);
-      if (act && b2f(yout[i]) <= 0.f) g = 0.f;
-      float xh = (b2f(x[i]) - mean) * invstd;
+      float xv = b2f(x[i]);
+      if (mask_mode == 1) {
+        if (b2f(yout[i]) <= 0.f) g = 0.f;
+      } else if (mask_mode == 2) {
+        if (fmaf(ga, xv, gb) <= 0.f) g = 0.f;
+      }
       a_dz += g;
-      a_dzx += g * xh;
+      a_dzx += g * (xv - mean) * invstd;
     }
   }
   sdz[mlane][threadIdx.x & 63] = a_dz;
@@ -158,19 +169,25 @@ __global__ __launch_bounds__(256) void k_bn_bwd_apply(
     const bf16* __restrict__ dy, const bf16* __restrict__ yout,
     const bf16* __restrict__ x, const float* __restrict__ save_mean,
     const float* __restrict__ save_invstd, const float* __restrict__ gamma,
-    const float* __restrict__ sum_dz, const float* __restrict__ sum_dzx,
-    bf16* __restrict__ dconv, bf16* __restrict__ dres, long M, int C,
-    int act) {
+    const float* __restrict__ beta, const float* __restrict__ sum_dz,
+    const float* __restrict__ sum_dzx, bf16* __restrict__ dconv,
+    bf16* __restrict__ dres, long M, int C, int mask_mode) {
   const float invM = 1.f / (float)M;
   long total = M * C;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += (long)gridDim.x * blockDim.x) {
     int c = (int)(i % C);
-    float g = b2f(dy[i]);
-    if (act && b2f(yout[i]) <= 0.f) g = 0.f;
-    if (dres != nullptr) dres[i] = f2b(g);
     float mean = save_mean[c], invstd = save_invstd[c];
-    float xh = (b2f(x[i]) - mean) * invstd;
+    float g = b2f(dy[i]);
+    float xv = b2f(x[i]);
+    if (mask_mode == 1) {
+      if (b2f(yout[i]) <= 0.f) g = 0.f;
+    } else if (mask_mode == 2) {
+      float ga = gamma[c] * invstd;
+      if (fmaf(ga, xv, beta[c] - mean * ga) <= 0.f) g = 0.f;
+    }
+    if (dres != nullptr) dres[i] = f2b(g);
+    float xh = (xv - mean) * invstd;
     float v = gamma[c] * invstd *
               (g - sum_dz[c] * invM - xh * sum_dzx[c] * invM);
     dconv[i] = f2b(v);
@@ -587,8 +604,9 @@ void launch_cast_f32_bf16(const float* src, void* dst, long n, int nsplit,
 
 void launch_bnact_bwd_reduce(const void* dy, const void* yout, const void* x,
                              const float* smean, const float* sinvstd,
+                             const float* gamma, const float* beta,
                              float* sum_dz, float* sum_dzx, long M, int C,
-                             int act, hipStream_t st) {
+                             int mask_mode, hipStream_t st) {
   // sum_dz/sum_dzx must be pre-zeroed (they are accumulated atomically)
   int cblocks = (C + 63) / 64;
   int msplit = (int)min((long)256, max((long)1, (long)(768 / cblocks)));
@@ -597,17 +615,19 @@ void launch_bnact_bwd_reduce(const void* dy, const void* yout, const void* x,
   dim3 grid(cblocks, msplit);
   k_bnact_bwd_reduce<<<grid, 256, 0, st>>>(
       (const bf16*)dy, (const bf16*)yout, (const bf16*)x, smean, sinvstd,
-      sum_dz, sum_dzx, M, C, act, mchunk);
+      gamma, beta, sum_dz, sum_dzx, M, C, mask_mode, mchunk);
 }
 
 void launch_bn_bwd_apply(const void* dy, const void* yout, const void* x,
                          const float* smean, const float* sinvstd,
-                         const float* gamma, const float* sum_dz,
-                         const float* sum_dzx, void* dconv, void* dres,
-                         long M, int C, int act, hipStream_t st) {
+                         const float* gamma, const float* beta,
+                         const float* sum_dz, const float* sum_dzx,
+                         void* dconv, void* dres, long M, int C,
+                         int mask_mode, hipStream_t st) {
   k_bn_bwd_apply<<<gsz(M * (long)C), 256, 0, st>>>(
       (const bf16*)dy, (const bf16*)yout, (const bf16*)x, smean, sinvstd,
-      gamma, sum_dz, sum_dzx, (bf16*)dconv, (bf16*)dres, M, C, act);
+      gamma, beta, sum_dz, sum_dzx, (bf16*)dconv, (bf16*)dres, M, C,
+      mask_mode);
 }
 
 void launch_maxpool_fwd(const void* x, void* y, unsigned char* idx, int Nb,
