@@ -106,8 +106,10 @@ def main():
             _ops.extension().flush_wgrad()    # wgrads complete before sync
             comm_buf.copy_(mgr.grad)          # pack f32 -> bf16 (half bytes)
             dist.all_reduce(comm_buf)         # RCCL over xGMI
-            mgr.grad.copy_(comm_buf).mul_(inv_world)
-        opt.step()                            # fused adam + zero_grad + rsck
+            # fused optimizer consumes the reduced bf16 buffer directly
+            opt.step(grad_bf16=comm_buf, grad_scale=inv_world)
+        else:
+            opt.step()                        # fused adam + zero_grad + rsck
         return loss
 
     # ---- warmup + graph capture -----------------------------------------

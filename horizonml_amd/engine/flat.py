@@ -108,13 +108,17 @@ class HorizonAdam:
         self.v = torch.zeros_like(mgr.master)
         self.step_t = torch.zeros(1, device=dev)
 
-    def step(self, zero_grad: bool = True):
+    def step(self, zero_grad: bool = True, grad_bf16=None,
+             grad_scale: float = 1.0):
+        """``grad_bf16``: consume an all-reduced bf16 gradient buffer
+        directly (× ``grad_scale``) — skips the DP unpack pass."""
         _ops.extension().flush_wgrad()  # batched deferred weight grads
         _ops.extension().adam_step(self.mgr.master, self.mgr.grad, self.m,
                                    self.v, self.mgr.shadow, self.step_t,
                                    self.lr, self.betas[0], self.betas[1],
                                    self.eps, self.wd, zero_grad,
-                                   self.mgr.stats_arena)
+                                   self.mgr.stats_arena, grad_bf16,
+                                   grad_scale)
         self.mgr.refresh_rsck()
 
 
@@ -125,10 +129,11 @@ class HorizonSGD:
         self.lr, self.mu, self.wd = lr, momentum, weight_decay
         self.mom = (torch.zeros_like(mgr.master) if momentum > 0 else None)
 
-    def step(self, zero_grad: bool = True):
+    def step(self, zero_grad: bool = True, grad_bf16=None,
+             grad_scale: float = 1.0):
         _ops.extension().flush_wgrad()  # batched deferred weight grads
         _ops.extension().sgd_step(self.mgr.master, self.mgr.grad, self.mom,
                                   self.mgr.shadow, self.lr, self.mu, self.wd,
-                                  zero_grad)
+                                  zero_grad, grad_bf16, grad_scale)
         self.mgr.stats_arena.zero_()
         self.mgr.refresh_rsck()

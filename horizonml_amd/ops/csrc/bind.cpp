@@ -62,9 +62,9 @@ void launch_ce_fwd_bwd(const float*, const long*, float*, float*, int, int,
                        hipStream_t);
 void launch_adam_step(float*, float*, float*, float*, void*, const float*,
                       long, float, float, float, float, float, int, float*,
-                      long, hipStream_t);
+                      long, const void*, float, hipStream_t);
 void launch_sgd_step(float*, float*, float*, void*, long, float, float, float,
-                     int, hipStream_t);
+                     int, const void*, float, hipStream_t);
 void launch_permute_krsc_rsck(const void*, void*, const int*, int, int,
                               hipStream_t);
 void launch_grad_divergence(const float*, float*, float*, float*, long, int,
@@ -579,7 +579,8 @@ Tensor wgrad_only(Tensor x, Tensor dz, int64_t K, int64_t R, int64_t S,
 void adam_step(Tensor master, Tensor grad, Tensor m, Tensor v,
                c10::optional<Tensor> shadow, Tensor step_t, double lr,
                double b1, double b2, double eps, double wd, bool zero_grad,
-               c10::optional<Tensor> extra_zero) {
+               c10::optional<Tensor> extra_zero,
+               c10::optional<Tensor> grad_bf16, double grad_scale) {
   check_f32(master, "master");
   launch_adam_step(master.data_ptr<float>(), grad.data_ptr<float>(),
                    m.data_ptr<float>(), v.data_ptr<float>(),
@@ -590,18 +591,22 @@ void adam_step(Tensor master, Tensor grad, Tensor m, Tensor v,
                    extra_zero.has_value() ? extra_zero->data_ptr<float>()
                                           : nullptr,
                    extra_zero.has_value() ? extra_zero->numel() : 0,
-                   cur_stream());
+                   grad_bf16.has_value() ? grad_bf16->data_ptr() : nullptr,
+                   (float)grad_scale, cur_stream());
 }
 
 void sgd_step(Tensor master, Tensor grad, c10::optional<Tensor> mom,
               c10::optional<Tensor> shadow, double lr, double mu, double wd,
-              bool zero_grad) {
+              bool zero_grad, c10::optional<Tensor> grad_bf16,
+              double grad_scale) {
   check_f32(master, "master");
   launch_sgd_step(master.data_ptr<float>(), grad.data_ptr<float>(),
                   mom.has_value() ? mom->data_ptr<float>() : nullptr,
                   shadow.has_value() ? shadow->data_ptr() : nullptr,
                   master.numel(), (float)lr, (float)mu, (float)wd,
-                  zero_grad ? 1 : 0, cur_stream());
+                  zero_grad ? 1 : 0,
+                  grad_bf16.has_value() ? grad_bf16->data_ptr() : nullptr,
+                  (float)grad_scale, cur_stream());
 }
 
 void permute_krsc_rsck(Tensor src, Tensor dst, Tensor meta,

@@ -438,12 +438,14 @@ __global__ __launch_bounds__(256) void k_ce_fwd_bwd(
 // Flat-buffer Adam: master f32, grad f32 (zeroed after), m/v f32, bf16 shadow
 // emitted for every element.  `step_t` is a device scalar so the kernel is
 // hipGraph-replayable (bias correction computed on device).
+// gsrc (optional): all-reduced bf16 gradient consumed directly with a
+// 1/world scale — skips the DP path's unpack copy+mul pass.
 __global__ __launch_bounds__(256) void k_adam_step(
     float* __restrict__ master, float* __restrict__ grad,
     float* __restrict__ m, float* __restrict__ v, bf16* __restrict__ shadow,
     const float* __restrict__ step_t, long n, float lr, float b1, float b2,
     float eps, float wd, int zero_grad, float* __restrict__ extra_zero,
-    long n_extra) {
+    long n_extra, const bf16* __restrict__ gsrc, float gscale) {
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n_extra;
        i += (long)gridDim.x * blockDim.x)
     extra_zero[i] = 0.f;
@@ -451,7 +453,7 @@ __global__ __launch_bounds__(256) void k_adam_step(
   float bc1 = 1.f - __powf(b1, t), bc2 = 1.f - __powf(b2, t);
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (long)gridDim.x * blockDim.x) {
-    float g = grad[i];
+    float g = gsrc != nullptr ? b2f(gsrc[i]) * gscale : grad[i];
     float w = master[i];
     if (wd != 0.f) g += wd * w;
     float mi = b1 * m[i] + (1.f - b1) * g;
@@ -468,10 +470,11 @@ __global__ __launch_bounds__(256) void k_adam_step(
 __global__ __launch_bounds__(256) void k_sgd_step(
     float* __restrict__ master, float* __restrict__ grad,
     float* __restrict__ mom, bf16* __restrict__ shadow, long n, float lr,
-    float mu, float wd, int zero_grad) {
+    float mu, float wd, int zero_grad, const bf16* __restrict__ gsrc,
+    float gscale) {
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (long)gridDim.x * blockDim.x) {
-    float g = grad[i];
+    float g = gsrc != nullptr ? b2f(gsrc[i]) * gscale : grad[i];
     float w = master[i];
     if (wd != 0.f) g += wd * w;
     float u = (mom != nullptr) ? (mu * mom[i] + g) : g;
@@ -683,18 +686,21 @@ void launch_ce_fwd_bwd(const float* logits, const long* target, float* loss,
 void launch_adam_step(float* master, float* grad, float* m, float* v,
                       void* shadow, const float* step_t, long n, float lr,
                       float b1, float b2, float eps, float wd, int zero_grad,
-                      float* extra_zero, long n_extra, hipStream_t st) {
+                      float* extra_zero, long n_extra, const void* gsrc,
+                      float gscale, hipStream_t st) {
   k_inc_step<<<1, 1, 0, st>>>((float*)step_t);
   k_adam_step<<<gsz(n), 256, 0, st>>>(master, grad, m, v, (bf16*)shadow,
                                       step_t, n, lr, b1, b2, eps, wd,
-                                      zero_grad, extra_zero, n_extra);
+                                      zero_grad, extra_zero, n_extra,
+                                      (const bf16*)gsrc, gscale);
 }
 
 void launch_sgd_step(float* master, float* grad, float* mom, void* shadow,
                      long n, float lr, float mu, float wd, int zero_grad,
-                     hipStream_t st) {
+                     const void* gsrc, float gscale, hipStream_t st) {
   k_sgd_step<<<gsz(n), 256, 0, st>>>(master, grad, mom, (bf16*)shadow, n, lr,
-                                     mu, wd, zero_grad);
+                                     mu, wd, zero_grad, (const bf16*)gsrc,
+                                     gscale);
 }
 
 void launch_permute_krsc_rsck(const void* src, void* dst, const int* meta,

@@ -197,7 +197,7 @@ def test_fused_adam_matches_torch():
     step_t = torch.zeros(1, device="cuda")
     for _ in range(3):
         _C().adam_step(mg, gg, m, v, shadow, step_t, 1e-3, 0.9, 0.999, 1e-8,
-                       0.0, False, None)
+                       0.0, False, None, None, 1.0)
     assert rel(mg, p.detach()) < 1e-5
     assert rel(shadow, p.detach()) < 1e-2
 
@@ -215,7 +215,7 @@ def test_fused_sgd_matches_torch():
     mom = torch.zeros(n, device="cuda")
     for _ in range(3):
         opt.step()
-        _C().sgd_step(mg, gg, mom, None, 0.1, 0.9, 0.0, False)
+        _C().sgd_step(mg, gg, mom, None, 0.1, 0.9, 0.0, False, None, 1.0)
     assert rel(mg, p.detach()) < 1e-5
 
 
@@ -401,3 +401,29 @@ def test_wgrad_accumulates_across_backwards():
         torch.cuda.synchronize()
         r = ((acc - ref).norm() / ref.norm()).item()
         assert r < 1e-4, f"defer={defer}: accumulation broken, rel {r}"
+
+
+def test_adam_bf16_grad_source():
+    """opt.step(grad_bf16=..., grad_scale=s) must equal the normal path on
+    bf16-representable grads (the DP all-reduce consumption path)."""
+    from horizonml_amd.engine.flat import FlatParamManager, HorizonAdam
+    from horizonml_amd.models import resnet18
+    dev = torch.device("cuda", 0)
+    out = []
+    for use_bf16 in (False, True):
+        torch.manual_seed(0)
+        model = resnet18(num_classes=10).to(dev)
+        mgr = FlatParamManager(model, dev)
+        opt = HorizonAdam(mgr, lr=1e-2)
+        torch.manual_seed(5)
+        g = torch.randn_like(mgr.grad).to(torch.bfloat16)
+        if use_bf16:
+            mgr.grad.zero_()
+            opt.step(grad_bf16=(g * 2.0), grad_scale=0.5)
+        else:
+            mgr.grad.copy_(g.float())
+            opt.step()
+        torch.cuda.synchronize()
+        out.append(mgr.master.clone())
+    assert torch.equal(out[0], out[1]), \
+        f"max diff {(out[0] - out[1]).abs().max().item()}"
