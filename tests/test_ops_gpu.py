@@ -427,3 +427,50 @@ def test_adam_bf16_grad_source():
         out.append(mgr.master.clone())
     assert torch.equal(out[0], out[1]), \
         f"max diff {(out[0] - out[1]).abs().max().item()}"
+
+
+def test_resnet50_gpu_matches_cpu_small():
+    """Bottleneck blocks (fused 3-conv residual path) vs CPU fp32 on a
+    small image — guards the hybrid DP×PP config's model on gfx950."""
+    from horizonml_amd.models import build_model
+    torch.manual_seed(0)
+    cpu = build_model("resnet50", num_classes=10)
+    gpu = build_model("resnet50", num_classes=10)
+    gpu.load_state_dict(cpu.state_dict())
+    gpu = gpu.cuda()
+    x = torch.randn(4, 3, 64, 64)
+    ref = cpu(x)
+    out = gpu(to_gpu_cl(x))
+    # bf16 noise through 50 layers: measured ~0.23 rel (the fused block
+    # path is closer to fp32 than the unfused GPU path at 0.27)
+    assert rel(out, ref) < 0.4, f"logits rel={rel(out, ref)}"
+    # backward runs end to end and produces finite grads everywhere
+    from horizonml_amd.models._functional_gpu import cross_entropy
+    y = torch.randint(0, 10, (4,)).cuda()
+    loss = cross_entropy(gpu(to_gpu_cl(x)), y)
+    loss.backward()
+    for n, p in gpu.named_parameters():
+        assert p.grad is not None and torch.isfinite(p.grad).all(), n
+
+
+def test_resnet50_imagenet_shape_trains():
+    """ResNet50 on 224x224 synthetic (BASELINE config #5 shape): loss drops
+    when overfitting one batch through the native kernels."""
+    from horizonml_amd.engine.flat import FlatParamManager, HorizonAdam
+    from horizonml_amd.models import build_model
+    from horizonml_amd.models._functional_gpu import cross_entropy
+    torch.manual_seed(0)
+    dev = torch.device("cuda", 0)
+    model = build_model("resnet50", num_classes=1000).to(dev)
+    mgr = FlatParamManager(model, dev)
+    opt = HorizonAdam(mgr, lr=1e-3)
+    x = torch.randn(8, 3, 224, 224, device=dev).to(
+        memory_format=torch.channels_last).to(torch.bfloat16)
+    y = torch.randint(0, 1000, (8,), device=dev)
+    losses = []
+    for _ in range(8):
+        loss = cross_entropy(model(x), y)
+        loss.backward()
+        opt.step()
+        losses.append(float(loss.detach()))
+    assert losses[-1] < losses[0], f"no learning: {losses}"
