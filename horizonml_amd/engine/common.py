@@ -47,22 +47,23 @@ class GradDivergenceProbe:
         if not self.params:
             raise ValueError("no trainable params for divergence probe")
         dev = self.params[0].device
-        total = sum(p.numel() for p in self.params)
-        self.prev = torch.zeros(total, dtype=torch.float32, device=dev)
+        self.prev_list = [torch.zeros_like(p, dtype=torch.float32)
+                          for p in self.params]
         self.sum = torch.zeros((), dtype=torch.float32, device=dev)
         self.n = 0
         self.first = True
 
     @torch.no_grad()
     def step(self):
-        grads = [(p.grad if p.grad is not None
-                  else torch.zeros_like(p)).flatten().float()
+        grads = [(p.grad if p.grad is not None else torch.zeros_like(p))
                  for p in self.params]
-        flat = torch.cat(grads)
         if not self.first:
-            self.sum += torch.linalg.vector_norm(flat - self.prev)
+            diffs = torch._foreach_sub(grads, self.prev_list)
+            norms = torch._foreach_norm(diffs)
+            self.sum += torch.stack([n.float() for n in norms]) \
+                .square().sum().sqrt()
             self.n += 1
-        self.prev.copy_(flat)
+        torch._foreach_copy_(self.prev_list, grads)
         self.first = False
 
     def epoch_value(self) -> float:
@@ -77,9 +78,12 @@ def build_optimizer(params, name: str = "adam", lr: float = 1e-3,
     """Reference default: Adam(lr=1e-3) (``data_parallel_train.py:205``).
     The north star also names the SGD step — both supported."""
     name = name.lower()
+    # foreach=True batches the per-parameter update into multi-tensor
+    # kernels — the eager engine paths are launch-bound otherwise
     if name == "adam":
-        return torch.optim.Adam(params, lr=lr, weight_decay=weight_decay)
+        return torch.optim.Adam(params, lr=lr, weight_decay=weight_decay,
+                                foreach=True)
     if name == "sgd":
         return torch.optim.SGD(params, lr=lr, momentum=momentum,
-                               weight_decay=weight_decay)
+                               weight_decay=weight_decay, foreach=True)
     raise ValueError(f"unknown optimizer {name!r}")

@@ -1,0 +1,39 @@
+"""Checkpoint/resume round-trip (an addition over the reference, which has
+no persistence — SURVEY.md §5.4)."""
+import torch
+
+from horizonml_amd.models import resnet18
+from horizonml_amd.utils.checkpoint import load_checkpoint, save_checkpoint
+
+
+def test_checkpoint_roundtrip(tmp_path):
+    torch.manual_seed(0)
+    m1 = resnet18(num_classes=10)
+    opt = torch.optim.Adam(m1.parameters(), lr=1e-3)
+    x = torch.randn(4, 3, 32, 32)
+    y = torch.randint(0, 10, (4,))
+    loss = torch.nn.functional.cross_entropy(m1(x), y)
+    loss.backward()
+    opt.step()
+    path = str(tmp_path / "ckpt.pt")
+    save_checkpoint(path, m1, opt, epoch=3, extra={"note": "t"})
+
+    torch.manual_seed(123)
+    m2 = resnet18(num_classes=10)
+    opt2 = torch.optim.Adam(m2.parameters(), lr=1e-3)
+    state = load_checkpoint(path, m2, opt2)
+    assert state["epoch"] == 3
+    assert state["extra"]["note"] == "t"
+    for (n1, p1), (n2, p2) in zip(m1.named_parameters(),
+                                  m2.named_parameters()):
+        assert torch.equal(p1, p2), n1
+    # optimizer moments restored
+    s1 = opt.state_dict()["state"]
+    s2 = opt2.state_dict()["state"]
+    assert set(s1.keys()) == set(s2.keys())
+    k = next(iter(s1))
+    assert torch.equal(s1[k]["exp_avg"], s2[k]["exp_avg"])
+    # training continues identically from the restored state
+    l1 = torch.nn.functional.cross_entropy(m1(x), y)
+    l2 = torch.nn.functional.cross_entropy(m2(x), y)
+    assert torch.allclose(l1, l2)
