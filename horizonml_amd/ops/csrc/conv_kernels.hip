@@ -152,8 +152,12 @@ __global__ __launch_bounds__(256) void k_conv_mfma(
     const bf16* __restrict__ A, const bf16* __restrict__ Bw,
     bf16* __restrict__ Y, float* __restrict__ ws_out,
     float* __restrict__ stats, ConvP p, int Ntot, int kchunk, int accum) {
-  __shared__ bf16 As[64 * LDA];
-  __shared__ bf16 Bs[64 * LDA];
+  // Double-buffered LDS, ONE barrier per K-step, 2-deep register
+  // prefetch: tile k+2's global loads are in flight while tile k computes,
+  // so the ~900-cycle HBM latency is fully hidden even at 1 block/CU
+  // (these conv grids are small — latency, not bandwidth, bounds them).
+  __shared__ bf16 As[2][64 * LDA];
+  __shared__ bf16 Bs[2][64 * LDA];
 
   const int m0 = blockIdx.y * 64, n0 = blockIdx.x * 64;
   const int kbeg = SPLIT ? blockIdx.z * kchunk : 0;
@@ -166,32 +170,48 @@ __global__ __launch_bounds__(256) void k_conv_mfma(
 
   f32x4 acc[2][2] = {};
 
-  // prologue: prefetch the first K-tile into registers
-  V8 a_nx = load8_a<MODE, VECA>(A, p, m0 + srow, kbeg + scol);
-  V8 b_nx = load8_b<MODE, VECB>(Bw, p, n0 + srow, kbeg + scol, Ntot);
+  // prologue: tile 0 -> LDS[0]; tile 1 -> registers
+  {
+    V8 a0 = load8_a<MODE, VECA>(A, p, m0 + srow, kbeg + scol);
+    V8 b0 = load8_b<MODE, VECB>(Bw, p, n0 + srow, kbeg + scol, Ntot);
+    *(V8*)&As[0][srow * LDA + scol] = a0;
+    *(V8*)&Bs[0][srow * LDA + scol] = b0;
+  }
+  V8 a_nx, b_nx;
+  if (kbeg + 32 < kend) {
+    a_nx = load8_a<MODE, VECA>(A, p, m0 + srow, kbeg + 32 + scol);
+    b_nx = load8_b<MODE, VECB>(Bw, p, n0 + srow, kbeg + 32 + scol, Ntot);
+  }
+  __syncthreads();
 
+  int buf = 0;
   for (int k0 = kbeg; k0 < kend; k0 += 32) {
-    *(V8*)&As[srow * LDA + scol] = a_nx;
-    *(V8*)&Bs[srow * LDA + scol] = b_nx;
-    __syncthreads();
-    // issue next tile's loads now: latency hides under ds_read + MFMA
+    // stage tile k+1 from registers into the OTHER buffer (its readers
+    // synchronized at the previous barrier), then issue tile k+2's loads
     if (k0 + 32 < kend) {
-      a_nx = load8_a<MODE, VECA>(A, p, m0 + srow, k0 + 32 + scol);
-      b_nx = load8_b<MODE, VECB>(Bw, p, n0 + srow, k0 + 32 + scol, Ntot);
+      *(V8*)&As[buf ^ 1][srow * LDA + scol] = a_nx;
+      *(V8*)&Bs[buf ^ 1][srow * LDA + scol] = b_nx;
+      if (k0 + 64 < kend) {
+        a_nx = load8_a<MODE, VECA>(A, p, m0 + srow, k0 + 64 + scol);
+        b_nx = load8_b<MODE, VECB>(Bw, p, n0 + srow, k0 + 64 + scol, Ntot);
+      }
     }
     bf16x8 af[2], bf[2];
 #pragma unroll
     for (int mi = 0; mi < 2; mi++)
-      af[mi] = *(const bf16x8*)&As[(wr * 32 + mi * 16 + fr) * LDA + fk * 8];
+      af[mi] =
+          *(const bf16x8*)&As[buf][(wr * 32 + mi * 16 + fr) * LDA + fk * 8];
 #pragma unroll
     for (int ni = 0; ni < 2; ni++)
-      bf[ni] = *(const bf16x8*)&Bs[(wc * 32 + ni * 16 + fr) * LDA + fk * 8];
+      bf[ni] =
+          *(const bf16x8*)&Bs[buf][(wc * 32 + ni * 16 + fr) * LDA + fk * 8];
 #pragma unroll
     for (int mi = 0; mi < 2; mi++)
 #pragma unroll
       for (int ni = 0; ni < 2; ni++)
         acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+    buf ^= 1;
     __syncthreads();
   }
 
