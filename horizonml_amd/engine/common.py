@@ -87,3 +87,15 @@ def build_optimizer(params, name: str = "adam", lr: float = 1e-3,
         return torch.optim.SGD(params, lr=lr, momentum=momentum,
                                weight_decay=weight_decay, foreach=True)
     raise ValueError(f"unknown optimizer {name!r}")
+
+
+def progress_iter(loader, desc: str, enabled: bool = True):
+    """tqdm per-worker progress bar (reference ``data_parallel_train.py:102``
+    console parity); auto-disabled on non-TTY output so logs/CI stay clean."""
+    if not enabled:
+        return loader
+    try:
+        from tqdm import tqdm
+        return tqdm(loader, desc=desc, leave=False, disable=None)
+    except ImportError:
+        return loader

@@ -24,7 +24,8 @@ import torch
 import torch.nn.functional as F
 
 from ..data import get_dataloader
-from ..engine.common import GradDivergenceProbe, Meters, build_optimizer
+from ..engine.common import (GradDivergenceProbe, Meters,
+                             build_optimizer, progress_iter)
 from ..models import build_model
 from ..parallel import BucketedDataParallel
 from ..profiling.metrics import (EpochMetrics, MetricsWriter,
@@ -80,7 +81,8 @@ def train_dp(ctx: DistContext, epochs: int, sample_size: int, logs_dir: str,
         epoch_start = time.time()
         cpu_samples, mem_samples = [], []
 
-        for step, (x, y) in enumerate(loader):
+        for step, (x, y) in enumerate(progress_iter(
+                loader, f"dp r{rank} e{epoch + 1}", log_progress)):
             prof.step_begin()
             cpu, mem = sample_host_resources(proc)
             cpu_samples.append(cpu)

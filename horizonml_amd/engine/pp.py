@@ -16,7 +16,8 @@ import torch
 import torch.nn.functional as F
 
 from ..data import get_dataloader
-from ..engine.common import GradDivergenceProbe, Meters, build_optimizer
+from ..engine.common import (GradDivergenceProbe, Meters,
+                             build_optimizer, progress_iter)
 from ..models import build_model, partition_model
 from ..parallel.pipeline import PipelineStage
 from ..profiling.metrics import (EpochMetrics, MetricsWriter,
@@ -101,7 +102,8 @@ def train_pp(ctx: DistContext, epochs: int, sample_size: int, logs_dir: str,
         cpu_samples, mem_samples = [], []
         loss_sum, correct, count = 0.0, 0, 0
 
-        for x, y in loader:
+        for x, y in progress_iter(loader, f"pp s{stage_idx} "
+                                  f"e{epoch + 1}", log_progress):
             prof.step_begin()
             cpu, mem = sample_host_resources(proc)
             cpu_samples.append(cpu)

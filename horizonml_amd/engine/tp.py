@@ -15,7 +15,8 @@ import torch
 import torch.nn.functional as F
 
 from ..data import get_dataloader
-from ..engine.common import GradDivergenceProbe, Meters, build_optimizer
+from ..engine.common import (GradDivergenceProbe, Meters,
+                             build_optimizer, progress_iter)
 from ..parallel import BucketedDataParallel
 from ..parallel.tensor_parallel import replicated_parameters
 from ..parallel.tp_models import build_tp_resnet18
@@ -62,7 +63,8 @@ def train_tp(ctx: DistContext, epochs: int, sample_size: int, logs_dir: str,
             barrier(ctx)
         epoch_start = time.time()
         cpu_samples, mem_samples = [], []
-        for x, y in loader:
+        for x, y in progress_iter(loader, f"tp r{rank} e{epoch + 1}",
+                                  log_progress):
             prof.step_begin()
             cpu, mem = sample_host_resources(proc)
             cpu_samples.append(cpu)
