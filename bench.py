@@ -68,8 +68,13 @@ def main():
     assert torch.cuda.is_available(), "bench.py needs an MI355X"
     torch.cuda.set_device(local_rank)
     dev = torch.device("cuda", local_rank)
-    if world > 1:
+    # HZ_FORCE_COMM=1: exercise the full RCCL comm path (all-reduce inside
+    # the captured graph) on a single-rank communicator — 1-GPU validation
+    # of the multi-GPU code path
+    force_comm = os.environ.get("HZ_FORCE_COMM") == "1" and world == 1
+    if world > 1 or force_comm:
         dist.init_process_group("nccl", rank=rank, world_size=world)
+    use_comm = world > 1 or force_comm
 
     from horizonml_amd import ops as _ops
     from horizonml_amd.engine.flat import (FlatParamManager, HorizonAdam,
@@ -95,14 +100,14 @@ def main():
     x_static = pool_x[0].clone()
     y_static = pool_y[0].clone()
     comm_buf = (torch.zeros_like(mgr.grad, dtype=torch.bfloat16)
-                if world > 1 else None)
+                if use_comm else None)
     inv_world = 1.0 / world
 
     def train_step():
         logits = model(x_static)
         loss = cross_entropy(logits, y_static)
         loss.backward()
-        if world > 1:
+        if use_comm:
             _ops.extension().flush_wgrad()    # wgrads complete before sync
             comm_buf.copy_(mgr.grad)          # pack f32 -> bf16 (half bytes)
             dist.all_reduce(comm_buf)         # RCCL over xGMI
@@ -217,7 +222,7 @@ def main():
             },
         }
         print(json.dumps(out), flush=True)
-    if world > 1:
+    if world > 1 or force_comm:
         dist.destroy_process_group()
 
 
