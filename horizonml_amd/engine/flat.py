@@ -15,7 +15,7 @@ buffers, single-kernel updates):
 """
 from __future__ import annotations
 
-from typing import List, Optional
+from typing import List
 
 import torch
 import torch.nn as nn

@@ -16,8 +16,8 @@ import torch
 import torch.nn.functional as F
 
 from ..data import get_dataloader
-from ..engine.common import (GradDivergenceProbe, Meters,
-                             build_optimizer, progress_iter)
+from ..engine.common import (GradDivergenceProbe, build_optimizer,
+                             progress_iter)
 from ..models import build_model, partition_model
 from ..parallel.pipeline import PipelineStage
 from ..profiling.metrics import (EpochMetrics, MetricsWriter,

@@ -10,8 +10,6 @@ classifier logits f32.
 """
 from __future__ import annotations
 
-from typing import Optional
-
 import torch
 
 from .. import ops as _ops

@@ -27,7 +27,6 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
-from .. import ops as _ops
 
 
 def _use_hip(x: torch.Tensor) -> bool:

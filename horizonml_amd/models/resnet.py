@@ -13,7 +13,7 @@ deliberate improvement, not the parity default).
 """
 from __future__ import annotations
 
-from typing import List, Optional, Tuple
+from typing import List, Tuple
 
 import torch
 import torch.nn as nn

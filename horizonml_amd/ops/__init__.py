@@ -15,7 +15,6 @@ Dispatch policy:
 from __future__ import annotations
 
 import importlib
-import os
 
 _EXT = None
 _EXT_ERR: Exception | None = None

@@ -15,7 +15,6 @@ import os
 import traceback
 
 import numpy as np
-import pandas as pd
 
 from data_parallel_train import run_data_parallel
 from layer_model_parallel_train import run_model_parallel

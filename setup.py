@@ -10,7 +10,6 @@ No hipify, no CUDA paths — the kernels are native CDNA4 HIP.
 """
 import os
 import subprocess
-import sys
 
 from setuptools import setup
 from torch.utils.cpp_extension import BuildExtension, CppExtension

@@ -6,11 +6,7 @@ BucketedDataParallel hooks, the pipeline relay degenerate case, sharded-TP
 modules, hipEvent profilers and GPU resource sampling, all running through
 the gfx950 kernels.
 """
-import os
-
-import pandas as pd
 import pytest
-import torch
 
 pytestmark = pytest.mark.gpu
 

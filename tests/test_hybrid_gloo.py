@@ -7,7 +7,6 @@ only holds if the per-stage all-reduce over the DP sub-communicator works).
 """
 import os
 
-import pandas as pd
 import pytest
 import torch
 

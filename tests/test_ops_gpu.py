@@ -3,8 +3,6 @@
 All comparisons are against CPU fp32 compositions of the same op; tolerances
 reflect bf16 inputs with f32 accumulation.
 """
-import math
-
 import pytest
 import torch
 import torch.nn.functional as F

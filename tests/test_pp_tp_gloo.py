@@ -1,7 +1,6 @@
 """Pipeline + tensor-parallel correctness over 2-process CPU/gloo."""
 import os
 
-import pytest
 import torch
 import torch.multiprocessing as mp
 import torch.nn as nn
