@@ -70,7 +70,8 @@ class ConvBNActFn(torch.autograd.Function):
         dx, dw, dgamma, dbeta, dres = _C().conv_bn_act_bwd(
             dy, y, x, w_bf16, w_rsck if need_dx else w_bf16, convout, gamma,
             beta, smean, sinvstd, mod.stride, mod.padding, mod.act, need_dx,
-            ctx.has_res, dw_out, dg_out, db_out, None)
+            ctx.has_res, dw_out, dg_out, db_out, None, None, 0, False, None,
+            None)
         if direct:
             return (dx if need_dx else None, None, None, None,
                     dres if ctx.has_res else None, None)
@@ -243,6 +244,7 @@ class ResBlockFn(torch.autograd.Function):
         dres = None
         grads = [None] * (3 * n + (3 if ds_mod is not None else 0))
 
+        next_sums = None  # (sum_dz, sum_dzx) computed by conv i+1's bwd
         for i in range(n - 1, -1, -1):
             m = mods[i]
             y, convout, smean, sinvstd, w_bf16, gamma, beta = per[i]
@@ -254,12 +256,33 @@ class ResBlockFn(torch.autograd.Function):
             # conv's dgrad accumulates into dres; for downsample blocks the
             # downsample dgrad accumulates into the main-path dx below
             dx_accum = dres if (i == 0 and ds_mod is None) else None
+            # intra-block edge fusion: while producing dx (= conv i-1's
+            # dy), also complete conv i-1's BN-backward channel sums
+            fuse_up, up_mask, sums = None, 0, None
+            if i >= 1:
+                m_up = mods[i - 1]
+                (y_u, convout_u, smean_u, sinvstd_u, _w_u, gamma_u,
+                 beta_u) = per[i - 1]
+                if (getattr(m_up, "_managed", False)
+                        and m_up.weight.grad is not None):
+                    sums = (m_up.bn_bias.grad, m_up.bn_weight.grad)
+                else:
+                    sums = (torch.zeros_like(m_up.bn_bias),
+                            torch.zeros_like(m_up.bn_weight))
+                up_mask = 2 if m_up.act else 0
+                fuse_up = [convout_u, y_u, smean_u, sinvstd_u, gamma_u,
+                           beta_u, sums[0], sums[1]]
+            sums_ready = next_sums is not None
             dx, dw, dgamma, dbeta, dres_i = _C().conv_bn_act_bwd(
                 cur, y, inp, w_bf16, _rsck(m, w_bf16), convout, gamma, beta,
                 smean, sinvstd, m.stride, m.padding, m.act, True, has_res,
                 m.weight.grad if direct else None,
                 m.bn_weight.grad if direct else None,
-                m.bn_bias.grad if direct else None, dx_accum)
+                m.bn_bias.grad if direct else None, dx_accum,
+                fuse_up, up_mask, sums_ready,
+                next_sums[0] if sums_ready else None,
+                next_sums[1] if sums_ready else None)
+            next_sums = sums
             if has_res:
                 dres = dres_i
             if not direct:
@@ -277,7 +300,8 @@ class ResBlockFn(torch.autograd.Function):
                 ds_mod.padding, False, True, False,
                 ds_mod.weight.grad if direct else None,
                 ds_mod.bn_weight.grad if direct else None,
-                ds_mod.bn_bias.grad if direct else None, cur)
+                ds_mod.bn_bias.grad if direct else None, cur, None, 0,
+                False, None, None)
             if not direct:
                 grads[3 * n: 3 * n + 3] = [dw, dgamma, dbeta]
             cur = dxds  # == cur buffer, accumulated

@@ -40,6 +40,9 @@ void launch_bn_apply_f32(const float*, const void*, void*, void*,
                          int, int, int, hipStream_t);
 void launch_stats_reduce(const float*, float*, long, int, int, hipStream_t);
 void launch_cast_f32_bf16(const float*, void*, long, int, int, hipStream_t);
+void launch_cast_bnact(const float*, void*, long, int, int, int, const void*,
+                       const void*, const float*, const float*, const float*,
+                       const float*, float*, float*, int, hipStream_t);
 void launch_bnact_bwd_reduce(const void*, const void*, const void*,
                              const float*, const float*, const float*,
                              const float*, float*, float*, long, int, int,
@@ -87,7 +90,7 @@ hipStream_t cur_stream() { return c10::hip::getCurrentHIPStream().stream(); }
 // 2 launches instead of ~40, full chip occupancy, no atomics.  Task tables
 // travel as kernel arguments by value, so the path is hipGraph-capture-safe
 // (no host staging buffers whose contents could change between replays).
-constexpr int WG_MAX_TASKS = 14;
+constexpr int WG_MAX_TASKS = 24;
 
 struct WgradTask {
   const void* X;
@@ -351,7 +354,18 @@ std::vector<Tensor> conv_bn_act_bwd(
     Tensor gamma, Tensor beta, Tensor save_mean, Tensor save_invstd,
     int64_t stride, int64_t pad, bool act, bool need_dx, bool has_res,
     c10::optional<Tensor> dw_out, c10::optional<Tensor> dgamma_out,
-    c10::optional<Tensor> dbeta_out, c10::optional<Tensor> dx_accum) {
+    c10::optional<Tensor> dbeta_out, c10::optional<Tensor> dx_accum,
+    c10::optional<std::vector<Tensor>> fuse_up, int64_t up_mask_mode,
+    bool sums_ready, c10::optional<Tensor> sum_dz_in,
+    c10::optional<Tensor> sum_dzx_in) {
+  // fuse_up = [x_up, y_up, smean_up, sinvstd_up, gamma_up, beta_up,
+  //            sum_dz_up, sum_dzx_up]: while producing dx (= the UPSTREAM
+  //            conv's dy), also complete that conv's BN-backward channel
+  //            sums — fused into the slab-sum cast when dgrad is split-K,
+  //            else run as the standalone reduce right after dgrad.  The
+  //            upstream conv's own backward then passes sums_ready=true
+  //            (with sum_dz_in/sum_dzx_in in non-direct-grad mode) and
+  //            skips its reduce pass.
   // Direct-grad mode: when dw_out/dgamma_out/dbeta_out are given they are
   // PRE-ZEROED flat .grad views — the kernels accumulate straight into
   // them, skipping autograd's per-parameter accumulate pass.
@@ -368,14 +382,21 @@ std::vector<Tensor> conv_bn_act_bwd(
   // ReLU-mask source: residual epilogue needs y; otherwise the mask is
   // recovered from convout via BN algebra and y is never read
   int mask_mode = act ? (has_res ? 1 : 2) : 0;
-  Tensor sum_dz = direct ? *dbeta_out : at::zeros({K}, fopt);
-  Tensor sum_dzx = direct ? *dgamma_out : at::zeros({K}, fopt);
-  launch_bnact_bwd_reduce(dy.data_ptr(), y.data_ptr(), convout.data_ptr(),
-                          save_mean.data_ptr<float>(),
-                          save_invstd.data_ptr<float>(),
-                          gamma.data_ptr<float>(), beta.data_ptr<float>(),
-                          sum_dz.data_ptr<float>(), sum_dzx.data_ptr<float>(),
-                          (long)p.M, K, mask_mode, st);
+  Tensor sum_dz, sum_dzx;
+  if (sums_ready) {  // computed by the downstream conv's dx producer
+    sum_dz = direct ? *dbeta_out : *sum_dz_in;
+    sum_dzx = direct ? *dgamma_out : *sum_dzx_in;
+  } else {
+    sum_dz = direct ? *dbeta_out : at::zeros({K}, fopt);
+    sum_dzx = direct ? *dgamma_out : at::zeros({K}, fopt);
+    launch_bnact_bwd_reduce(dy.data_ptr(), y.data_ptr(), convout.data_ptr(),
+                            save_mean.data_ptr<float>(),
+                            save_invstd.data_ptr<float>(),
+                            gamma.data_ptr<float>(), beta.data_ptr<float>(),
+                            sum_dz.data_ptr<float>(),
+                            sum_dzx.data_ptr<float>(),
+                            (long)p.M, K, mask_mode, st);
+  }
 
   Tensor dconv = empty_cl_bf16(p.Nb, K, p.Ho, p.Wo, x);
   Tensor dres;
@@ -422,17 +443,41 @@ std::vector<Tensor> conv_bn_act_bwd(
     pd.M = p.Nb * p.H * p.W;
     pd.Kd = R * S * K;
     int splitk = conv_dgrad_splitk(p);
+    const bool fuse = fuse_up.has_value();
     if (splitk > 1) {
       int kchunk = ((pd.Kd + splitk - 1) / splitk + 31) / 32 * 32;
       splitk = (pd.Kd + kchunk - 1) / kchunk;
       Tensor wsd = at::empty({(long)splitk * pd.M * p.C}, fopt);
       launch_conv_dgrad(dconv.data_ptr(), w_rsck.data_ptr(), nullptr,
                         wsd.data_ptr<float>(), splitk, pd, 0, st);
-      launch_cast_f32_bf16(wsd.data_ptr<float>(), dx.data_ptr(),
-                           (long)pd.M * p.C, splitk, accum, st);
+      if (fuse) {
+        auto& f = *fuse_up;
+        launch_cast_bnact(wsd.data_ptr<float>(), dx.data_ptr(), (long)pd.M,
+                          p.C, splitk, accum, f[0].data_ptr(),
+                          f[1].defined() ? f[1].data_ptr() : f[0].data_ptr(),
+                          f[2].data_ptr<float>(), f[3].data_ptr<float>(),
+                          f[4].data_ptr<float>(), f[5].data_ptr<float>(),
+                          f[6].data_ptr<float>(), f[7].data_ptr<float>(),
+                          (int)up_mask_mode, st);
+      } else {
+        launch_cast_f32_bf16(wsd.data_ptr<float>(), dx.data_ptr(),
+                             (long)pd.M * p.C, splitk, accum, st);
+      }
     } else {
       launch_conv_dgrad(dconv.data_ptr(), w_rsck.data_ptr(), dx.data_ptr(),
                         nullptr, 1, pd, accum, st);
+      if (fuse) {
+        auto& f = *fuse_up;
+        launch_bnact_bwd_reduce(dx.data_ptr(), f[1].defined()
+                                    ? f[1].data_ptr() : f[0].data_ptr(),
+                                f[0].data_ptr(), f[2].data_ptr<float>(),
+                                f[3].data_ptr<float>(),
+                                f[4].data_ptr<float>(),
+                                f[5].data_ptr<float>(),
+                                f[6].data_ptr<float>(),
+                                f[7].data_ptr<float>(), (long)pd.M, p.C,
+                                (int)up_mask_mode, st);
+      }
     }
   }
   // dgamma = Σ dz·xhat, dbeta = Σ dz

@@ -434,7 +434,7 @@ __global__ __launch_bounds__(256) void k_wgrad_reduce(
 // (mchunk ≈ 512 rows) and the grid still fills 256 CUs without atomics.
 // Task tables travel as kernel arguments by value — no staging buffers,
 // hipGraph-capture-safe.
-#define WG_MAX_TASKS 14
+#define WG_MAX_TASKS 24
 
 struct WgradTask {
   const bf16* X;

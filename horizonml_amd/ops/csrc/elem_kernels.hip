@@ -112,6 +112,59 @@ __global__ __launch_bounds__(256) void k_cast_f32_bf16(
   }
 }
 
+// Fused split-dgrad slab-sum + upstream BN-backward reduce: while summing
+// the dgrad slabs into dy (bf16), also accumulate the UPSTREAM conv's
+// per-channel Σdz and Σ(dz·xhat) — its k_bnact_bwd_reduce launch is then
+// skipped entirely, and the sums use the unrounded f32 dy (closer to the
+// fp32 reference than the two-kernel version).  accum: dy accumulates into
+// dst (residual junction).  grid: (cdiv(C,64), msplit) like the reduce.
+__global__ __launch_bounds__(256) void k_cast_bnact(
+    const float* __restrict__ src, bf16* __restrict__ dst, long M, int C,
+    int nsplit, int accum, const bf16* __restrict__ x_up,
+    const bf16* __restrict__ y_up, const float* __restrict__ save_mean,
+    const float* __restrict__ save_invstd, const float* __restrict__ gamma,
+    const float* __restrict__ beta, float* __restrict__ sum_dz,
+    float* __restrict__ sum_dzx, int mask_mode, long mchunk) {
+  __shared__ float sdz[4][64];
+  __shared__ float sdzx[4][64];
+  const long n = M * (long)C;
+  const int c = blockIdx.x * 64 + (threadIdx.x & 63);
+  const int mlane = threadIdx.x >> 6;
+  const long mbeg = (long)blockIdx.y * mchunk;
+  const long mend = min((long)M, mbeg + mchunk);
+  float a_dz = 0.f, a_dzx = 0.f;
+  if (c < C) {
+    const float mean = save_mean[c], invstd = save_invstd[c];
+    const float ga = gamma[c] * invstd;
+    const float gb = beta[c] - mean * ga;
+    for (long m = mbeg + mlane; m < mend; m += 4) {
+      long i = m * C + c;
+      float v = src[i];
+      for (int z = 1; z < nsplit; z++) v += src[z * n + i];
+      if (accum) v += b2f(dst[i]);
+      dst[i] = f2b(v);
+      float g = v;
+      float xv = b2f(x_up[i]);
+      if (mask_mode == 1) {
+        if (b2f(y_up[i]) <= 0.f) g = 0.f;
+      } else if (mask_mode == 2) {
+        if (fmaf(ga, xv, gb) <= 0.f) g = 0.f;
+      }
+      a_dz += g;
+      a_dzx += g * (xv - mean) * invstd;
+    }
+  }
+  sdz[mlane][threadIdx.x & 63] = a_dz;
+  sdzx[mlane][threadIdx.x & 63] = a_dzx;
+  __syncthreads();
+  if (mlane == 0 && c < C) {
+    atomicAdd(&sum_dz[c], sdz[0][threadIdx.x] + sdz[1][threadIdx.x] +
+                              sdz[2][threadIdx.x] + sdz[3][threadIdx.x]);
+    atomicAdd(&sum_dzx[c], sdzx[0][threadIdx.x] + sdzx[1][threadIdx.x] +
+                               sdzx[2][threadIdx.x] + sdzx[3][threadIdx.x]);
+  }
+}
+
 // ------------------------------------------------------ BN+act backward ----
 // Pass 1: per-channel Σdz and Σ(dz·xhat) where dz = dy·relu'(y).
 // grid: (cdiv(C,64), msplit); block 256 = 4 m-lanes × 64 channels.
@@ -603,6 +656,23 @@ void launch_stats_reduce(const float* ws, float* stats, long M, int C,
 void launch_cast_f32_bf16(const float* src, void* dst, long n, int nsplit,
                           int accum, hipStream_t st) {
   k_cast_f32_bf16<<<gsz(n), 256, 0, st>>>(src, (bf16*)dst, n, nsplit, accum);
+}
+
+void launch_cast_bnact(const float* src, void* dst, long M, int C,
+                       int nsplit, int accum, const void* x_up,
+                       const void* y_up, const float* smean,
+                       const float* sinvstd, const float* gamma,
+                       const float* beta, float* sum_dz, float* sum_dzx,
+                       int mask_mode, hipStream_t st) {
+  int cblocks = (C + 63) / 64;
+  int msplit = (int)min((long)256, max((long)1, (long)(768 / cblocks)));
+  long mchunk = (M + msplit - 1) / msplit;
+  msplit = (int)((M + mchunk - 1) / mchunk);
+  dim3 grid(cblocks, msplit);
+  k_cast_bnact<<<grid, 256, 0, st>>>(src, (bf16*)dst, M, C, nsplit, accum,
+                                     (const bf16*)x_up, (const bf16*)y_up,
+                                     smean, sinvstd, gamma, beta, sum_dz,
+                                     sum_dzx, mask_mode, mchunk);
 }
 
 void launch_bnact_bwd_reduce(const void* dy, const void* yout, const void* x,
