@@ -541,9 +541,22 @@ extern "C" int conv_fwd_splitk(ConvP p) {
 }
 
 extern "C" int conv_dgrad_splitk(ConvP p) {
-  // dgrad geometry: M=Nb*H*W rows, Kd=R*S*K, N=C
+  // dgrad geometry: M=Nb*H*W rows, Kd=R*S*K, N=C.  dgrad has its own Kd
+  // threshold: its split path costs only the cast pass (often fused with
+  // the upstream conv's BN reduce), so splitting pays at smaller Kd than
+  // the forward's split (which drags a stats pass + slab re-reads).
+  static int kd_min_dg = [] {
+    const char* e = getenv("HZ_SK_KD_DG");
+    return e ? atoi(e) : 512;
+  }();
   int tiles = cdiv_h(p.C, 64) * cdiv_h(p.Nb * p.H * p.W, 64);
-  return pick_splitk(tiles, p.R * p.S * p.K);
+  int Kd = p.R * p.S * p.K;
+  if (tiles >= 96 || Kd < kd_min_dg) return 1;
+  int sk = cdiv_h(256, tiles);
+  int maxsk = cdiv_h(Kd, 32);
+  if (sk > maxsk) sk = maxsk;
+  if (sk > 32) sk = 32;
+  return sk;
 }
 
 // Non-split forward (bf16 out + fused stats).
