@@ -48,3 +48,14 @@ def test_tp_engine_gpu(tmp_path):
     df = run_tensor_parallel(1, 1, 128, str(tmp_path / "tp"), batch_size=32,
                              synthetic=True)
     _check(df, 1, bandwidth=True)
+
+
+@pytest.mark.timeout(300)
+def test_pp_engine_microbatches_gpu(tmp_path):
+    """Pipeline with microbatching (bubble reduction) through the fused
+    blocks — exercises gradient accumulation across microbatch backwards
+    in non-direct-grad mode on the gfx950 kernels."""
+    from layer_model_parallel_train import run_model_parallel
+    df = run_model_parallel(1, 1, 128, str(tmp_path / "ppm"), batch_size=64,
+                            synthetic=True, microbatches=4)
+    _check(df, 1, bandwidth=True)

@@ -115,3 +115,22 @@ def test_build_model_names():
         assert build_model(name, 10) is not None
     with pytest.raises(ValueError):
         build_model("nope")
+
+
+def test_partition_resnet50_8_stages():
+    """BASELINE config #3/#5 shapes: resnet50 split into 8 pipeline stages
+    must preserve the forward exactly."""
+    import torch
+    from horizonml_amd.models import build_model, partition_model
+    torch.manual_seed(0)
+    model = build_model("resnet50", num_classes=10)
+    model.eval()
+    segs = partition_model(model, 8)
+    assert len(segs) == 8
+    x = torch.randn(2, 3, 32, 32)
+    with torch.no_grad():
+        ref = model(x)
+        h = x
+        for s in segs:
+            h = s(h)
+    assert torch.allclose(h, ref, atol=1e-5), (h - ref).abs().max()
