@@ -113,11 +113,38 @@ def build_dataset(data_dir: str = "./data", synthetic: Optional[bool] = None,
                             num_classes=num_classes)
 
 
+class RawView(Dataset):
+    """uint8 view of a dataset: normalization deferred to the consumer
+    (GPU engines normalize on-device — 4x fewer host-copy bytes and no CPU
+    elementwise work in the hot loop)."""
+
+    def __init__(self, base):
+        self.base = base
+
+    def __len__(self):
+        return len(self.base)
+
+    def __getitem__(self, i: int):
+        return self.base.images[i], self.base.labels[i]
+
+    def __getitems__(self, idxs):
+        idx = torch.as_tensor(idxs)
+        return list(zip(self.base.images[idx].unbind(0),
+                        self.base.labels[idx].unbind(0)))
+
+
+def normalize_uint8(x: torch.Tensor) -> torch.Tensor:
+    """The reference transform ((0.5,0.5,0.5) mean/std) for raw uint8
+    batches, applied on whatever device x lives on."""
+    return x.to(torch.float32).div_(255.0).sub_(_MEAN).div_(_STD)
+
+
 def get_dataloader(rank: int, world_size: int, batch_size: int = 64,
                    sample_size: int = 1000, strategy: str = "dp",
                    data_dir: str = "./data", synthetic: Optional[bool] = None,
                    seed: int = DEFAULT_SEED, drop_last: bool = False,
-                   image_size: int = 32, num_classes: int = 10
+                   image_size: int = 32, num_classes: int = 10,
+                   raw: bool = False
                    ) -> Tuple[DataLoader, Optional[DistributedSampler]]:
     """Reference-parity dataloader.
 
@@ -135,6 +162,8 @@ def get_dataloader(rank: int, world_size: int, batch_size: int = 64,
                        n=max(50000, sample_size), seed=seed,
                        image_size=image_size, num_classes=num_classes)
     idx = shared_subset_indices(len(ds), sample_size, seed=seed)
+    if raw:
+        ds = RawView(ds)
     subset = Subset(ds, idx.tolist())
     if strategy in ("dp", "hybrid"):
         sampler = DistributedSampler(subset, num_replicas=world_size,

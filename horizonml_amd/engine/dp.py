@@ -39,6 +39,9 @@ from ..utils.seed import seed_everything
 def _to_device(x, y, ctx: DistContext):
     if ctx.is_gpu:
         x = x.to(ctx.device, non_blocking=True)
+        if x.dtype == torch.uint8:  # raw loader: normalize on-device
+            from ..data.cifar import normalize_uint8
+            x = normalize_uint8(x)
         x = x.to(memory_format=torch.channels_last).to(torch.bfloat16)
         y = y.to(ctx.device, non_blocking=True)
     return x, y
@@ -54,7 +57,7 @@ def train_dp(ctx: DistContext, epochs: int, sample_size: int, logs_dir: str,
     seed_everything(rank=rank)
     loader, sampler = get_dataloader(rank, world, batch_size, sample_size,
                                      strategy="dp", data_dir=data_dir,
-                                     synthetic=synthetic)
+                                     synthetic=synthetic, raw=ctx.is_gpu)
 
     model = build_model(model_name, num_classes=10)
     if ctx.is_gpu:

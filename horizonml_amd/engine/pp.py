@@ -47,13 +47,13 @@ def train_pp(ctx: DistContext, epochs: int, sample_size: int, logs_dir: str,
                                    sample_size, strategy="hybrid",
                                    data_dir=data_dir, synthetic=synthetic,
                                    image_size=image_size,
-                                   num_classes=num_classes)
+                                   num_classes=num_classes, raw=ctx.is_gpu)
     else:
         loader, _ = get_dataloader(rank, world, batch_size, sample_size,
                                    strategy="mp", data_dir=data_dir,
                                    synthetic=synthetic,
                                    image_size=image_size,
-                                   num_classes=num_classes)
+                                   num_classes=num_classes, raw=ctx.is_gpu)
 
     model = build_model(model_name, num_classes=num_classes)
     segments = partition_model(model, n_stages)
@@ -109,8 +109,11 @@ def train_pp(ctx: DistContext, epochs: int, sample_size: int, logs_dir: str,
             cpu_samples.append(cpu)
             mem_samples.append(mem)
             if ctx.is_gpu:
-                x = x.to(ctx.device, non_blocking=True).to(
-                    memory_format=torch.channels_last).to(torch.bfloat16)
+                x = x.to(ctx.device, non_blocking=True)
+                if x.dtype == torch.uint8:
+                    from ..data.cifar import normalize_uint8
+                    x = normalize_uint8(x)
+                x = x.to(memory_format=torch.channels_last).to(torch.bfloat16)
                 y = y.to(ctx.device, non_blocking=True)
             if optimizer is not None:
                 optimizer.zero_grad(set_to_none=False)

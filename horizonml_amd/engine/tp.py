@@ -38,7 +38,7 @@ def train_tp(ctx: DistContext, epochs: int, sample_size: int, logs_dir: str,
     torch.manual_seed(1234 + rank)  # shard params differ per rank by design
     loader, _ = get_dataloader(rank, world, batch_size, sample_size,
                                strategy="tp", data_dir=data_dir,
-                               synthetic=synthetic)
+                               synthetic=synthetic, raw=ctx.is_gpu)
 
     model = build_tp_resnet18(world, rank, num_classes=10, mode=tp_mode)
     if ctx.is_gpu:
@@ -70,8 +70,11 @@ def train_tp(ctx: DistContext, epochs: int, sample_size: int, logs_dir: str,
             cpu_samples.append(cpu)
             mem_samples.append(mem)
             if ctx.is_gpu:
-                x = x.to(ctx.device, non_blocking=True).to(
-                    memory_format=torch.channels_last).to(torch.bfloat16)
+                x = x.to(ctx.device, non_blocking=True)
+                if x.dtype == torch.uint8:
+                    from ..data.cifar import normalize_uint8
+                    x = normalize_uint8(x)
+                x = x.to(memory_format=torch.channels_last).to(torch.bfloat16)
                 y = y.to(ctx.device, non_blocking=True)
             with prof.compute():
                 optimizer.zero_grad(set_to_none=False)

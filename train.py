@@ -37,7 +37,7 @@ def main():
     try:
         loader, sampler = get_dataloader(ctx.rank, ctx.world_size,
                                          batch_size, sample_size,
-                                         strategy="dp",
+                                         strategy="dp", raw=ctx.is_gpu,
                                          synthetic=os.environ.get(
                                              "SYNTHETIC") == "1" or None)
         name = "resnet18" if model_type.startswith("resnet") else "mobilenet_v2"
@@ -55,7 +55,11 @@ def main():
             loss_sum, correct, count = 0.0, 0, 0
             for x, y in loader:
                 if ctx.is_gpu:
-                    x = x.to(ctx.device).to(
+                    x = x.to(ctx.device)
+                    if x.dtype == torch.uint8:
+                        from horizonml_amd.data.cifar import normalize_uint8
+                        x = normalize_uint8(x)
+                    x = x.to(
                         memory_format=torch.channels_last).to(torch.bfloat16)
                     y = y.to(ctx.device)
                 opt.zero_grad(set_to_none=False)
