@@ -66,6 +66,7 @@ def main():
     world = int(os.environ.get("WORLD_SIZE", 1))
     local_rank = int(os.environ.get("LOCAL_RANK", 0))
     assert torch.cuda.is_available(), "bench.py needs an MI355X"
+    torch.set_num_threads(8)  # tiny host-side work; avoid pool wake-storms
     torch.cuda.set_device(local_rank)
     dev = torch.device("cuda", local_rank)
     # HZ_FORCE_COMM=1: exercise the full RCCL comm path (all-reduce inside
