@@ -42,9 +42,15 @@ def main():
     ap.add_argument("--synthetic", action="store_true", default=None,
                     help="force synthetic CIFAR-shaped data")
     ap.add_argument("--lr", type=float, default=1e-3)
+    ap.add_argument("--deterministic", action="store_true",
+                    help="torch.use_deterministic_algorithms (GPU kernels "
+                         "remain reproducible only to bf16/atomic rounding)")
     ap.add_argument("--optimizer", type=str, default="adam",
                     choices=["adam", "sgd"])
     args = ap.parse_args()
+    if args.deterministic:
+        import os
+        os.environ["HZ_DETERMINISTIC"] = "1"
     df = run_data_parallel(args.world_size, args.epochs, args.sample_size,
                            args.logs_dir, args.batch_size, args.model,
                            args.backend, args.synthetic, args.lr,

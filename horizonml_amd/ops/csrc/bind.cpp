@@ -429,6 +429,9 @@ std::vector<Tensor> conv_bn_act_bwd(
                  p, st);
   }
 
+  TORCH_CHECK(need_dx || !fuse_up.has_value(),
+              "fuse_up requires need_dx (the fused reduce rides the dx "
+              "production)");
   Tensor dx;
   if (need_dx) {
     TORCH_CHECK(w_rsck.is_contiguous() &&

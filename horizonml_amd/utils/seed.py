@@ -26,6 +26,9 @@ def seed_everything(seed: int = DEFAULT_SEED, rank: int = 0) -> None:
     torch.manual_seed(s)
     if torch.cuda.is_available():
         torch.cuda.manual_seed_all(s)
+    import os
+    if os.environ.get("HZ_DETERMINISTIC") == "1":
+        enable_deterministic()
 
 
 def shared_subset_indices(dataset_len: int, sample_size: int,
@@ -40,3 +43,18 @@ def shared_subset_indices(dataset_len: int, sample_size: int,
     g = torch.Generator().manual_seed(int(seed))
     n = min(int(sample_size), int(dataset_len))
     return torch.randperm(dataset_len, generator=g)[:n]
+
+
+def enable_deterministic(warn_only: bool = True) -> None:
+    """Best-effort deterministic mode (SURVEY.md §5.2 suggested this as the
+    rebuild's answer to the reference's absent seed discipline).
+
+    Caveat, documented honestly: the gfx950 kernels accumulate BatchNorm
+    batch statistics and channel sums with f32 atomics, whose ordering
+    varies run to run — GPU training is reproducible only to bf16 rounding
+    (the measured effect is ~1e-7 per value, amplified by depth; see
+    tests/test_ops_gpu.py).  CPU runs are fully deterministic.
+    """
+    import os
+    os.environ.setdefault("CUBLAS_WORKSPACE_CONFIG", ":4096:8")
+    torch.use_deterministic_algorithms(True, warn_only=warn_only)

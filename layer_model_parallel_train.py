@@ -43,11 +43,17 @@ def main():
                     choices=[None, "nccl", "gloo"])
     ap.add_argument("--synthetic", action="store_true", default=None)
     ap.add_argument("--lr", type=float, default=1e-3)
+    ap.add_argument("--deterministic", action="store_true",
+                    help="torch.use_deterministic_algorithms (GPU kernels "
+                         "remain reproducible only to bf16/atomic rounding)")
     ap.add_argument("--optimizer", type=str, default="adam",
                     choices=["adam", "sgd"])
     ap.add_argument("--microbatches", type=int, default=1,
                     help="pipeline microbatches per step (bubble reduction)")
     args = ap.parse_args()
+    if args.deterministic:
+        import os
+        os.environ["HZ_DETERMINISTIC"] = "1"
     df = run_model_parallel(args.world_size, args.epochs, args.sample_size,
                             args.logs_dir, args.batch_size, args.model,
                             args.backend, args.synthetic, args.lr,

@@ -41,6 +41,9 @@ def main():
                     choices=[None, "nccl", "gloo"])
     ap.add_argument("--synthetic", action="store_true", default=None)
     ap.add_argument("--lr", type=float, default=1e-3)
+    ap.add_argument("--deterministic", action="store_true",
+                    help="torch.use_deterministic_algorithms (GPU kernels "
+                         "remain reproducible only to bf16/atomic rounding)")
     ap.add_argument("--optimizer", type=str, default="adam",
                     choices=["adam", "sgd"])
     ap.add_argument("--tp_mode", type=str, default="fc",
@@ -48,6 +51,9 @@ def main():
                     help="fc: reference-parity classifier shard; "
                          "full: sharded conv2d + all-gather")
     args = ap.parse_args()
+    if args.deterministic:
+        import os
+        os.environ["HZ_DETERMINISTIC"] = "1"
     df = run_tensor_parallel(args.world_size, args.epochs, args.sample_size,
                              args.logs_dir, args.batch_size, args.backend,
                              args.synthetic, args.lr, args.optimizer,
