@@ -11,13 +11,14 @@
 //
 // Latency design (these conv shapes are tiny — batch 64 CIFAR gives
 // M=64..16k, so kernels are latency-bound, not throughput-bound):
-//  * register-staged prefetch: tile t+1's global loads are issued right
-//    after the barrier so their latency hides under tile t's ds_read+MFMA
-//    (guide §5.5 T14 minimum form);
-//  * split-K: when the tile grid would underfill the 256 CUs, the K
-//    reduction is partitioned over blockIdx.z and f32 partials are
-//    atomically combined in an [M][N] f32 workspace; BN statistics +
-//    bf16 conversion then happen in the bn_apply pass (bind.cpp decides).
+//  * double-buffered LDS with ONE barrier per K-step and 2-deep register
+//    prefetch: tile k+2's loads are in flight while tile k computes, so
+//    the ~900-cycle HBM latency hides even at 1 block/CU;
+//  * split-K: when a long K loop would serialize on an underfilled grid
+//    (thresholds measured per direction — see pick_splitk /
+//    conv_dgrad_splitk), the K reduction is partitioned over blockIdx.z
+//    into per-split f32 slabs summed by the consumer pass (stats_reduce /
+//    bn_apply forward, cast[+fused BN reduce] backward) — no atomics.
 //
 // The non-split forward epilogue also accumulates per-channel BN batch
 // statistics (Σy, Σy²) via wave shuffle reduction + one atomicAdd per
