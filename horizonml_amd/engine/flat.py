@@ -21,7 +21,7 @@ import torch
 import torch.nn as nn
 
 from .. import ops as _ops
-from ..models.layers import ConvBNAct, Linear
+from ..models.layers import ConvBNAct, DepthwiseConvBNAct, Linear
 
 
 class FlatParamManager:
@@ -52,7 +52,7 @@ class FlatParamManager:
         # wire bf16 shadow views into the modules; collect conv RSCK metadata
         convs = []
         for m in model.modules():
-            if isinstance(m, (ConvBNAct, Linear)):
+            if isinstance(m, (ConvBNAct, Linear, DepthwiseConvBNAct)):
                 woff, wn = self.slices[id(m.weight)]
                 m.weight_bf16 = self.shadow[woff:woff + wn].view(
                     m.weight.shape)

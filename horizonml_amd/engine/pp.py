@@ -128,6 +128,9 @@ def train_pp(ctx: DistContext, epochs: int, sample_size: int, logs_dir: str,
             if optimizer is not None:
                 with prof.compute():
                     optimizer.step()
+                    if ctx.is_gpu:
+                        from ..models import refresh_all_shadows
+                        refresh_all_shadows(seg)
             if probe is not None:
                 probe.step()
             if stage.is_last and total_loss is not None:

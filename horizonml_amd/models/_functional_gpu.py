@@ -84,6 +84,42 @@ def conv_bn_act(x, mod, residual=None):
                              residual, mod)
 
 
+class DwConvBNActFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bn_weight, bn_bias, residual, mod):
+        x = _cl(x)
+        w_bf16 = mod._shadow()
+        if residual is not None:
+            residual = _cl(residual)
+        y, convout, smean, sinvstd = _C().dw_conv_bn_fwd(
+            x, w_bf16, bn_weight, bn_bias, mod.running_mean,
+            mod.running_var, mod.stride, mod.padding, mod.momentum, mod.eps,
+            mod.training, mod.act, residual)
+        ctx.save_for_backward(x, y, convout, w_bf16, bn_weight, bn_bias,
+                              smean, sinvstd)
+        ctx.mod = mod
+        ctx.has_res = residual is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, y, convout, w_bf16, gamma, beta, smean, sinvstd = \
+            ctx.saved_tensors
+        mod = ctx.mod
+        need_dx = ctx.needs_input_grad[0]
+        dy = _cl(dy.to(torch.bfloat16) if dy.dtype != torch.bfloat16 else dy)
+        dx, dw, dgamma, dbeta, dres = _C().dw_conv_bn_bwd(
+            dy, y, x, w_bf16, convout, gamma, beta, smean, sinvstd,
+            mod.stride, mod.padding, mod.act, need_dx, ctx.has_res)
+        return (dx if need_dx else None, dw, dgamma, dbeta,
+                dres if ctx.has_res else None, None)
+
+
+def dw_conv_bn_act(x, mod, residual=None):
+    return DwConvBNActFn.apply(x, mod.weight, mod.bn_weight, mod.bn_bias,
+                               residual, mod)
+
+
 class MaxPoolFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x):
@@ -269,7 +305,8 @@ class ResBlockFn(torch.autograd.Function):
                 else:
                     sums = (torch.zeros_like(m_up.bn_bias),
                             torch.zeros_like(m_up.bn_weight))
-                up_mask = 2 if m_up.act else 0
+                up_mask = (2 if m_up.act == 1
+                           else 4 if m_up.act == 2 else 0)
                 fuse_up = [convout_u, y_u, smean_u, sinvstd_u, gamma_u,
                            beta_u, sums[0], sums[1]]
             sums_ready = next_sums is not None

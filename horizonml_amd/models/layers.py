@@ -41,15 +41,18 @@ class ConvBNAct(nn.Module):
     (fused into the conv epilogue on GPU).
     """
 
+    ACT = {False: 0, "none": 0, True: 1, "relu": 1, "relu6": 2}
+
     def __init__(self, in_ch: int, out_ch: int, kernel_size: int = 3,
                  stride: int = 1, padding: Optional[int] = None,
-                 act: bool = True, eps: float = 1e-5, momentum: float = 0.1):
+                 act=True, eps: float = 1e-5, momentum: float = 0.1):
         super().__init__()
         if padding is None:
             padding = kernel_size // 2
         self.in_ch, self.out_ch = in_ch, out_ch
         self.kernel_size, self.stride, self.padding = kernel_size, stride, padding
-        self.act = act
+        # activation kind: 0 none, 1 ReLU, 2 ReLU6 (mobilenet)
+        self.act = self.ACT[act] if not isinstance(act, int) else act
         self.eps, self.momentum = eps, momentum
         # KRSC layout (out, kh, kw, in)
         self.weight = nn.Parameter(
@@ -107,8 +110,79 @@ class ConvBNAct(nn.Module):
                          self.training, self.momentum, self.eps)
         if residual is not None:
             y = y + residual
-        if self.act:
+        if self.act == 1:
             y = F.relu(y)
+        elif self.act == 2:
+            y = F.hardtanh(y, 0.0, 6.0)  # ReLU6
+        return y
+
+
+class DepthwiseConvBNAct(nn.Module):
+    """Fused depthwise Conv2d (groups=C) + BatchNorm2d [+ act] — the
+    MobileNet inverted-residual middle op, on the c-blocked gfx950 stencil
+    kernels (no MFMA: depthwise is bandwidth-bound).  Weight layout
+    [R, S, C] (channels innermost, matching NHWC activations)."""
+
+    def __init__(self, ch: int, kernel_size: int = 3, stride: int = 1,
+                 padding: Optional[int] = None, act="relu6",
+                 eps: float = 1e-5, momentum: float = 0.1):
+        super().__init__()
+        if padding is None:
+            padding = kernel_size // 2
+        self.ch = ch
+        self.kernel_size, self.stride, self.padding = (kernel_size, stride,
+                                                       padding)
+        self.act = ConvBNAct.ACT[act] if not isinstance(act, int) else act
+        self.eps, self.momentum = eps, momentum
+        self.weight = nn.Parameter(
+            torch.empty(kernel_size, kernel_size, ch))
+        self.bn_weight = nn.Parameter(torch.ones(ch))
+        self.bn_bias = nn.Parameter(torch.zeros(ch))
+        self.register_buffer("running_mean", torch.zeros(ch))
+        self.register_buffer("running_var", torch.ones(ch))
+        self.register_buffer("weight_bf16", torch.empty(0), persistent=False)
+        with torch.no_grad():
+            fan = kernel_size * kernel_size
+            self.weight.normal_(0, math.sqrt(2.0 / fan))
+
+    def _shadow(self) -> torch.Tensor:
+        if getattr(self, "_managed", False):
+            return self.weight_bf16
+        if (self.weight_bf16.numel() != self.weight.numel()
+                or self.weight_bf16.device != self.weight.device):
+            with torch.no_grad():
+                self.weight_bf16 = self.weight.detach() \
+                    .to(torch.bfloat16).contiguous()
+        return self.weight_bf16
+
+    def refresh_shadow(self):
+        if not getattr(self, "_managed", False):
+            with torch.no_grad():
+                self.weight_bf16 = self.weight.detach() \
+                    .to(torch.bfloat16).contiguous()
+
+    def extra_repr(self):
+        return (f"{self.ch}, k={self.kernel_size}, s={self.stride}, "
+                f"act={self.act}")
+
+    def forward(self, x: torch.Tensor,
+                residual: Optional[torch.Tensor] = None) -> torch.Tensor:
+        if _use_hip(x):
+            from . import _functional_gpu as FG
+            return FG.dw_conv_bn_act(x, self, residual)
+        # CPU reference path: [R,S,C] -> (C,1,R,S) grouped conv
+        w = self.weight.permute(2, 0, 1).unsqueeze(1)
+        y = F.conv2d(x, w, None, stride=self.stride, padding=self.padding,
+                     groups=self.ch)
+        y = F.batch_norm(y, self.running_mean, self.running_var,
+                         self.bn_weight, self.bn_bias, self.training,
+                         self.momentum, self.eps)
+        if residual is not None:
+            y = y + residual
+        if self.act == 1:
+            y = F.relu(y)
+        elif self.act == 2:
+            y = F.hardtanh(y, 0.0, 6.0)
         return y
 
 
@@ -176,5 +250,5 @@ class Linear(nn.Module):
 def refresh_all_shadows(model: nn.Module):
     """Refresh every bf16 weight shadow in `model` (call after param updates)."""
     for m in model.modules():
-        if isinstance(m, (ConvBNAct, Linear)):
+        if isinstance(m, (ConvBNAct, Linear, DepthwiseConvBNAct)):
             m.refresh_shadow()

@@ -51,6 +51,12 @@ void launch_bn_bwd_apply(const void*, const void*, const void*, const float*,
                          const float*, const float*, const float*,
                          const float*, const float*, void*, void*, long, int,
                          int, hipStream_t);
+void launch_dw_fwd(const void*, const void*, void*, float*, int, int, int,
+                   int, int, int, int, int, int, int, hipStream_t);
+void launch_dw_dgrad(const void*, const void*, void*, int, int, int, int,
+                     int, int, int, int, int, int, hipStream_t);
+void launch_dw_wgrad(const void*, const void*, float*, int, int, int, int,
+                     int, int, int, int, int, int, hipStream_t);
 void launch_maxpool_fwd(const void*, void*, unsigned char*, int, int, int,
                         int, int, int, hipStream_t);
 void launch_maxpool_bwd(const void*, const unsigned char*, void*, int, int,
@@ -280,7 +286,7 @@ ConvP make_convp(const Tensor& x, int K, int R, int S, int stride, int pad) {
 std::vector<Tensor> conv_bn_act_fwd(
     Tensor x, Tensor w, Tensor gamma, Tensor beta, Tensor running_mean,
     Tensor running_var, int64_t stride, int64_t pad, double momentum,
-    double eps, bool training, bool act,
+    double eps, bool training, int64_t act,
     c10::optional<Tensor> residual, c10::optional<Tensor> stats_buf) {
   check_cl(x, "x");
   TORCH_CHECK(w.is_cuda() && w.scalar_type() == torch::kBFloat16 &&
@@ -333,7 +339,7 @@ std::vector<Tensor> conv_bn_act_fwd(
                         training ? smean.data_ptr<float>() : nullptr,
                         training ? sinvstd.data_ptr<float>() : nullptr,
                         (long)p.M, K, (float)momentum, (float)eps,
-                        training ? 1 : 0, act ? 1 : 0, splitk, st);
+                        training ? 1 : 0, (int)act, splitk, st);
   } else {
     launch_conv_fwd(x.data_ptr(), w.data_ptr(), convout.data_ptr(),
                     stats_ptr, p, st);
@@ -344,7 +350,7 @@ std::vector<Tensor> conv_bn_act_fwd(
                     training ? smean.data_ptr<float>() : nullptr,
                     training ? sinvstd.data_ptr<float>() : nullptr,
                     (long)p.M, K, (float)momentum, (float)eps,
-                    training ? 1 : 0, act ? 1 : 0, st);
+                    training ? 1 : 0, (int)act, st);
   }
   return {y, convout, smean, sinvstd};
 }
@@ -352,7 +358,7 @@ std::vector<Tensor> conv_bn_act_fwd(
 std::vector<Tensor> conv_bn_act_bwd(
     Tensor dy, Tensor y, Tensor x, Tensor w, Tensor w_rsck, Tensor convout,
     Tensor gamma, Tensor beta, Tensor save_mean, Tensor save_invstd,
-    int64_t stride, int64_t pad, bool act, bool need_dx, bool has_res,
+    int64_t stride, int64_t pad, int64_t act, bool need_dx, bool has_res,
     c10::optional<Tensor> dw_out, c10::optional<Tensor> dgamma_out,
     c10::optional<Tensor> dbeta_out, c10::optional<Tensor> dx_accum,
     c10::optional<std::vector<Tensor>> fuse_up, int64_t up_mask_mode,
@@ -381,7 +387,8 @@ std::vector<Tensor> conv_bn_act_bwd(
   bool direct = dw_out.has_value();
   // ReLU-mask source: residual epilogue needs y; otherwise the mask is
   // recovered from convout via BN algebra and y is never read
-  int mask_mode = act ? (has_res ? 1 : 2) : 0;
+  int mask_mode = act == 1 ? (has_res ? 1 : 2)
+                 : act == 2 ? (has_res ? 3 : 4) : 0;
   Tensor sum_dz, sum_dzx;
   if (sums_ready) {  // computed by the downstream conv's dx producer
     sum_dz = direct ? *dbeta_out : *sum_dz_in;
@@ -484,6 +491,103 @@ std::vector<Tensor> conv_bn_act_bwd(
     }
   }
   // dgamma = Σ dz·xhat, dbeta = Σ dz
+  return {dx, dw, sum_dzx, sum_dz, dres};
+}
+
+// --------------------------------------------------------- depthwise conv --
+std::vector<Tensor> dw_conv_bn_fwd(
+    Tensor x, Tensor w, Tensor gamma, Tensor beta, Tensor running_mean,
+    Tensor running_var, int64_t stride, int64_t pad, double momentum,
+    double eps, bool training, int64_t act, c10::optional<Tensor> residual) {
+  check_cl(x, "x");
+  TORCH_CHECK(w.is_cuda() && w.scalar_type() == torch::kBFloat16 &&
+                  w.is_contiguous() && w.dim() == 3,
+              "w must be contiguous bf16 [R,S,C]");
+  int Nb = (int)x.size(0), C = (int)x.size(1), H = (int)x.size(2),
+      W = (int)x.size(3);
+  int R = (int)w.size(0), S = (int)w.size(1);
+  TORCH_CHECK((int)w.size(2) == C, "depthwise channel mismatch");
+  int Ho = (H + 2 * (int)pad - R) / (int)stride + 1;
+  int Wo = (W + 2 * (int)pad - S) / (int)stride + 1;
+  long M = (long)Nb * Ho * Wo;
+  auto fopt = x.options().dtype(torch::kFloat32);
+  Tensor convout = empty_cl_bf16(Nb, C, Ho, Wo, x);
+  Tensor y = empty_cl_bf16(Nb, C, Ho, Wo, x);
+  Tensor stats, smean, sinvstd;
+  float* stats_ptr = nullptr;
+  if (training) {
+    stats = at::zeros({2L * C}, fopt);
+    smean = at::empty({C}, fopt);
+    sinvstd = at::empty({C}, fopt);
+    stats_ptr = stats.data_ptr<float>();
+  } else {
+    smean = at::empty({0}, fopt);
+    sinvstd = at::empty({0}, fopt);
+  }
+  const void* res_ptr = nullptr;
+  if (residual.has_value()) {
+    check_cl(*residual, "residual");
+    res_ptr = residual->data_ptr();
+  }
+  auto st = cur_stream();
+  launch_dw_fwd(x.data_ptr(), w.data_ptr(), convout.data_ptr(), stats_ptr,
+                Nb, H, W, C, Ho, Wo, R, S, (int)stride, (int)pad, st);
+  launch_bn_apply(convout.data_ptr(), res_ptr, y.data_ptr(), stats_ptr,
+                  gamma.data_ptr<float>(), beta.data_ptr<float>(),
+                  running_mean.data_ptr<float>(),
+                  running_var.data_ptr<float>(),
+                  training ? smean.data_ptr<float>() : nullptr,
+                  training ? sinvstd.data_ptr<float>() : nullptr, M, C,
+                  (float)momentum, (float)eps, training ? 1 : 0, (int)act,
+                  st);
+  return {y, convout, smean, sinvstd};
+}
+
+std::vector<Tensor> dw_conv_bn_bwd(
+    Tensor dy, Tensor y, Tensor x, Tensor w, Tensor convout, Tensor gamma,
+    Tensor beta, Tensor save_mean, Tensor save_invstd, int64_t stride,
+    int64_t pad, int64_t act, bool need_dx, bool has_res) {
+  check_cl(dy, "dy");
+  check_cl(x, "x");
+  int Nb = (int)x.size(0), C = (int)x.size(1), H = (int)x.size(2),
+      W = (int)x.size(3);
+  int R = (int)w.size(0), S = (int)w.size(1);
+  int Ho = (int)dy.size(2), Wo = (int)dy.size(3);
+  long M = (long)Nb * Ho * Wo;
+  auto fopt = x.options().dtype(torch::kFloat32);
+  auto st = cur_stream();
+  int mask_mode = act == 1 ? (has_res ? 1 : 2)
+                 : act == 2 ? (has_res ? 3 : 4) : 0;
+  Tensor sum_dz = at::zeros({C}, fopt);
+  Tensor sum_dzx = at::zeros({C}, fopt);
+  launch_bnact_bwd_reduce(dy.data_ptr(), y.data_ptr(), convout.data_ptr(),
+                          save_mean.data_ptr<float>(),
+                          save_invstd.data_ptr<float>(),
+                          gamma.data_ptr<float>(), beta.data_ptr<float>(),
+                          sum_dz.data_ptr<float>(), sum_dzx.data_ptr<float>(),
+                          M, C, mask_mode, st);
+  Tensor dconv = empty_cl_bf16(Nb, C, Ho, Wo, x);
+  Tensor dres;
+  void* dres_ptr = nullptr;
+  if (has_res) {
+    dres = empty_cl_bf16(Nb, C, Ho, Wo, x);
+    dres_ptr = dres.data_ptr();
+  }
+  launch_bn_bwd_apply(dy.data_ptr(), y.data_ptr(), convout.data_ptr(),
+                      save_mean.data_ptr<float>(),
+                      save_invstd.data_ptr<float>(), gamma.data_ptr<float>(),
+                      beta.data_ptr<float>(), sum_dz.data_ptr<float>(),
+                      sum_dzx.data_ptr<float>(), dconv.data_ptr(), dres_ptr,
+                      M, C, mask_mode, st);
+  Tensor dw = at::zeros({(int64_t)R, (int64_t)S, (int64_t)C}, fopt);
+  launch_dw_wgrad(x.data_ptr(), dconv.data_ptr(), dw.data_ptr<float>(), Nb,
+                  H, W, C, Ho, Wo, R, S, (int)stride, (int)pad, st);
+  Tensor dx;
+  if (need_dx) {
+    dx = empty_cl_bf16(Nb, C, H, W, x);
+    launch_dw_dgrad(dconv.data_ptr(), w.data_ptr(), dx.data_ptr(), Nb, H, W,
+                    C, Ho, Wo, R, S, (int)stride, (int)pad, st);
+  }
   return {dx, dw, sum_dzx, sum_dz, dres};
 }
 
@@ -677,6 +781,8 @@ void grad_divergence(Tensor g, Tensor prev, Tensor sumsq, Tensor out,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv_bn_act_fwd", &conv_bn_act_fwd);
   m.def("conv_bn_act_bwd", &conv_bn_act_bwd);
+  m.def("dw_conv_bn_fwd", &dw_conv_bn_fwd);
+  m.def("dw_conv_bn_bwd", &dw_conv_bn_bwd);
   m.def("maxpool_fwd", &maxpool_fwd);
   m.def("maxpool_bwd", &maxpool_bwd);
   m.def("avgpool_fwd", &avgpool_fwd);

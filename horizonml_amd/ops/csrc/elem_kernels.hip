@@ -76,7 +76,8 @@ __global__ __launch_bounds__(256) void k_bn_apply(
       if (F32SRC && convout != nullptr) convout[i] = f2b(xi);
       float v = (xi - mean) * invstd * gamma[c] + beta[c];
       if (res != nullptr) v += b2f(res[i]);
-      if (act) v = fmaxf(v, 0.f);
+      if (act == 1) v = fmaxf(v, 0.f);
+      else if (act == 2) v = fminf(fmaxf(v, 0.f), 6.f);
       y[i] = f2b(v);
     }
     return;
@@ -131,7 +132,8 @@ __global__ __launch_bounds__(256) void k_bn_apply(
     for (int e = 0; e < 8; e++) {
       float v = (xi.f[e] - mean8.f[e]) * istd8.f[e] * g8.f[e] + b8.f[e];
       if (res != nullptr) v += b2f(r8.e[e]);
-      if (act) v = fmaxf(v, 0.f);
+      if (act == 1) v = fmaxf(v, 0.f);
+      else if (act == 2) v = fminf(fmaxf(v, 0.f), 6.f);  // ReLU6
       out.e[e] = f2b(v);
     }
     *(uint4*)(y + i) = out.u;
@@ -249,6 +251,12 @@ __global__ __launch_bounds__(256) void k_cast_bnact(
         if (b2f(y_up[i]) <= 0.f) g = 0.f;
       } else if (mask_mode == 2) {
         if (fmaf(ga, xv, gb) <= 0.f) g = 0.f;
+      } else if (mask_mode == 3) {
+        float yv = b2f(y_up[i]);
+        if (yv <= 0.f || yv >= 6.f) g = 0.f;
+      } else if (mask_mode == 4) {
+        float bn = fmaf(ga, xv, gb);
+        if (bn <= 0.f || bn >= 6.f) g = 0.f;
       }
       a_dz += g;
       a_dzx += g * (xv - mean) * invstd;
@@ -297,6 +305,12 @@ __global__ __launch_bounds__(256) void k_bnact_bwd_reduce(
         if (b2f(yout[i]) <= 0.f) g = 0.f;
       } else if (mask_mode == 2) {
         if (fmaf(ga, xv, gb) <= 0.f) g = 0.f;
+      } else if (mask_mode == 3) {
+        float yv = b2f(yout[i]);
+        if (yv <= 0.f || yv >= 6.f) g = 0.f;
+      } else if (mask_mode == 4) {
+        float bn = fmaf(ga, xv, gb);
+        if (bn <= 0.f || bn >= 6.f) g = 0.f;
       }
       a_dz += g;
       a_dzx += g * (xv - mean) * invstd;
@@ -340,6 +354,13 @@ __global__ __launch_bounds__(256) void k_bn_bwd_apply(
       } else if (mask_mode == 2) {
         float ga = gamma[c] * invstd;
         if (fmaf(ga, xv, beta[c] - mean * ga) <= 0.f) g = 0.f;
+      } else if (mask_mode == 3) {
+        float yv = b2f(yout[i]);
+        if (yv <= 0.f || yv >= 6.f) g = 0.f;
+      } else if (mask_mode == 4) {
+        float ga = gamma[c] * invstd;
+        float bn = fmaf(ga, xv, beta[c] - mean * ga);
+        if (bn <= 0.f || bn >= 6.f) g = 0.f;
       }
       if (dres != nullptr) dres[i] = f2b(g);
       float xh = (xv - mean) * invstd;
@@ -360,9 +381,9 @@ __global__ __launch_bounds__(256) void k_bn_bwd_apply(
     V8 dy8, x8, y8;
     dy8.u = *(const uint4*)(dy + i);
     x8.u = *(const uint4*)(x + i);
-    if (mask_mode == 1) y8.u = *(const uint4*)(yout + i);
+    if (mask_mode == 1 || mask_mode == 3) y8.u = *(const uint4*)(yout + i);
     F8 b8;
-    if (mask_mode == 2) b8 = load_f8(beta + c0);
+    if (mask_mode == 2 || mask_mode == 4) b8 = load_f8(beta + c0);
     V8 dc8, dr8;
 #pragma unroll
     for (int e = 0; e < 8; e++) {
@@ -373,6 +394,13 @@ __global__ __launch_bounds__(256) void k_bn_bwd_apply(
       } else if (mask_mode == 2) {
         float ga = g8.f[e] * istd8.f[e];
         if (fmaf(ga, xv, b8.f[e] - mean8.f[e] * ga) <= 0.f) g = 0.f;
+      } else if (mask_mode == 3) {
+        float yv = b2f(y8.e[e]);
+        if (yv <= 0.f || yv >= 6.f) g = 0.f;
+      } else if (mask_mode == 4) {
+        float ga = g8.f[e] * istd8.f[e];
+        float bn = fmaf(ga, xv, b8.f[e] - mean8.f[e] * ga);
+        if (bn <= 0.f || bn >= 6.f) g = 0.f;
       }
       dr8.e[e] = f2b(g);
       float xh = (xv - mean8.f[e]) * istd8.f[e];
@@ -382,6 +410,145 @@ __global__ __launch_bounds__(256) void k_bn_bwd_apply(
     }
     if (dres != nullptr) *(uint4*)(dres + i) = dr8.u;
     *(uint4*)(dconv + i) = dc8.u;
+  }
+}
+
+// ------------------------------------------------------- depthwise conv ----
+// MobileNet-family depthwise 3x3 (generic R,S ≤ 5): channel c maps to
+// itself, so the kernel is a per-channel stencil — bandwidth-bound, no
+// MFMA.  c-blocked layout (64 channels × 4 m-lanes per block, the proven
+// BN-reduce geometry): weights [R,S,C] bf16 are register-cached per
+// thread for the whole m-loop, per-channel BN batch stats (Σy,Σy²)
+// accumulate via LDS reduce + one atomic pair per block.
+#define DW_MAXRS 25
+
+__global__ __launch_bounds__(256) void k_dw_fwd(
+    const bf16* __restrict__ x, const bf16* __restrict__ w,
+    bf16* __restrict__ y, float* __restrict__ stats, int Nb, int H, int W,
+    int C, int Ho, int Wo, int R, int S, int str, int pad, long mchunk) {
+  __shared__ float s1[4][64];
+  __shared__ float s2[4][64];
+  const int c = blockIdx.x * 64 + (threadIdx.x & 63);
+  const int mlane = threadIdx.x >> 6;
+  const long M = (long)Nb * Ho * Wo;
+  const long mbeg = (long)blockIdx.y * mchunk;
+  const long mend = min(M, mbeg + mchunk);
+  float wreg[DW_MAXRS];
+  const int RS = R * S;
+  if (c < C)
+    for (int t = 0; t < RS; t++) wreg[t] = b2f(w[t * C + c]);
+  float a1 = 0.f, a2 = 0.f;
+  if (c < C) {
+    for (long m = mbeg + mlane; m < mend; m += 4) {
+      int n = (int)(m / (Ho * Wo)), hw = (int)(m % (Ho * Wo));
+      int ho = hw / Wo, wo = hw % Wo;
+      float acc = 0.f;
+      for (int r = 0; r < R; r++) {
+        int hi = ho * str - pad + r;
+        if (hi < 0 || hi >= H) continue;
+        for (int sx = 0; sx < S; sx++) {
+          int wi = wo * str - pad + sx;
+          if (wi < 0 || wi >= W) continue;
+          acc = fmaf(b2f(x[((long)(n * H + hi) * W + wi) * C + c]),
+                     wreg[r * S + sx], acc);
+        }
+      }
+      y[m * C + c] = f2b(acc);
+      a1 += acc;
+      a2 += acc * acc;
+    }
+  }
+  if (stats != nullptr) {
+    s1[mlane][threadIdx.x & 63] = a1;
+    s2[mlane][threadIdx.x & 63] = a2;
+    __syncthreads();
+    if (mlane == 0 && c < C) {
+      atomicAdd(&stats[c], s1[0][threadIdx.x] + s1[1][threadIdx.x] +
+                               s1[2][threadIdx.x] + s1[3][threadIdx.x]);
+      atomicAdd(&stats[C + c], s2[0][threadIdx.x] + s2[1][threadIdx.x] +
+                                   s2[2][threadIdx.x] + s2[3][threadIdx.x]);
+    }
+  }
+}
+
+// dx[n,hi,wi,c] = Σ_{r,s reaching it} dz[n,ho,wo,c] · w[r,s,c]
+__global__ __launch_bounds__(256) void k_dw_dgrad(
+    const bf16* __restrict__ dz, const bf16* __restrict__ w,
+    bf16* __restrict__ dx, int Nb, int H, int W, int C, int Ho, int Wo,
+    int R, int S, int str, int pad, long mchunk) {
+  const int c = blockIdx.x * 64 + (threadIdx.x & 63);
+  const int mlane = threadIdx.x >> 6;
+  const long M = (long)Nb * H * W;
+  const long mbeg = (long)blockIdx.y * mchunk;
+  const long mend = min(M, mbeg + mchunk);
+  float wreg[DW_MAXRS];
+  const int RS = R * S;
+  if (c < C)
+    for (int t = 0; t < RS; t++) wreg[t] = b2f(w[t * C + c]);
+  if (c >= C) return;
+  for (long m = mbeg + mlane; m < mend; m += 4) {
+    int n = (int)(m / (H * W)), hw = (int)(m % (H * W));
+    int hi = hw / W, wi = hw % W;
+    float acc = 0.f;
+    for (int r = 0; r < R; r++) {
+      int hs = hi + pad - r;
+      if (hs < 0 || hs % str != 0) continue;
+      int ho = hs / str;
+      if (ho >= Ho) continue;
+      for (int sx = 0; sx < S; sx++) {
+        int ws = wi + pad - sx;
+        if (ws < 0 || ws % str != 0) continue;
+        int wo = ws / str;
+        if (wo >= Wo) continue;
+        acc = fmaf(b2f(dz[((long)(n * Ho + ho) * Wo + wo) * C + c]),
+                   wreg[r * S + sx], acc);
+      }
+    }
+    dx[m * C + c] = f2b(acc);
+  }
+}
+
+// dW[r,s,c] += Σ_m x[...]·dz[...]; 9 register partials per thread, LDS
+// reduce over the 4 m-lanes, one atomicAdd per (r,s,c) per block.
+__global__ __launch_bounds__(256) void k_dw_wgrad(
+    const bf16* __restrict__ x, const bf16* __restrict__ dz,
+    float* __restrict__ dw, int Nb, int H, int W, int C, int Ho, int Wo,
+    int R, int S, int str, int pad, long mchunk) {
+  __shared__ float red[4][64];
+  const int c = blockIdx.x * 64 + (threadIdx.x & 63);
+  const int mlane = threadIdx.x >> 6;
+  const long M = (long)Nb * Ho * Wo;
+  const long mbeg = (long)blockIdx.y * mchunk;
+  const long mend = min(M, mbeg + mchunk);
+  const int RS = R * S;
+  float acc[DW_MAXRS];
+  for (int t = 0; t < RS; t++) acc[t] = 0.f;
+  if (c < C) {
+    for (long m = mbeg + mlane; m < mend; m += 4) {
+      int n = (int)(m / (Ho * Wo)), hw = (int)(m % (Ho * Wo));
+      int ho = hw / Wo, wo = hw % Wo;
+      float g = b2f(dz[m * C + c]);
+      for (int r = 0; r < R; r++) {
+        int hi = ho * str - pad + r;
+        if (hi < 0 || hi >= H) continue;
+        for (int sx = 0; sx < S; sx++) {
+          int wi = wo * str - pad + sx;
+          if (wi < 0 || wi >= W) continue;
+          acc[r * S + sx] = fmaf(
+              b2f(x[((long)(n * H + hi) * W + wi) * C + c]), g,
+              acc[r * S + sx]);
+        }
+      }
+    }
+  }
+  for (int t = 0; t < RS; t++) {
+    red[mlane][threadIdx.x & 63] = acc[t];
+    __syncthreads();
+    if (mlane == 0 && c < C)
+      atomicAdd(&dw[t * C + c], red[0][threadIdx.x] + red[1][threadIdx.x] +
+                                    red[2][threadIdx.x] +
+                                    red[3][threadIdx.x]);
+    __syncthreads();
   }
 }
 
@@ -840,6 +1007,47 @@ void launch_bn_bwd_apply(const void* dy, const void* yout, const void* x,
       (const bf16*)dy, (const bf16*)yout, (const bf16*)x, smean, sinvstd,
       gamma, beta, sum_dz, sum_dzx, (bf16*)dconv, (bf16*)dres, M, C,
       mask_mode);
+}
+
+void launch_dw_fwd(const void* x, const void* w, void* y, float* stats,
+                   int Nb, int H, int W, int C, int Ho, int Wo, int R,
+                   int S, int str, int pad, hipStream_t st) {
+  int cblocks = (C + 63) / 64;
+  long M = (long)Nb * Ho * Wo;
+  int msplit = (int)min((long)256, max((long)1, (long)(768 / cblocks)));
+  long mchunk = (M + msplit - 1) / msplit;
+  msplit = (int)((M + mchunk - 1) / mchunk);
+  dim3 grid(cblocks, msplit);
+  k_dw_fwd<<<grid, 256, 0, st>>>((const bf16*)x, (const bf16*)w, (bf16*)y,
+                                 stats, Nb, H, W, C, Ho, Wo, R, S, str, pad,
+                                 mchunk);
+}
+
+void launch_dw_dgrad(const void* dz, const void* w, void* dx, int Nb, int H,
+                     int W, int C, int Ho, int Wo, int R, int S, int str,
+                     int pad, hipStream_t st) {
+  int cblocks = (C + 63) / 64;
+  long M = (long)Nb * H * W;
+  int msplit = (int)min((long)256, max((long)1, (long)(768 / cblocks)));
+  long mchunk = (M + msplit - 1) / msplit;
+  msplit = (int)((M + mchunk - 1) / mchunk);
+  dim3 grid(cblocks, msplit);
+  k_dw_dgrad<<<grid, 256, 0, st>>>((const bf16*)dz, (const bf16*)w,
+                                   (bf16*)dx, Nb, H, W, C, Ho, Wo, R, S,
+                                   str, pad, mchunk);
+}
+
+void launch_dw_wgrad(const void* x, const void* dz, float* dw, int Nb,
+                     int H, int W, int C, int Ho, int Wo, int R, int S,
+                     int str, int pad, hipStream_t st) {
+  int cblocks = (C + 63) / 64;
+  long M = (long)Nb * Ho * Wo;
+  int msplit = (int)min((long)64, max((long)1, (long)(512 / cblocks)));
+  long mchunk = (M + msplit - 1) / msplit;
+  msplit = (int)((M + mchunk - 1) / mchunk);
+  dim3 grid(cblocks, msplit);
+  k_dw_wgrad<<<grid, 256, 0, st>>>((const bf16*)x, (const bf16*)dz, dw, Nb,
+                                   H, W, C, Ho, Wo, R, S, str, pad, mchunk);
 }
 
 void launch_maxpool_fwd(const void* x, void* y, unsigned char* idx, int Nb,

@@ -37,9 +37,14 @@ def test_dp_engine_gpu(tmp_path):
 @pytest.mark.timeout(300)
 def test_pp_engine_gpu(tmp_path):
     from layer_model_parallel_train import run_model_parallel
-    df = run_model_parallel(1, 1, 128, str(tmp_path / "pp"), batch_size=32,
+    df = run_model_parallel(1, 4, 128, str(tmp_path / "pp"), batch_size=32,
                             synthetic=True)
-    _check(df, 1, bandwidth=True)
+    _check(df, 4, bandwidth=True)
+    # regression: conv-weight bf16 shadows must refresh after optimizer
+    # steps — with stale shadows only BN params train and loss stays flat
+    by_epoch = df.groupby("epoch")["loss"].mean()
+    assert by_epoch.iloc[-1] < by_epoch.iloc[0] * 0.9, \
+        f"pipeline not learning: {by_epoch.tolist()}"
 
 
 @pytest.mark.timeout(300)

@@ -73,6 +73,9 @@ def main():
                 loss.backward()
                 ddp.finalize_backward()
                 opt.step()
+                if ctx.is_gpu:
+                    from horizonml_amd.models import refresh_all_shadows
+                    refresh_all_shadows(model)
                 bs = y.shape[0]
                 loss_sum += float(loss.detach()) * bs
                 correct += int((logits.detach().argmax(1) == y).sum())
