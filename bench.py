@@ -41,6 +41,8 @@ def parse_args():
     ap.add_argument("--optimizer", type=str, default="adam")
     ap.add_argument("--no-graph", action="store_true",
                     help="disable hipGraph capture (eager fallback)")
+    ap.add_argument("--infer", action="store_true",
+                    help="serving mode: eval-only forward throughput")
     return ap.parse_args()
 
 
@@ -86,7 +88,10 @@ def main():
 
     torch.manual_seed(1234)  # identical replicas on every rank
     model = build_model(args.model, num_classes=10).to(dev)
-    model.train()
+    if args.infer:
+        model.eval()
+    else:
+        model.train()
     mgr = FlatParamManager(model, dev)
     opt = (HorizonAdam(mgr, lr=1e-3) if args.optimizer == "adam"
            else HorizonSGD(mgr, lr=0.1))
@@ -105,6 +110,10 @@ def main():
     inv_world = 1.0 / world
 
     def train_step():
+        if args.infer:  # serving: eval-mode forward only
+            with torch.no_grad():
+                logits = model(x_static)
+            return logits.float().sum()
         logits = model(x_static)
         loss = cross_entropy(logits, y_static)
         loss.backward()
@@ -211,7 +220,8 @@ def main():
             "dtype": "bf16",
             "data": "synthetic",
             "config": {
-                "model": "resnet18_cifar10",
+                "model": args.model + ("_infer" if args.infer else
+                                       "_cifar10"),
                 "global_batch": global_batch,
                 "seq_len": None,
                 "image": "3x32x32",
