@@ -96,13 +96,39 @@ hipStream_t cur_stream() { return c10::hip::getCurrentHIPStream().stream(); }
 // 2 launches instead of ~40, full chip occupancy, no atomics.  Task tables
 // travel as kernel arguments by value, so the path is hipGraph-capture-safe
 // (no host staging buffers whose contents could change between replays).
-constexpr int WG_MAX_TASKS = 24;
+constexpr int WG_MAX_TASKS = 20;
+
+struct MagicP {
+  unsigned m[4];
+  int s[4];
+  unsigned d[4];
+};
+
+static void magic_u32_h(unsigned d, unsigned* m, int* sh) {
+  if (d <= 1) { *m = 0; *sh = 0; return; }
+  int p = 0;
+  while ((1ull << p) < d) p++;
+  *m = (unsigned)((((unsigned long long)((1ull << p) - d) << 32) / d) + 1);
+  *sh = p;
+}
+
+static MagicP make_magic_mode1(const ConvP& p) {
+  MagicP mg{};
+  unsigned d[4] = {(unsigned)(p.Ho * p.Wo), (unsigned)p.Wo,
+                   (unsigned)(p.S * p.C), (unsigned)p.C};
+  for (int i = 0; i < 4; i++) {
+    mg.d[i] = d[i] ? d[i] : 1;
+    magic_u32_h(mg.d[i], &mg.m[i], &mg.s[i]);
+  }
+  return mg;
+}
 
 struct WgradTask {
   const void* X;
   const void* Dz;
   float* out;
   ConvP p;
+  MagicP mg;
   int Ntot, mchunk, msplit, tx, ty;
   int base;
   int vec;
@@ -207,6 +233,7 @@ void flush_one_group(int group_lo, int group_hi) {
       t.out = msplit[lo + j] > 1 ? ws_base + ws_off[lo + j]
                                  : pw.dw.data_ptr<float>();
       t.p = pw.p;
+      t.mg = make_magic_mode1(pw.p);
       t.Ntot = pw.p.K;
       t.mchunk = mchunk[lo + j];
       t.msplit = msplit[lo + j];

@@ -31,3 +31,21 @@ DEV float wave_reduce_sum(float x) {
     x += __shfl_xor(x, off, 64);
   return x;
 }
+
+// Fast division by a runtime constant (cutlass FastDivmod form):
+// q = (umulhi(n, m) + n) >> s for 1 < d < 2^31, 0 <= n < 2^31; d == 1 is
+// the identity.  The implicit-GEMM gathers decompose linear indices with
+// 4-5 divisions per vec8 — emulated u32 division costs ~25-40 VALU each.
+struct MagicP {
+  unsigned m[4];
+  int s[4];
+  unsigned d[4];
+};
+
+DEV unsigned fdiv(unsigned n, const MagicP& mg, int i) {
+  return mg.d[i] == 1 ? n : (__umulhi(n, mg.m[i]) + n) >> mg.s[i];
+}
+
+DEV unsigned fmod(unsigned n, unsigned q, const MagicP& mg, int i) {
+  return n - q * mg.d[i];
+}

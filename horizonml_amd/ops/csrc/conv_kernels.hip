@@ -33,55 +33,91 @@ struct ConvP {
   int M, Kd;               // GEMM rows, reduction size
 };
 
+// ------------------------------------------------------------- fast div ----
+static void magic_u32(unsigned d, unsigned* m, int* sh) {
+  if (d <= 1) { *m = 0; *sh = 0; return; }
+  int p = 0;
+  while ((1ull << p) < d) p++;
+  *m = (unsigned)((((unsigned long long)((1ull << p) - d) << 32) / d) + 1);
+  *sh = p;
+}
+
+// gather divisors per mode: MODE1 {Ho*Wo, Wo, S*C, C};
+// MODE2 {H*W, W, S*K, K} (index 3 also serves b_addr's /K %K)
+static MagicP make_magic(const ConvP& p, int mode) {
+  MagicP mg{};
+  unsigned d[4];
+  if (mode == 2) {
+    d[0] = (unsigned)(p.H * p.W);
+    d[1] = (unsigned)p.W;
+    d[2] = (unsigned)(p.S * p.K);
+    d[3] = (unsigned)p.K;
+  } else {
+    d[0] = (unsigned)(p.Ho * p.Wo);
+    d[1] = (unsigned)p.Wo;
+    d[2] = (unsigned)(p.S * p.C);
+    d[3] = (unsigned)p.C;
+  }
+  for (int i = 0; i < 4; i++) {
+    mg.d[i] = d[i] ? d[i] : 1;
+    magic_u32(mg.d[i], &mg.m[i], &mg.s[i]);
+  }
+  return mg;
+}
+
 // ---------------------------------------------------------------- staging --
 template <int MODE>
-DEV int a_addr(const ConvP& p, int m, int k, bool& valid) {
+DEV int a_addr(const ConvP& p, const MagicP& mg, int m, int k, bool& valid) {
   if (MODE == 0) {
     valid = (m < p.M) & (k < p.Kd);
     return m * p.Kd + k;
   } else if (MODE == 1) {
-    // x gather: m -> (n,ho,wo), k -> (r,s,c)
-    int n = m / (p.Ho * p.Wo), hw = m % (p.Ho * p.Wo);
-    int ho = hw / p.Wo, wo = hw % p.Wo;
-    int r = k / (p.S * p.C), rm = k % (p.S * p.C);
-    int s = rm / p.C, c = rm % p.C;
-    int hi = ho * p.str - p.pad + r, wi = wo * p.str - p.pad + s;
+    // x gather: m -> (n,ho,wo), k -> (r,s,c); index decompositions via
+    // magic-number division (the emulated u32 divides dominated the VALU)
+    unsigned n = fdiv(m, mg, 0), hw = fmod(m, n, mg, 0);
+    unsigned ho = fdiv(hw, mg, 1), wo = fmod(hw, ho, mg, 1);
+    unsigned r = fdiv(k, mg, 2), rm = fmod(k, r, mg, 2);
+    unsigned sx = fdiv(rm, mg, 3), c = fmod(rm, sx, mg, 3);
+    int hi = (int)ho * p.str - p.pad + (int)r;
+    int wi = (int)wo * p.str - p.pad + (int)sx;
     valid = (m < p.M) & (k < p.Kd) & (hi >= 0) & (hi < p.H) & (wi >= 0) & (wi < p.W);
-    return ((n * p.H + hi) * p.W + wi) * p.C + c;
+    return (((int)n * p.H + hi) * p.W + wi) * p.C + (int)c;
   } else {
     // dz gather (dgrad): m -> (n,hi,wi) over input dims, k -> (r,s,ko)
-    int n = m / (p.H * p.W), hw = m % (p.H * p.W);
-    int hi = hw / p.W, wi = hw % p.W;
-    int r = k / (p.S * p.K), rm = k % (p.S * p.K);
-    int s = rm / p.K, ko = rm % p.K;
-    int hs = hi + p.pad - r, ws = wi + p.pad - s;
+    unsigned n = fdiv(m, mg, 0), hw = fmod(m, n, mg, 0);
+    unsigned hi = fdiv(hw, mg, 1), wi = fmod(hw, hi, mg, 1);
+    unsigned r = fdiv(k, mg, 2), rm = fmod(k, r, mg, 2);
+    unsigned sx = fdiv(rm, mg, 3), ko = fmod(rm, sx, mg, 3);
+    int hs = (int)hi + p.pad - (int)r, ws = (int)wi + p.pad - (int)sx;
     bool ok = (m < p.M) & (k < p.Kd) & (hs >= 0) & (ws >= 0) &&
               (hs % p.str == 0) && (ws % p.str == 0);
-    int ho = hs / p.str, wo = ws / p.str;
+    int ho = hs / p.str, wo = ws / p.str;  // str is 1 or 2: cheap
     ok = ok && (ho < p.Ho) && (wo < p.Wo);
     valid = ok;
-    return ((n * p.Ho + ho) * p.Wo + wo) * p.K + ko;
+    return (((int)n * p.Ho + ho) * p.Wo + wo) * p.K + (int)ko;
   }
 }
 
 template <int MODE>
-DEV int b_addr(const ConvP& p, int n, int k, int Ntot, bool& valid) {
+DEV int b_addr(const ConvP& p, const MagicP& mg, int n, int k, int Ntot,
+               bool& valid) {
   if (MODE == 2) {
     // W_rsck[(r*S+s)*C + c][ko] with k = (r*S+s)*K + ko, n = c
-    int rs = k / p.K, ko = k % p.K;
+    unsigned rs = fdiv(k, mg, 3), ko = fmod(k, rs, mg, 3);
     valid = (n < Ntot) & (k < p.Kd);
-    return (rs * p.C + n) * p.K + ko;
+    return ((int)rs * p.C + n) * p.K + (int)ko;
   }
   valid = (n < Ntot) & (k < p.Kd);
   return n * p.Kd + k;
 }
 
 template <int MODE, bool VEC>
-DEV V8 load8_a(const bf16* __restrict__ src, const ConvP& p, int m, int k) {
+DEV V8 load8_a(const bf16* __restrict__ src, const ConvP& p,
+               const MagicP& mg, int m, int k) {
   V8 v;
   if (VEC) {
     bool ok;
-    int a = a_addr<MODE>(p, m, k, ok);
+    int a = a_addr<MODE>(p, mg, m, k, ok);
     if (ok) v.u = *(const uint4*)(src + a);
     else v.u = uint4{0, 0, 0, 0};
   } else if (MODE == 1) {
@@ -116,7 +152,7 @@ DEV V8 load8_a(const bf16* __restrict__ src, const ConvP& p, int m, int k) {
   } else {
     for (int e = 0; e < 8; e++) {
       bool ok;
-      int a = a_addr<MODE>(p, m, k + e, ok);
+      int a = a_addr<MODE>(p, mg, m, k + e, ok);
       v.e[e] = ok ? src[a] : (bf16)0.f;
     }
   }
@@ -124,18 +160,18 @@ DEV V8 load8_a(const bf16* __restrict__ src, const ConvP& p, int m, int k) {
 }
 
 template <int MODE, bool VEC>
-DEV V8 load8_b(const bf16* __restrict__ src, const ConvP& p, int n, int k,
-               int Ntot) {
+DEV V8 load8_b(const bf16* __restrict__ src, const ConvP& p,
+               const MagicP& mg, int n, int k, int Ntot) {
   V8 v;
   if (VEC) {
     bool ok;
-    int a = b_addr<MODE>(p, n, k, Ntot, ok);
+    int a = b_addr<MODE>(p, mg, n, k, Ntot, ok);
     if (ok) v.u = *(const uint4*)(src + a);
     else v.u = uint4{0, 0, 0, 0};
   } else {
     for (int e = 0; e < 8; e++) {
       bool ok;
-      int a = b_addr<MODE>(p, n, k + e, Ntot, ok);
+      int a = b_addr<MODE>(p, mg, n, k + e, Ntot, ok);
       v.e[e] = ok ? src[a] : (bf16)0.f;
     }
   }
@@ -152,7 +188,8 @@ template <int MODE, bool VECA, bool VECB, bool STATS, bool SPLIT>
 __global__ __launch_bounds__(256) void k_conv_mfma(
     const bf16* __restrict__ A, const bf16* __restrict__ Bw,
     bf16* __restrict__ Y, float* __restrict__ ws_out,
-    float* __restrict__ stats, ConvP p, int Ntot, int kchunk, int accum) {
+    float* __restrict__ stats, ConvP p, MagicP mg, int Ntot, int kchunk,
+    int accum) {
   // Double-buffered LDS, ONE barrier per K-step, 2-deep register
   // prefetch: tile k+2's global loads are in flight while tile k computes,
   // so the ~900-cycle HBM latency is fully hidden even at 1 block/CU
@@ -173,15 +210,15 @@ __global__ __launch_bounds__(256) void k_conv_mfma(
 
   // prologue: tile 0 -> LDS[0]; tile 1 -> registers
   {
-    V8 a0 = load8_a<MODE, VECA>(A, p, m0 + srow, kbeg + scol);
-    V8 b0 = load8_b<MODE, VECB>(Bw, p, n0 + srow, kbeg + scol, Ntot);
+    V8 a0 = load8_a<MODE, VECA>(A, p, mg, m0 + srow, kbeg + scol);
+    V8 b0 = load8_b<MODE, VECB>(Bw, p, mg, n0 + srow, kbeg + scol, Ntot);
     *(V8*)&As[0][srow * LDA + scol] = a0;
     *(V8*)&Bs[0][srow * LDA + scol] = b0;
   }
   V8 a_nx, b_nx;
   if (kbeg + 32 < kend) {
-    a_nx = load8_a<MODE, VECA>(A, p, m0 + srow, kbeg + 32 + scol);
-    b_nx = load8_b<MODE, VECB>(Bw, p, n0 + srow, kbeg + 32 + scol, Ntot);
+    a_nx = load8_a<MODE, VECA>(A, p, mg, m0 + srow, kbeg + 32 + scol);
+    b_nx = load8_b<MODE, VECB>(Bw, p, mg, n0 + srow, kbeg + 32 + scol, Ntot);
   }
   __syncthreads();
 
@@ -193,8 +230,9 @@ __global__ __launch_bounds__(256) void k_conv_mfma(
       *(V8*)&As[buf ^ 1][srow * LDA + scol] = a_nx;
       *(V8*)&Bs[buf ^ 1][srow * LDA + scol] = b_nx;
       if (k0 + 64 < kend) {
-        a_nx = load8_a<MODE, VECA>(A, p, m0 + srow, k0 + 64 + scol);
-        b_nx = load8_b<MODE, VECB>(Bw, p, n0 + srow, k0 + 64 + scol, Ntot);
+        a_nx = load8_a<MODE, VECA>(A, p, mg, m0 + srow, k0 + 64 + scol);
+        b_nx = load8_b<MODE, VECB>(Bw, p, mg, n0 + srow, k0 + 64 + scol,
+                                   Ntot);
       }
     }
     bf16x8 af[2], bf[2];
@@ -292,9 +330,9 @@ union WgradSmem {
 
 template <bool VECA>
 DEV void wgrad_tile(const bf16* __restrict__ X, const bf16* __restrict__ Dz,
-                    float* __restrict__ out, const ConvP& p, int Ntot,
-                    int mchunk, int tx, int ty, int z, bool split,
-                    WgradSmem& smem) {
+                    float* __restrict__ out, const ConvP& p,
+                    const MagicP& mg, int Ntot, int mchunk, int tx, int ty,
+                    int z, bool split, WgradSmem& smem) {
   bf16* At = smem.s.At;   // [k3][m]
   bf16* Dt = smem.s.Dt;   // [ko][m]
   const int k3_0 = tx * 64, n0 = ty * 64;
@@ -308,7 +346,7 @@ DEV void wgrad_tile(const bf16* __restrict__ X, const bf16* __restrict__ Dz,
 
   f32x4 acc[2][2] = {};
 
-  V8 a_nx = load8_a<1, VECA>(X, p, mbeg + sm, k3_0 + sv);
+  V8 a_nx = load8_a<1, VECA>(X, p, mg, mbeg + sm, k3_0 + sv);
   V8 d_nx;
   {
     int m = mbeg + sm, n = n0 + sv;
@@ -333,7 +371,7 @@ DEV void wgrad_tile(const bf16* __restrict__ X, const bf16* __restrict__ Dz,
     }
     __syncthreads();
     if (m0 + 32 < mend) {
-      a_nx = load8_a<1, VECA>(X, p, m0 + 32 + sm, k3_0 + sv);
+      a_nx = load8_a<1, VECA>(X, p, mg, m0 + 32 + sm, k3_0 + sv);
       int m = m0 + 32 + sm, n = n0 + sv;
       if (m < p.M && n < Ntot)
         d_nx.u = *(const uint4*)(Dz + (long)m * Ntot + n);
@@ -396,9 +434,9 @@ DEV void wgrad_tile(const bf16* __restrict__ X, const bf16* __restrict__ Dz,
 template <bool VECA>
 __global__ __launch_bounds__(256) void k_wgrad(
     const bf16* __restrict__ X, const bf16* __restrict__ Dz,
-    float* __restrict__ dW, ConvP p, int Ntot, int mchunk) {
+    float* __restrict__ dW, ConvP p, MagicP mg, int Ntot, int mchunk) {
   __shared__ WgradSmem smem;
-  wgrad_tile<VECA>(X, Dz, dW, p, Ntot, mchunk, blockIdx.x, blockIdx.y,
+  wgrad_tile<VECA>(X, Dz, dW, p, mg, Ntot, mchunk, blockIdx.x, blockIdx.y,
                    blockIdx.z, gridDim.z > 1, smem);
 }
 
@@ -435,13 +473,14 @@ __global__ __launch_bounds__(256) void k_wgrad_reduce(
 // (mchunk ≈ 512 rows) and the grid still fills 256 CUs without atomics.
 // Task tables travel as kernel arguments by value — no staging buffers,
 // hipGraph-capture-safe.
-#define WG_MAX_TASKS 24
+#define WG_MAX_TASKS 20
 
 struct WgradTask {
   const bf16* X;
   const bf16* Dz;
   float* out;      // dW when msplit==1, else slab base [msplit][Ntot][Kd]
   ConvP p;
+  MagicP mg;
   int Ntot, mchunk, msplit, tx, ty;  // tile counts
   int base;        // first block id of this task
   int vec;
@@ -463,11 +502,11 @@ __global__ __launch_bounds__(256) void k_wgrad_batched(WgradBatchArgs a) {
   int ty = rem / t.tx, tx = rem - ty * t.tx;
   __shared__ WgradSmem smem;
   if (t.vec)
-    wgrad_tile<true>(t.X, t.Dz, t.out, t.p, t.Ntot, t.mchunk, tx, ty, z,
-                     t.msplit > 1, smem);
+    wgrad_tile<true>(t.X, t.Dz, t.out, t.p, t.mg, t.Ntot, t.mchunk, tx, ty,
+                     z, t.msplit > 1, smem);
   else
-    wgrad_tile<false>(t.X, t.Dz, t.out, t.p, t.Ntot, t.mchunk, tx, ty, z,
-                      t.msplit > 1, smem);
+    wgrad_tile<false>(t.X, t.Dz, t.out, t.p, t.mg, t.Ntot, t.mchunk, tx,
+                      ty, z, t.msplit > 1, smem);
 }
 
 struct WredTask {
@@ -569,9 +608,10 @@ extern "C" void launch_conv_fwd(const void* x, const void* w, void* y,
   auto A = (const bf16*)x;
   auto B = (const bf16*)w;
   auto Y = (bf16*)y;
+  MagicP mg = make_magic(p, 1);
 #define CASE(VA, VB, ST)                                     \
   k_conv_mfma<1, VA, VB, ST, false><<<grid, 256, 0, st>>>(   \
-      A, B, Y, nullptr, stats, p, p.K, 0, 0)
+      A, B, Y, nullptr, stats, p, mg, p.K, 0, 0)
   if (vec && s) CASE(true, true, true);
   else if (vec) CASE(true, true, false);
   else if (s) CASE(false, false, true);
@@ -591,12 +631,12 @@ extern "C" void launch_conv_fwd_splitk(const void* x, const void* w,
   auto B = (const bf16*)w;
   if (vec)
     k_conv_mfma<1, true, true, false, true>
-        <<<grid, 256, 0, st>>>(A, B, nullptr, ws, nullptr, p, p.K, kchunk,
-                               0);
+        <<<grid, 256, 0, st>>>(A, B, nullptr, ws, nullptr, p,
+                               make_magic(p, 1), p.K, kchunk, 0);
   else
     k_conv_mfma<1, false, false, false, true>
-        <<<grid, 256, 0, st>>>(A, B, nullptr, ws, nullptr, p, p.K, kchunk,
-                               0);
+        <<<grid, 256, 0, st>>>(A, B, nullptr, ws, nullptr, p,
+                               make_magic(p, 1), p.K, kchunk, 0);
 }
 
 extern "C" void launch_conv_dgrad(const void* dz, const void* w_rsck,
@@ -612,22 +652,22 @@ extern "C" void launch_conv_dgrad(const void* dz, const void* w_rsck,
     dim3 grid(cdiv_h(p.C, 64), cdiv_h(p.M, 64), splitk);
     if (vec)
       k_conv_mfma<2, true, true, false, true>
-          <<<grid, 256, 0, st>>>(A, B, nullptr, ws, nullptr, p, p.C, kchunk,
-                                 0);
+          <<<grid, 256, 0, st>>>(A, B, nullptr, ws, nullptr, p,
+                                 make_magic(p, 2), p.C, kchunk, 0);
     else
       k_conv_mfma<2, false, false, false, true>
-          <<<grid, 256, 0, st>>>(A, B, nullptr, ws, nullptr, p, p.C, kchunk,
-                                 0);
+          <<<grid, 256, 0, st>>>(A, B, nullptr, ws, nullptr, p,
+                                 make_magic(p, 2), p.C, kchunk, 0);
   } else {
     dim3 grid(cdiv_h(p.C, 64), cdiv_h(p.M, 64));
     if (vec)
       k_conv_mfma<2, true, true, false, false>
-          <<<grid, 256, 0, st>>>(A, B, (bf16*)dx, nullptr, nullptr, p, p.C,
-                                 0, accum);
+          <<<grid, 256, 0, st>>>(A, B, (bf16*)dx, nullptr, nullptr, p,
+                                 make_magic(p, 2), p.C, 0, accum);
     else
       k_conv_mfma<2, false, false, false, false>
-          <<<grid, 256, 0, st>>>(A, B, (bf16*)dx, nullptr, nullptr, p, p.C,
-                                 0, accum);
+          <<<grid, 256, 0, st>>>(A, B, (bf16*)dx, nullptr, nullptr, p,
+                                 make_magic(p, 2), p.C, 0, accum);
   }
 }
 
@@ -638,14 +678,15 @@ extern "C" void launch_gemm_bf16(const void* a, const void* b, void* c, int M,
   p.Kd = K;
   dim3 grid(cdiv_h(N, 64), cdiv_h(M, 64));
   bool vec = (K % 8) == 0;
+  MagicP mg{};
   if (vec)
     k_conv_mfma<0, true, true, false, false>
         <<<grid, 256, 0, st>>>((const bf16*)a, (const bf16*)b, (bf16*)c,
-                               nullptr, nullptr, p, N, 0, 0);
+                               nullptr, nullptr, p, mg, N, 0, 0);
   else
     k_conv_mfma<0, false, false, false, false>
         <<<grid, 256, 0, st>>>((const bf16*)a, (const bf16*)b, (bf16*)c,
-                               nullptr, nullptr, p, N, 0, 0);
+                               nullptr, nullptr, p, mg, N, 0, 0);
 }
 
 extern "C" void launch_wgrad_batched(const void* args, int blocks,
@@ -678,12 +719,13 @@ extern "C" void launch_wgrad(const void* x, const void* dz, float* dw,
   dim3 grid(cdiv_h(p.Kd, 64), cdiv_h(p.K, 64), msplit);
   bool vec = (p.C % 8) == 0;
   float* out = msplit > 1 ? ws : dw;
+  MagicP mg = make_magic(p, 1);
   if (vec)
     k_wgrad<true><<<grid, 256, 0, st>>>((const bf16*)x, (const bf16*)dz, out,
-                                        p, p.K, mchunk);
+                                        p, mg, p.K, mchunk);
   else
-    k_wgrad<false><<<grid, 256, 0, st>>>((const bf16*)x, (const bf16*)dz, out,
-                                         p, p.K, mchunk);
+    k_wgrad<false><<<grid, 256, 0, st>>>((const bf16*)x, (const bf16*)dz,
+                                         out, p, mg, p.K, mchunk);
   if (msplit > 1) {
     long n = (long)p.K * p.Kd;
     int blocks = (int)min((n / 4 + 255) / 256, (long)1024);
