@@ -23,12 +23,20 @@ import torch.distributed as dist
 DEFAULT_TIMEOUT_S = 300  # reference parity: 300 s gloo op timeout (C1)
 
 
-def auto_backend(prefer: Optional[str] = None) -> str:
+def auto_backend(prefer: Optional[str] = None,
+                 world_size: Optional[int] = None) -> str:
+    """nccl (RCCL) on GPU hosts, gloo otherwise — and gloo whenever the
+    world oversubscribes the visible GPUs (RCCL rejects two ranks on one
+    device, e.g. the reference's default world_size=5 on a 1-GPU box)."""
     if prefer in ("nccl", "rccl"):
         return "nccl"
     if prefer == "gloo":
         return "gloo"
-    return "nccl" if torch.cuda.is_available() else "gloo"
+    if not torch.cuda.is_available():
+        return "gloo"
+    if world_size is not None and world_size > torch.cuda.device_count():
+        return "gloo"
+    return "nccl"
 
 
 def bind_gpu(rank: int) -> Optional[torch.device]:
@@ -64,7 +72,7 @@ def setup_distributed(rank: int, world_size: int, port: int,
     # keep explicit values authoritative for spawned workers
     os.environ["MASTER_ADDR"] = master_addr
     os.environ["MASTER_PORT"] = str(port)
-    be = auto_backend(backend)
+    be = auto_backend(backend, world_size)
     device = bind_gpu(rank) if be == "nccl" else None
     dist.init_process_group(backend=be, rank=rank, world_size=world_size,
                             timeout=datetime.timedelta(seconds=timeout_s))
@@ -81,7 +89,7 @@ def setup_from_env(backend: Optional[str] = None,
     legacy Docker entry mode (``train.py:15-41``) and torchrun."""
     rank = int(os.environ.get("RANK", 0))
     world = int(os.environ.get("WORLD_SIZE", 1))
-    be = auto_backend(backend)
+    be = auto_backend(backend, world)
     device = bind_gpu(rank) if be == "nccl" else None
     dist.init_process_group(backend=be, rank=rank, world_size=world,
                             timeout=datetime.timedelta(seconds=timeout_s))
