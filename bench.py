@@ -103,19 +103,17 @@ def main():
               .to(memory_format=torch.channels_last).to(torch.bfloat16)
               for _ in range(pool_n)]
     pool_y = [torch.randint(0, 10, (bs,), device=dev) for _ in range(pool_n)]
-    x_static = pool_x[0].clone()
-    y_static = pool_y[0].clone()
     comm_buf = (torch.zeros_like(mgr.grad, dtype=torch.bfloat16)
                 if use_comm else None)
     inv_world = 1.0 / world
 
-    def train_step():
+    def train_step(xb, yb):
         if args.infer:  # serving: eval-mode forward only
             with torch.no_grad():
-                logits = model(x_static)
+                logits = model(xb)
             return logits.float().sum()
-        logits = model(x_static)
-        loss = cross_entropy(logits, y_static)
+        logits = model(xb)
+        loss = cross_entropy(logits, yb)
         loss.backward()
         if use_comm:
             _ops.extension().flush_wgrad()    # wgrads complete before sync
@@ -131,53 +129,53 @@ def main():
     # hipGraph capture recipe (torch "whole-network capture"): warm up on a
     # SIDE stream so AccumulateGrad nodes are not bound to the default
     # stream, drop every reference to the warmup autograd graph, then
-    # capture one full step.
+    # capture one graph PER POOL ENTRY (each binds its own input tensors —
+    # no per-step staging copies at replay time).
     mode = "eager" if args.no_graph else "graph"
     side = torch.cuda.Stream()
     side.wait_stream(torch.cuda.current_stream())
     with torch.cuda.stream(side):
         for i in range(3):
-            x_static.copy_(pool_x[i % pool_n])
-            y_static.copy_(pool_y[i % pool_n])
-            loss = train_step()
+            loss = train_step(pool_x[i % pool_n], pool_y[i % pool_n])
     torch.cuda.current_stream().wait_stream(side)
     torch.cuda.synchronize()
     del loss
-    loss_static = None
+    losses_static = []
 
-    graph = None
+    graphs = None
     if mode == "graph":
         try:
-            graph = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(graph):
-                loss_static = train_step()
+            graphs = []
+            for i in range(pool_n):
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g):
+                    losses_static.append(train_step(pool_x[i], pool_y[i]))
+                graphs.append(g)
         except Exception as e:  # noqa: BLE001
             if rank == 0:
                 print(f"[bench] graph capture failed ({e!r}); eager fallback",
                       file=sys.stderr)
-            graph = None
+            graphs = None
             mode = "eager"
         if world > 1:
             # all ranks must agree on the execution mode BEFORE any replay:
             # a rank-divergent capture failure would otherwise deadlock the
             # captured collectives (replay on one side only)
-            ok = torch.tensor([1.0 if graph is not None else 0.0],
+            ok = torch.tensor([1.0 if graphs is not None else 0.0],
                               device=dev)
             dist.all_reduce(ok, op=dist.ReduceOp.MIN)
             if float(ok) < 1.0:
-                graph = None
+                graphs = None
                 mode = "eager"
-        if graph is not None:
-            graph.replay()
+        if graphs is not None:
+            graphs[0].replay()
             torch.cuda.synchronize()
 
     def run_step(i):
-        x_static.copy_(pool_x[i % pool_n])
-        y_static.copy_(pool_y[i % pool_n])
-        if graph is not None:
-            graph.replay()
+        if graphs is not None:
+            graphs[i % pool_n].replay()
         else:
-            train_step()
+            train_step(pool_x[i % pool_n], pool_y[i % pool_n])
 
     for i in range(args.warmup):
         run_step(i)
@@ -200,10 +198,11 @@ def main():
     ms_per_step = elapsed_s / args.steps * 1000.0
     global_batch = bs * world
     ips = global_batch * args.steps / elapsed_s
-    if graph is not None:
-        final_loss = float(loss_static.detach().float().cpu())
+    if graphs is not None:
+        final_loss = float(losses_static[-1].detach().float().cpu())
     else:
-        final_loss = float(train_step().detach().float().cpu())
+        final_loss = float(train_step(pool_x[0], pool_y[0])
+                           .detach().float().cpu())
 
     if rank == 0:
         out = {
