@@ -134,3 +134,18 @@ def test_partition_resnet50_8_stages():
         for s in segs:
             h = s(h)
     assert torch.allclose(h, ref, atol=1e-5), (h - ref).abs().max()
+
+
+def test_raw_loader_matches_normalized():
+    """raw=True (GPU engines: uint8 batches + on-device normalize) must
+    yield exactly the normalized loader's values."""
+    import torch
+    from horizonml_amd.data import get_dataloader
+    from horizonml_amd.data.cifar import normalize_uint8
+    a, _ = get_dataloader(0, 1, 16, 64, strategy="mp", synthetic=True)
+    b, _ = get_dataloader(0, 1, 16, 64, strategy="mp", synthetic=True,
+                          raw=True)
+    for (xa, ya), (xb, yb) in zip(a, b):
+        assert xb.dtype == torch.uint8
+        assert torch.equal(ya, yb)
+        assert torch.allclose(xa, normalize_uint8(xb), atol=1e-6)
