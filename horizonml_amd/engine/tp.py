@@ -90,13 +90,13 @@ def train_tp(ctx: DistContext, epochs: int, sample_size: int, logs_dir: str,
                     ddp.finalize_backward()
             with prof.compute():
                 optimizer.step()
+                if ctx.is_gpu:
+                    from ..models import refresh_all_shadows
+                    refresh_all_shadows(model)
             meters.update(loss, logits, y)
             if probe is not None:
                 probe.step()
             prof.step_end()
-        if ctx.is_gpu:
-            from ..models import refresh_all_shadows
-            refresh_all_shadows(model)
         epoch_time = time.time() - epoch_start
         loss_v, acc_v = meters.epoch_values()
         t = prof.epoch_end()

@@ -86,7 +86,12 @@ class ConvBNAct(nn.Module):
         if getattr(self, "_managed", False):
             return
         with torch.no_grad():
-            self.weight_bf16 = self.weight.detach().to(torch.bfloat16).contiguous()
+            if (self.weight_bf16.shape == self.weight.shape
+                    and self.weight_bf16.device == self.weight.device):
+                self.weight_bf16.copy_(self.weight.detach())
+            else:
+                self.weight_bf16 = self.weight.detach() \
+                    .to(torch.bfloat16).contiguous()
 
     def _shadow(self) -> torch.Tensor:
         if getattr(self, "_managed", False):
@@ -252,7 +257,26 @@ class Linear(nn.Module):
 
 
 def refresh_all_shadows(model: nn.Module):
-    """Refresh every bf16 weight shadow in `model` (call after param updates)."""
+    """Refresh every bf16 weight shadow in `model`.
+
+    Called after EVERY optimizer step by the eager engines; in-place
+    shadows + one ``_foreach_copy_`` keep the per-step cost to a couple of
+    multi-tensor kernels."""
+    dsts, srcs = [], []
     for m in model.modules():
-        if isinstance(m, (ConvBNAct, Linear, DepthwiseConvBNAct)):
-            m.refresh_shadow()
+        if isinstance(m, (ConvBNAct, Linear, DepthwiseConvBNAct)) \
+                and not getattr(m, "_managed", False):
+            if (m.weight_bf16.shape != m.weight.shape
+                    or m.weight_bf16.device != m.weight.device):
+                m.refresh_shadow()  # (re)allocate
+            else:
+                dsts.append(m.weight_bf16)
+                srcs.append(m.weight.detach())
+    if not dsts:
+        return
+    with torch.no_grad():
+        try:
+            torch._foreach_copy_(dsts, srcs)
+        except (RuntimeError, AttributeError):
+            for d, sr in zip(dsts, srcs):
+                d.copy_(sr)
