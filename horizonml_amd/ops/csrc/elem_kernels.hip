@@ -140,6 +140,34 @@ __global__ __launch_bounds__(256) void k_bn_apply(
   }
 }
 
+// Deterministic per-channel (Σx, Σx²) over a bf16 [M][C] tensor: ONE
+// block per 64 channels, fixed-order serial m-loop per lane + ordered LDS
+// reduce — no atomics, bitwise-reproducible (deterministic mode).
+__global__ __launch_bounds__(256) void k_stats_bf16_det(
+    const bf16* __restrict__ x, float* __restrict__ stats, long M, int C) {
+  __shared__ float s1[4][64];
+  __shared__ float s2[4][64];
+  const int c = blockIdx.x * 64 + (threadIdx.x & 63);
+  const int mlane = threadIdx.x >> 6;
+  float a1 = 0.f, a2 = 0.f;
+  if (c < C) {
+    for (long m = mlane; m < M; m += 4) {
+      float v = b2f(x[m * C + c]);
+      a1 += v;
+      a2 += v * v;
+    }
+  }
+  s1[mlane][threadIdx.x & 63] = a1;
+  s2[mlane][threadIdx.x & 63] = a2;
+  __syncthreads();
+  if (mlane == 0 && c < C) {
+    stats[c] = s1[0][threadIdx.x] + s1[1][threadIdx.x] +
+               s1[2][threadIdx.x] + s1[3][threadIdx.x];
+    stats[C + c] = s2[0][threadIdx.x] + s2[1][threadIdx.x] +
+                   s2[2][threadIdx.x] + s2[3][threadIdx.x];
+  }
+}
+
 // Per-channel (Σx, Σx²) over `nsplit` stacked f32 [M][C] slabs (split-K
 // conv path — slabs are summed here).  grid: (cdiv(C,64), msplit);
 // stats must be pre-zeroed.
@@ -769,6 +797,37 @@ __global__ __launch_bounds__(256) void k_linear_bwd_dw(
 }
 
 // ------------------------------------------------------- fused CE loss -----
+// Deterministic variant: ONE block, per-thread strided rows, ordered
+// block reduce of the loss (the default kernel's atomicAdd order varies).
+__global__ __launch_bounds__(256) void k_ce_fwd_bwd_det(
+    const float* __restrict__ logits, const long* __restrict__ target,
+    float* __restrict__ loss, float* __restrict__ dlogits, int B, int NC) {
+  __shared__ float part[256];
+  float acc = 0.f;
+  for (int b = threadIdx.x; b < B; b += 256) {
+    const float* row = logits + (long)b * NC;
+    float mx = row[0];
+    for (int j = 1; j < NC; j++) mx = fmaxf(mx, row[j]);
+    float se = 0.f;
+    for (int j = 0; j < NC; j++) se += __expf(row[j] - mx);
+    float lse = __logf(se) + mx;
+    int t = (int)target[b];
+    acc += (lse - row[t]) / (float)B;
+    float invB = 1.f / (float)B;
+    for (int j = 0; j < NC; j++) {
+      float pj = __expf(row[j] - lse);
+      dlogits[(long)b * NC + j] = (pj - (j == t ? 1.f : 0.f)) * invB;
+    }
+  }
+  part[threadIdx.x] = acc;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float s = 0.f;
+    for (int i = 0; i < 256; i++) s += part[i];
+    loss[0] = s;
+  }
+}
+
 // logits[B,NC] f32 -> mean NLL loss (atomic into loss[0]) + dlogits f32
 // (softmax − onehot)/B.  One thread per row (NC ≤ 32).
 __global__ __launch_bounds__(256) void k_ce_fwd_bwd(
@@ -925,6 +984,15 @@ static inline int gsz(long total, int block = 256, int cap = 4096) {
 
 extern "C" {
 
+static int g_det_kernels = 0;
+void set_kernels_deterministic(int on) { g_det_kernels = on; }
+
+void launch_stats_bf16_det(const void* x, float* stats, long M, int C,
+                           hipStream_t st) {
+  k_stats_bf16_det<<<(C + 63) / 64, 256, 0, st>>>((const bf16*)x, stats, M,
+                                                  C);
+}
+
 void launch_bn_apply(const void* x, const void* res, void* y,
                      const float* stats, const float* gamma,
                      const float* beta, float* rmean, float* rvar,
@@ -952,6 +1020,7 @@ void launch_stats_reduce(const float* ws, float* stats, long M, int C,
                          int nsplit, hipStream_t st) {
   int cblocks = (C + 63) / 64;
   int msplit = (int)min((long)256, max((long)1, (long)(768 / cblocks)));
+  if (g_det_kernels) msplit = 1;  // fixed cross-block reduction order
   long mchunk = (M + msplit - 1) / msplit;
   msplit = (int)((M + mchunk - 1) / mchunk);
   dim3 grid(cblocks, msplit);
@@ -972,6 +1041,7 @@ void launch_cast_bnact(const float* src, void* dst, long M, int C,
                        int mask_mode, hipStream_t st) {
   int cblocks = (C + 63) / 64;
   int msplit = (int)min((long)256, max((long)1, (long)(768 / cblocks)));
+  if (g_det_kernels) msplit = 1;
   long mchunk = (M + msplit - 1) / msplit;
   msplit = (int)((M + mchunk - 1) / mchunk);
   dim3 grid(cblocks, msplit);
@@ -989,6 +1059,7 @@ void launch_bnact_bwd_reduce(const void* dy, const void* yout, const void* x,
   // sum_dz/sum_dzx must be pre-zeroed (they are accumulated atomically)
   int cblocks = (C + 63) / 64;
   int msplit = (int)min((long)256, max((long)1, (long)(768 / cblocks)));
+  if (g_det_kernels) msplit = 1;
   long mchunk = (M + msplit - 1) / msplit;
   msplit = (int)((M + mchunk - 1) / mchunk);
   dim3 grid(cblocks, msplit);
@@ -1015,6 +1086,7 @@ void launch_dw_fwd(const void* x, const void* w, void* y, float* stats,
   int cblocks = (C + 63) / 64;
   long M = (long)Nb * Ho * Wo;
   int msplit = (int)min((long)256, max((long)1, (long)(768 / cblocks)));
+  if (g_det_kernels) msplit = 1;
   long mchunk = (M + msplit - 1) / msplit;
   msplit = (int)((M + mchunk - 1) / mchunk);
   dim3 grid(cblocks, msplit);
@@ -1043,6 +1115,7 @@ void launch_dw_wgrad(const void* x, const void* dz, float* dw, int Nb,
   int cblocks = (C + 63) / 64;
   long M = (long)Nb * Ho * Wo;
   int msplit = (int)min((long)64, max((long)1, (long)(512 / cblocks)));
+  if (g_det_kernels) msplit = 1;
   long mchunk = (M + msplit - 1) / msplit;
   msplit = (int)((M + mchunk - 1) / mchunk);
   dim3 grid(cblocks, msplit);
@@ -1097,7 +1170,12 @@ void launch_linear_bwd(const float* dy, const void* x, const void* w,
 
 void launch_ce_fwd_bwd(const float* logits, const long* target, float* loss,
                        float* dlogits, int B, int NC, hipStream_t st) {
-  k_ce_fwd_bwd<<<gsz(B), 256, 0, st>>>(logits, target, loss, dlogits, B, NC);
+  if (g_det_kernels)
+    k_ce_fwd_bwd_det<<<1, 256, 0, st>>>(logits, target, loss, dlogits, B,
+                                        NC);
+  else
+    k_ce_fwd_bwd<<<gsz(B), 256, 0, st>>>(logits, target, loss, dlogits, B,
+                                         NC);
 }
 
 void launch_adam_step(float* master, float* grad, float* m, float* v,

@@ -667,3 +667,38 @@ def test_checkpoint_flat_manager_roundtrip(tmp_path):
         a = m1(x)
         b = m2(x)
     assert torch.equal(a, b)
+
+
+def test_deterministic_mode_bitwise_reproducible():
+    """set_deterministic(True): two identical training runs must produce
+    BITWISE-equal losses and gradients (ordered reductions replace every
+    atomic-order dependence)."""
+    from horizonml_amd.engine.flat import FlatParamManager, HorizonAdam
+    from horizonml_amd.models import resnet18
+    from horizonml_amd.models._functional_gpu import cross_entropy
+    dev = torch.device("cuda", 0)
+    _C().set_deterministic(True)
+    try:
+        outs = []
+        for _ in range(2):
+            torch.manual_seed(0)
+            model = resnet18(num_classes=10).to(dev)
+            mgr = FlatParamManager(model, dev)
+            opt = HorizonAdam(mgr, lr=1e-3)
+            torch.manual_seed(1)
+            x = torch.randn(16, 3, 32, 32, device=dev).to(
+                memory_format=torch.channels_last).to(torch.bfloat16)
+            y = torch.randint(0, 10, (16,), device=dev)
+            losses = []
+            for _ in range(4):
+                loss = cross_entropy(model(x), y)
+                loss.backward()
+                opt.step()
+                losses.append(float(loss.detach()))
+            torch.cuda.synchronize()
+            outs.append((losses, mgr.master.clone(), mgr.grad.clone()))
+        assert outs[0][0] == outs[1][0], \
+            f"losses differ: {outs[0][0]} vs {outs[1][0]}"
+        assert torch.equal(outs[0][1], outs[1][1]), "masters differ"
+    finally:
+        _C().set_deterministic(False)
