@@ -11,7 +11,6 @@ reference paths.
 """
 from __future__ import annotations
 
-import torch
 import torch.nn as nn
 
 from .layers import ConvBNAct, DepthwiseConvBNAct, GlobalAvgPool, Linear

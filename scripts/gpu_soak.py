@@ -1,9 +1,5 @@
 """Memory-stability soak: 3000 graph-replayed steps, assert no growth."""
-import sys
 import torch
-sys.argv = ["bench.py", "--steps", "1", "--warmup", "1"]
-import bench  # noqa
-# run the bench main once manually would exit; instead inline a small soak
 from horizonml_amd.engine.flat import FlatParamManager, HorizonAdam
 from horizonml_amd.models import build_model
 from horizonml_amd.models._functional_gpu import cross_entropy
