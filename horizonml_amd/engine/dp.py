@@ -93,7 +93,7 @@ def train_dp(ctx: DistContext, epochs: int, sample_size: int, logs_dir: str,
 
             x, y = _to_device(x, y, ctx)
             with prof.compute():
-                optimizer.zero_grad(set_to_none=False)
+                optimizer.zero_grad(set_to_none=True)
                 logits = ddp(x)
                 if logits.is_cuda:
                     from ..models._functional_gpu import cross_entropy

@@ -116,7 +116,7 @@ def train_pp(ctx: DistContext, epochs: int, sample_size: int, logs_dir: str,
                 x = x.to(memory_format=torch.channels_last).to(torch.bfloat16)
                 y = y.to(ctx.device, non_blocking=True)
             if optimizer is not None:
-                optimizer.zero_grad(set_to_none=False)
+                optimizer.zero_grad(set_to_none=True)
             with prof.compute():
                 total_loss, n, corr = stage.forward_backward(
                     x if stage.is_first else None,

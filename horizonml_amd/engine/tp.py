@@ -77,7 +77,7 @@ def train_tp(ctx: DistContext, epochs: int, sample_size: int, logs_dir: str,
                 x = x.to(memory_format=torch.channels_last).to(torch.bfloat16)
                 y = y.to(ctx.device, non_blocking=True)
             with prof.compute():
-                optimizer.zero_grad(set_to_none=False)
+                optimizer.zero_grad(set_to_none=True)
                 logits = model(x)
                 if logits.is_cuda:
                     from ..models._functional_gpu import cross_entropy

@@ -62,7 +62,7 @@ def main():
                     x = x.to(
                         memory_format=torch.channels_last).to(torch.bfloat16)
                     y = y.to(ctx.device)
-                opt.zero_grad(set_to_none=False)
+                opt.zero_grad(set_to_none=True)
                 logits = ddp(x)
                 if logits.is_cuda:
                     from horizonml_amd.models._functional_gpu import \
