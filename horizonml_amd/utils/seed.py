@@ -52,9 +52,10 @@ def enable_deterministic(warn_only: bool = True) -> None:
     Besides ``torch.use_deterministic_algorithms``, this switches the
     gfx950 kernels to fixed-order reductions (BN batch stats, backward
     channel sums, CE loss) in place of their atomic accumulations — GPU
-    training becomes bitwise run-to-run reproducible at a measured
-    ~1.5-2x cost on the affected (small) kernels; weight gradients are
-    atomic-free in both modes.
+    training becomes bitwise run-to-run reproducible.  Measured cost at
+    batch 64: ~9x step time (the single-block serial reductions dominate
+    at small batches) — a debugging/reproducibility tool, not a production
+    mode.  Weight gradients are atomic-free in both modes.
     """
     import os
     os.environ.setdefault("CUBLAS_WORKSPACE_CONFIG", ":4096:8")
