@@ -334,7 +334,7 @@ class ResBlockFn(torch.autograd.Function):
             dxds, dw, dgamma, dbeta, _ = _C().conv_bn_act_bwd(
                 dres, y_ds, x, w_ds, _rsck(ds_mod, w_ds), convout_ds,
                 gamma_ds, beta_ds, smean_ds, sinvstd_ds, ds_mod.stride,
-                ds_mod.padding, False, True, False,
+                ds_mod.padding, ds_mod.act, True, False,
                 ds_mod.weight.grad if direct else None,
                 ds_mod.bn_weight.grad if direct else None,
                 ds_mod.bn_bias.grad if direct else None, cur, None, 0,
