@@ -239,7 +239,7 @@ class ResBlockFn(torch.autograd.Function):
                 x, ds_mod._shadow(), ds_mod.bn_weight, ds_mod.bn_bias,
                 ds_mod.running_mean, ds_mod.running_var, ds_mod.stride,
                 ds_mod.padding, ds_mod.momentum, ds_mod.eps, training,
-                False, None,
+                ds_mod.act, None,
                 getattr(ds_mod, "_stats_buf", None) if training else None)
             identity = y_ds
         else:
