@@ -1,0 +1,76 @@
+"""Unit tests for the small runtime/util layers (CPU-only)."""
+import os
+
+import pytest
+import torch
+
+from horizonml_amd.models.partition import partition_units, split_counts
+from horizonml_amd.runtime.distributed import auto_backend
+from horizonml_amd.runtime.launcher import timeout_for
+from horizonml_amd.utils.ports import find_free_port
+from horizonml_amd.utils.seed import shared_subset_indices
+
+
+def test_split_counts_balanced_contiguous():
+    assert split_counts(5, 5) == [1, 1, 1, 1, 1]
+    assert split_counts(5, 3) == [2, 2, 1]      # remainder spread first
+    assert split_counts(10, 8) == [2, 2, 1, 1, 1, 1, 1, 1]
+    assert split_counts(3, 5) == [1, 1, 1, 0, 0]  # empty tail stages
+    with pytest.raises(ValueError):
+        split_counts(4, 0)
+
+
+def test_partition_units_empty_stage_is_identity():
+    import torch.nn as nn
+    units = [(f"u{i}", nn.Linear(2, 2)) for i in range(3)]
+    segs = partition_units(units, 5)
+    assert len(segs) == 5
+    out = torch.randn(1, 2)
+    for s in segs[3:]:
+        assert torch.equal(s(out), out)  # identity stages pass through
+
+
+def test_timeout_for_reference_scaling():
+    # reference: max(base, base * n / 1000) — data_parallel_train.py:252
+    assert timeout_for(1000) == 120
+    assert timeout_for(500) == 120
+    assert timeout_for(5000) == 600
+    assert timeout_for(1000, base=400) == 400
+
+
+def test_find_free_port_binds():
+    import socket
+    p = find_free_port()
+    s = socket.socket()
+    s.bind(("127.0.0.1", p))  # must be bindable right after
+    s.close()
+
+
+def test_auto_backend_rules():
+    assert auto_backend("gloo") == "gloo"
+    assert auto_backend("nccl") == "nccl"
+    assert auto_backend("rccl") == "nccl"
+    if not torch.cuda.is_available():
+        assert auto_backend(None) == "gloo"
+        assert auto_backend(None, world_size=8) == "gloo"
+
+
+def test_shared_subset_deterministic_and_shared():
+    a = shared_subset_indices(50000, 1000, seed=7)
+    b = shared_subset_indices(50000, 1000, seed=7)
+    assert torch.equal(a, b)            # Q1 fix: every rank derives the same
+    c = shared_subset_indices(50000, 1000, seed=8)
+    assert not torch.equal(a, c)
+    assert len(set(a.tolist())) == 1000  # no duplicates
+    assert shared_subset_indices(10, 50).numel() == 10  # clamped
+
+
+def test_checkpoint_atomic_tmp(tmp_path):
+    """save_checkpoint writes tmp + rename — no partial file left behind."""
+    from horizonml_amd.models import resnet18
+    from horizonml_amd.utils.checkpoint import save_checkpoint
+    m = resnet18(num_classes=10)
+    path = str(tmp_path / "c.pt")
+    save_checkpoint(path, m, epoch=1)
+    assert os.path.isfile(path)
+    assert not os.path.exists(path + ".tmp")
