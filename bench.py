@@ -33,8 +33,10 @@ BASELINE_IMAGES_PER_SEC = 1000.0 / 19.6  # DP, 1000 samples, 5 ep
 def parse_args():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=50)
-    ap.add_argument("--warmup", type=int, default=10)
+    # long enough (~0.3 s timed region at 1 ms/step) that driver-side GPU
+    # sampling and clock cross-checks see a busy device
+    ap.add_argument("--steps", type=int, default=300)
+    ap.add_argument("--warmup", type=int, default=20)
     ap.add_argument("--batch-size", type=int, default=64,
                     help="per-GPU batch (reference parity: 64)")
     ap.add_argument("--model", type=str, default="resnet18")

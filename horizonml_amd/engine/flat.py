@@ -121,6 +121,21 @@ class HorizonAdam:
                                    grad_scale)
         self.mgr.refresh_rsck()
 
+    # -- checkpoint round-trip (utils/checkpoint.py) ----------------------
+    def state_dict(self):
+        return {"kind": "horizon_adam", "lr": self.lr, "betas": self.betas,
+                "eps": self.eps, "weight_decay": self.wd,
+                "m": self.m.detach().cpu(), "v": self.v.detach().cpu(),
+                "step_t": self.step_t.detach().cpu()}
+
+    def load_state_dict(self, state):
+        if state.get("kind") != "horizon_adam":
+            raise ValueError("checkpoint optimizer state is not HorizonAdam")
+        with torch.no_grad():
+            self.m.copy_(state["m"].to(self.m.device))
+            self.v.copy_(state["v"].to(self.v.device))
+            self.step_t.copy_(state["step_t"].to(self.step_t.device))
+
 
 class HorizonSGD:
     def __init__(self, mgr: FlatParamManager, lr: float = 0.1,
@@ -137,3 +152,18 @@ class HorizonSGD:
                                   zero_grad, grad_bf16, grad_scale)
         self.mgr.stats_arena.zero_()
         self.mgr.refresh_rsck()
+
+    def state_dict(self):
+        return {"kind": "horizon_sgd", "lr": self.lr, "momentum": self.mu,
+                "weight_decay": self.wd,
+                "mom": (self.mom.detach().cpu()
+                        if self.mom is not None else None)}
+
+    def load_state_dict(self, state):
+        if state.get("kind") != "horizon_sgd":
+            raise ValueError("checkpoint optimizer state is not HorizonSGD")
+        if (state.get("mom") is None) != (self.mom is None):
+            raise ValueError("momentum buffer mismatch with checkpoint")
+        if self.mom is not None:
+            with torch.no_grad():
+                self.mom.copy_(state["mom"].to(self.mom.device))

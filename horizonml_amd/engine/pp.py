@@ -80,7 +80,12 @@ def train_pp(ctx: DistContext, epochs: int, sample_size: int, logs_dir: str,
     ddp = None
     if dp_group is not None and has_params:
         from ..parallel import BucketedDataParallel
-        ddp = BucketedDataParallel(seg, profiler=prof, process_group=dp_group)
+        # microbatches>1 runs SEVERAL backward() calls before the step:
+        # hook-launched all-reduces would ship first-microbatch partial
+        # grads and race with later packs — defer to finalize_backward,
+        # which packs the fully-accumulated p.grad once per step
+        ddp = BucketedDataParallel(seg, profiler=prof, process_group=dp_group,
+                                   defer_reduction=microbatches > 1)
 
     writer = MetricsWriter(logs_dir, rank, sample_size, with_bandwidth=True,
                            with_gpu=ctx.is_gpu)

@@ -53,11 +53,20 @@ class BucketedDataParallel(nn.Module):
     def __init__(self, module: nn.Module, bucket_cap_mb: float = 25.0,
                  comm_dtype: Optional[torch.dtype] = None,
                  process_group=None, profiler=None,
-                 broadcast_params: bool = True, parameters=None):
+                 broadcast_params: bool = True, parameters=None,
+                 defer_reduction: bool = False):
+        """``defer_reduction``: do NOT launch bucket all-reduces from the
+        grad hooks; pack everything from the fully-accumulated ``p.grad`` in
+        ``finalize_backward`` instead.  Required whenever one optimizer step
+        spans SEVERAL ``backward()`` calls (pipeline microbatching in hybrid
+        DP×PP): hook-triggered launches would all-reduce first-microbatch
+        partial gradients and race with later microbatches packing into the
+        same flat buffer."""
         super().__init__()
         self.module = module
         self.group = process_group
         self.profiler = profiler
+        self.defer_reduction = defer_reduction
         self.world_size = (dist.get_world_size(process_group)
                            if dist.is_initialized() else 1)
         params = (list(parameters) if parameters is not None
@@ -117,7 +126,7 @@ class BucketedDataParallel(nn.Module):
 
     # -- backward-hook machinery ------------------------------------------
     def _grad_ready(self, p: torch.Tensor):
-        if self.world_size <= 1:
+        if self.world_size <= 1 or self.defer_reduction:
             return
         b, i = self._param_bucket[id(p)]
         b.views[i].copy_(p.grad.detach())  # cast into comm dtype
@@ -138,13 +147,19 @@ class BucketedDataParallel(nn.Module):
         inv = 1.0 / self.world_size
         for b in self.buckets:
             if b.pending != 0:
-                # grads for some params never materialized (e.g. frozen
-                # subgraph this step) — reduce what we have to stay collective
+                # deferred mode, or grads for some params never materialized
+                # (frozen subgraph this step) — (re)pack from the accumulated
+                # ``p.grad`` and reduce, staying collective across ranks
                 for v, p in zip(b.views, b.params):
                     if p.grad is None:
                         v.zero_()
+                    else:
+                        v.copy_(p.grad.detach())
                 b.work = dist.all_reduce(b.flat, op=dist.ReduceOp.SUM,
                                          group=self.group, async_op=True)
+                if self.profiler is not None:
+                    self.profiler.add_bytes(b.flat.numel()
+                                            * b.flat.element_size())
             if b.work is not None:
                 b.work.wait()
         for b in self.buckets:

@@ -24,17 +24,33 @@ DEFAULT_TIMEOUT_S = 300  # reference parity: 300 s gloo op timeout (C1)
 
 
 def auto_backend(prefer: Optional[str] = None,
-                 world_size: Optional[int] = None) -> str:
-    """nccl (RCCL) on GPU hosts, gloo otherwise — and gloo whenever the
-    world oversubscribes the visible GPUs (RCCL rejects two ranks on one
-    device, e.g. the reference's default world_size=5 on a 1-GPU box)."""
+                 world_size: Optional[int] = None,
+                 verbose: bool = True) -> str:
+    """nccl (RCCL) on GPU hosts, gloo otherwise — and gloo whenever this
+    node's rank count oversubscribes its visible GPUs (RCCL rejects two
+    ranks on one device, e.g. the reference's default world_size=5 on a
+    1-GPU box).
+
+    Oversubscription is judged on the ranks *per node*: under a multi-node
+    launcher ``LOCAL_WORLD_SIZE`` is authoritative (a 16-rank job over two
+    8-GPU nodes must stay on RCCL); in the single-node spawn path there is
+    no env and ``world_size`` == ranks on this node.  The fallback decision
+    is printed once so an accidental CPU run is never silent.
+    """
     if prefer in ("nccl", "rccl"):
         return "nccl"
     if prefer == "gloo":
         return "gloo"
     if not torch.cuda.is_available():
         return "gloo"
-    if world_size is not None and world_size > torch.cuda.device_count():
+    local_ranks = os.environ.get("LOCAL_WORLD_SIZE")
+    local_ranks = int(local_ranks) if local_ranks is not None else world_size
+    if local_ranks is not None and local_ranks > torch.cuda.device_count():
+        if verbose:
+            print(f"[horizonml] {local_ranks} ranks on this node > "
+                  f"{torch.cuda.device_count()} visible GPU(s) — falling "
+                  "back to gloo/CPU (pass --backend nccl to force RCCL)",
+                  flush=True)
         return "gloo"
     return "nccl"
 

@@ -39,7 +39,7 @@ def test_hybrid_dp2_pp2_end_to_end(tmp_path):
     assert (first["loss"] == 0).all()
 
 
-def _hybrid_param_worker(rank, world, port, q):
+def _hybrid_param_worker(rank, world, port, q, microbatches=1):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     from horizonml_amd.engine.pp import train_pp
     from horizonml_amd.runtime.distributed import (make_hybrid_groups,
@@ -53,6 +53,7 @@ def _hybrid_param_worker(rank, world, port, q):
                  model_name="resnet18", synthetic=True, lr=1e-3,
                  group=pp_group, n_stages=2, stage_idx=pp_stage,
                  dp_group=dp_group, data_rank=dp_rank, data_world=2,
+                 microbatches=microbatches,
                  log_progress=False, probe_divergence=False)
     seg = train_pp.last_segment
     flat = torch.cat([p.detach().flatten() for p in seg.parameters()]) \
@@ -63,13 +64,18 @@ def _hybrid_param_worker(rank, world, port, q):
 
 
 @pytest.mark.timeout(300)
-def test_hybrid_dp_sync_and_rank_layout():
+@pytest.mark.parametrize("microbatches", [1, 2])
+def test_hybrid_dp_sync_and_rank_layout(microbatches):
+    """DP-replica equality after training; microbatches=2 exercises the
+    deferred-reduction path (multiple backwards per step
+    must not all-reduce partial gradients)."""
     import torch.multiprocessing as mp
     from horizonml_amd.utils.ports import find_free_port
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
     port = find_free_port()
-    procs = [ctx.Process(target=_hybrid_param_worker, args=(r, 4, port, q))
+    procs = [ctx.Process(target=_hybrid_param_worker,
+                         args=(r, 4, port, q, microbatches))
              for r in range(4)]
     for p in procs:
         p.start()

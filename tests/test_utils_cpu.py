@@ -46,13 +46,25 @@ def test_find_free_port_binds():
     s.close()
 
 
-def test_auto_backend_rules():
+def test_auto_backend_rules(monkeypatch, capsys):
     assert auto_backend("gloo") == "gloo"
     assert auto_backend("nccl") == "nccl"
     assert auto_backend("rccl") == "nccl"
     if not torch.cuda.is_available():
         assert auto_backend(None) == "gloo"
         assert auto_backend(None, world_size=8) == "gloo"
+    # oversubscription is judged per NODE: a multi-node job (global world 16,
+    # LOCAL_WORLD_SIZE 8 on an 8-GPU node) must stay on RCCL
+    monkeypatch.setattr(torch.cuda, "is_available", lambda: True)
+    monkeypatch.setattr(torch.cuda, "device_count", lambda: 8)
+    monkeypatch.setenv("LOCAL_WORLD_SIZE", "8")
+    assert auto_backend(None, world_size=16) == "nccl"
+    # single-node spawn path: world_size IS the per-node rank count, and the
+    # gloo/CPU fallback is announced, never silent
+    monkeypatch.delenv("LOCAL_WORLD_SIZE", raising=False)
+    capsys.readouterr()
+    assert auto_backend(None, world_size=16) == "gloo"
+    assert "falling back to gloo" in capsys.readouterr().out
 
 
 def test_shared_subset_deterministic_and_shared():
