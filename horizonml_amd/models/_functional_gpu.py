@@ -19,6 +19,17 @@ def _C():
     return _ops.extension()
 
 
+def _bwd_done(mod):
+    """Backward-complete notification for the bucketed DP reducer
+    (``parallel/flat_reducer.py``): by the time a unit's autograd backward
+    returns, its BN/bias grads are launched and its deferred wgrad tasks
+    registered, so the unit's flat-gradient range may be flushed + reduced.
+    No-op unless a scheduler installed ``_bwd_done_cb`` on the module."""
+    cb = getattr(mod, "_bwd_done_cb", None)
+    if cb is not None:
+        cb()
+
+
 def _cl(t: torch.Tensor) -> torch.Tensor:
     if t.dim() == 4 and not t.is_contiguous(memory_format=torch.channels_last):
         return t.contiguous(memory_format=torch.channels_last)
@@ -72,6 +83,7 @@ class ConvBNActFn(torch.autograd.Function):
             beta, smean, sinvstd, mod.stride, mod.padding, mod.act, need_dx,
             ctx.has_res, dw_out, dg_out, db_out, None, None, 0, False, None,
             None)
+        _bwd_done(mod)
         if direct:
             return (dx if need_dx else None, None, None, None,
                     dres if ctx.has_res else None, None)
@@ -111,6 +123,7 @@ class DwConvBNActFn(torch.autograd.Function):
         dx, dw, dgamma, dbeta, dres = _C().dw_conv_bn_bwd(
             dy, y, x, w_bf16, convout, gamma, beta, smean, sinvstd,
             mod.stride, mod.padding, mod.act, need_dx, ctx.has_res)
+        _bwd_done(mod)
         return (dx if need_dx else None, dw, dgamma, dbeta,
                 dres if ctx.has_res else None, None)
 
@@ -181,6 +194,7 @@ class LinearFn(torch.autograd.Function):
                   else None)
         dx, dw, db = _C().linear_bwd(dy, x, w_bf16, need_dx, ctx.need_db,
                                      dw_out, db_out)
+        _bwd_done(mod)
         if direct:
             return (dx if need_dx else None, None, None, None)
         return (dx if need_dx else None, dw,
@@ -343,6 +357,7 @@ class ResBlockFn(torch.autograd.Function):
                 grads[3 * n: 3 * n + 3] = [dw, dgamma, dbeta]
             cur = dxds  # == cur buffer, accumulated
 
+        _bwd_done(mods[0])  # whole-block unit: callback lives on conv1
         return (cur, None, None, *grads)
 
 

@@ -209,6 +209,30 @@ void flush_wgrad() {
   g_pending.clear();
 }
 
+// Flush only the pending wgrads whose dW target lies inside the flat
+// gradient range [lo_elem, hi_elem) of `flat` (the FlatParamManager grad
+// buffer) — the per-bucket flush of the overlapped DP all-reduce
+// (parallel/flat_reducer.py).  Entries outside the range stay pending for
+// their own bucket.  Reuses flush_wgrad's dup-partitioned batched launch
+// by swapping the matching run into g_pending.
+void flush_wgrad_range(torch::Tensor flat, int64_t lo_elem, int64_t hi_elem) {
+  if (g_pending.empty()) return;
+  const char* base = (const char*)flat.data_ptr();
+  const char* plo = base + lo_elem * flat.element_size();
+  const char* phi = base + hi_elem * flat.element_size();
+  std::vector<PendingWgrad> take, keep;
+  for (auto& p : g_pending) {
+    const char* q = (const char*)p.dw.data_ptr();
+    if (q >= plo && q < phi)
+      take.push_back(std::move(p));
+    else
+      keep.push_back(std::move(p));
+  }
+  g_pending = std::move(take);
+  flush_wgrad();  // clears g_pending
+  g_pending = std::move(keep);
+}
+
 void flush_one_group(int group_lo, int group_hi) {
   auto st = cur_stream();
   const int n = group_hi - group_lo;
@@ -849,5 +873,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("wgrad_defer_enabled", &wgrad_defer_enabled);
   m.def("wgrad_pending", &wgrad_pending);
   m.def("flush_wgrad", &flush_wgrad);
+  m.def("flush_wgrad_range", &flush_wgrad_range);
   m.def("wgrad_only", &wgrad_only);
 }

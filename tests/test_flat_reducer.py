@@ -1,0 +1,139 @@
+"""Bucketed flat all-reduce (parallel/flat_reducer.py): schedule unit tests
++ 2-process gloo equivalence (bucketed == single-buffer, bit-for-bit f32).
+VERDICT r01 next-round item 2."""
+import os
+
+import pytest
+import torch
+import torch.multiprocessing as mp
+
+from horizonml_amd.models import resnet18
+from horizonml_amd.parallel.flat_reducer import (BackwardBucketScheduler,
+                                                 FlatBucketReducer,
+                                                 build_bucket_schedule,
+                                                 grad_units,
+                                                 partition_unit_sizes)
+
+
+def _fake_slices(model):
+    """id(param) -> (offset, numel) in model.parameters() order — the layout
+    FlatParamManager builds on GPU, reproduced without the extension."""
+    slices, off = {}, 0
+    for p in model.parameters():
+        slices[id(p)] = (off, p.numel())
+        off += p.numel()
+    return slices, off
+
+
+def test_partition_unit_sizes():
+    assert partition_unit_sizes([1, 1, 1, 1], 2) == [2, 2]
+    assert partition_unit_sizes([10, 1, 1], 3) == [1, 1, 1]
+    assert partition_unit_sizes([1, 1, 10], 2) == [2, 1]
+    assert sum(partition_unit_sizes(list(range(1, 12)), 4)) == 11
+    # never an empty bucket; clamps when more buckets than units
+    assert partition_unit_sizes([5, 5], 8) == [1, 1]
+
+
+def test_grad_units_tile_resnet18():
+    m = resnet18(num_classes=10)
+    units = grad_units(m)
+    # stem + 8 basic blocks + fc
+    assert len(units) == 10
+    unit_params = [id(p) for _, ps in units for p in ps]
+    assert unit_params == [id(p) for p in m.parameters()]
+
+
+@pytest.mark.parametrize("n_buckets", [1, 2, 4, 8])
+def test_build_bucket_schedule_resnet18(n_buckets):
+    m = resnet18(num_classes=10)
+    slices, total = _fake_slices(m)
+    ranges, mods = build_bucket_schedule(m, slices, n_buckets)
+    assert len(ranges) == len(mods) == min(n_buckets, 10)
+    # reverse-layer order: bucket 0 ends at the flat tail (ready first)
+    assert ranges[0][1] == total
+    assert ranges[-1][0] == 0
+    # disjoint exact tiling
+    assert sorted(ranges) == ranges[::-1]
+    cover = 0
+    for lo, hi in sorted(ranges):
+        assert lo == cover
+        cover = hi
+    assert cover == total
+    # fc (the model tail) readies bucket 0
+    assert m.tail.fc in mods[0]
+
+
+def test_scheduler_fires_on_last_unit():
+    m = resnet18(num_classes=10)
+    slices, total = _fake_slices(m)
+    grad = torch.randn(total)
+    ranges, mods = build_bucket_schedule(m, slices, 4)
+    fired = []
+    red = FlatBucketReducer(grad, ranges, comm_dtype=torch.float32)
+    red.reduce_bucket = lambda i: fired.append(i)  # spy
+    sched = BackwardBucketScheduler(red, mods)
+    # simulate backward: units complete in reverse forward order
+    for ms in mods:
+        for mm in ms:
+            mm._bwd_done_cb()
+    assert fired == [0, 1, 2, 3]
+    # second step: begin_step resets the counters
+    sched.begin_step()
+    fired.clear()
+    for ms in mods:
+        for mm in ms:
+            mm._bwd_done_cb()
+    assert fired == [0, 1, 2, 3]
+
+
+def test_reducer_validates_tiling():
+    g = torch.zeros(10)
+    with pytest.raises(ValueError):
+        FlatBucketReducer(g, [(0, 4), (5, 10)])  # gap
+    with pytest.raises(ValueError):
+        FlatBucketReducer(g, [(0, 4), (4, 9)])  # short
+    FlatBucketReducer(g, [(4, 10), (0, 4)], comm_dtype=torch.float32)  # ok
+
+
+def _bucket_equiv_worker(rank, world, port, q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    from horizonml_amd.runtime.distributed import (setup_distributed,
+                                                   teardown_distributed)
+    ctx = setup_distributed(rank, world, port, backend="gloo")
+    import torch.distributed as dist
+
+    g = torch.Generator().manual_seed(100 + rank)  # rank-distinct grads
+    grad = torch.randn(1000, generator=g)
+
+    # single-buffer reference: one all-reduce of the whole buffer
+    ref = grad.clone()
+    dist.all_reduce(ref)
+
+    # bucketed: 4 uneven slices in "reverse-layer" order, fired out of order
+    ranges = [(700, 1000), (450, 700), (100, 450), (0, 100)]
+    red = FlatBucketReducer(grad, ranges, comm_dtype=torch.float32)
+    red.begin_step()
+    for i in (0, 2, 1, 3):  # completion order need not match schedule order
+        red.reduce_bucket(i)
+    red.wait()
+
+    q.put((rank, torch.equal(red.comm, ref),
+           float((red.comm - ref).abs().max())))
+    teardown_distributed(ctx)
+
+
+def test_bucketed_equals_single_buffer_gloo():
+    mp_ctx = mp.get_context("spawn")
+    q = mp_ctx.Queue()
+    from horizonml_amd.utils.ports import find_free_port
+    port = find_free_port()
+    procs = [mp_ctx.Process(target=_bucket_equiv_worker, args=(r, 2, port, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=120) for _ in range(2)]
+    for p in procs:
+        p.join(timeout=60)
+    for rank, exact, maxdiff in results:
+        assert exact, (f"rank {rank}: bucketed != single-buffer "
+                       f"(max diff {maxdiff})")
