@@ -109,18 +109,44 @@ def main():
                 if use_comm else None)
     inv_world = 1.0 / world
 
+    # Overlapped bucketed all-reduce (default at world>1): the flat grad is
+    # reduced in reverse-layer buckets launched DURING backward (per-unit
+    # callbacks), each bucket's deferred wgrads flushed just before its
+    # pack, so xGMI communication overlaps the remaining backward + the
+    # next bucket's wgrad GEMMs.  HZ_BUCKETS=1 keeps the single-buffer
+    # post-backward fallback.
+    n_buckets = int(os.environ.get("HZ_BUCKETS", "4")) if use_comm else 1
+    sched = None
+    if use_comm and n_buckets > 1 and not args.infer:
+        from horizonml_amd.parallel.flat_reducer import (
+            BackwardBucketScheduler, FlatBucketReducer, build_bucket_schedule)
+        ranges, bucket_mods = build_bucket_schedule(model, mgr.slices,
+                                                    n_buckets)
+        reducer = FlatBucketReducer(
+            mgr.grad, ranges, comm_buf=comm_buf,
+            flush_range_fn=lambda lo, hi:
+                _ops.extension().flush_wgrad_range(mgr.grad, lo, hi))
+        sched = BackwardBucketScheduler(reducer, bucket_mods)
+
     def train_step(xb, yb):
         if args.infer:  # serving: eval-mode forward only
             with torch.no_grad():
                 logits = model(xb)
             return logits.float().sum()
+        if sched is not None:
+            sched.begin_step()
         logits = model(xb)
         loss = cross_entropy(logits, yb)
         loss.backward()
         if use_comm:
-            _ops.extension().flush_wgrad()    # wgrads complete before sync
-            comm_buf.copy_(mgr.grad)          # pack f32 -> bf16 (half bytes)
-            dist.all_reduce(comm_buf)         # RCCL over xGMI
+            if sched is not None:
+                # buckets launched from backward callbacks; fence them
+                sched.reducer.reduce_all()  # safety net: fire stragglers
+                sched.reducer.wait()
+            else:
+                _ops.extension().flush_wgrad()  # wgrads complete before sync
+                comm_buf.copy_(mgr.grad)        # pack f32 -> bf16
+                dist.all_reduce(comm_buf)       # RCCL over xGMI
             # fused optimizer consumes the reduced bf16 buffer directly
             opt.step(grad_bf16=comm_buf, grad_scale=inv_world)
         else:

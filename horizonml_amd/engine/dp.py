@@ -17,6 +17,7 @@ dominate the microsecond-scale steps.
 """
 from __future__ import annotations
 
+import os
 import time
 from typing import Optional
 
@@ -45,6 +46,251 @@ def _to_device(x, y, ctx: DistContext):
         x = x.to(memory_format=torch.channels_last).to(torch.bfloat16)
         y = y.to(ctx.device, non_blocking=True)
     return x, y
+
+
+def select_engine(engine: str, ctx: DistContext, model_name: str) -> str:
+    """``auto`` → ``flat`` (hipGraph fast path) on GPU for models the flat
+    manager supports, ``eager`` otherwise.  Explicit choices are honored
+    (``flat`` on CPU raises — it is a GPU-only path by design)."""
+    if engine == "auto":
+        if ctx.is_gpu and model_name.lower().startswith(
+                ("resnet", "mobilenet")):
+            return "flat"
+        return "eager"
+    if engine == "flat" and not ctx.is_gpu:
+        raise RuntimeError("--engine flat requires a GPU (hipGraph + HIP "
+                           "extension path); use --engine eager on CPU")
+    return engine
+
+
+def train_dp_flat(ctx: DistContext, epochs: int, sample_size: int,
+                  logs_dir: str, batch_size: int = 64,
+                  model_name: str = "resnet18", lr: float = 1e-3,
+                  optimizer_name: str = "adam",
+                  synthetic: Optional[bool] = None, data_dir: str = "./data",
+                  probe_divergence: bool = True, log_progress: bool = True,
+                  use_graph: bool = True):
+    """DP training on the flat fast path — the entrypoint-facing version of
+    the bench.py flagship step (the parity entrypoint
+    should run the best path we have).
+
+    * flat f32 master / bf16 shadow params (``FlatParamManager``), fused
+      Adam/SGD step, batched deferred wgrads;
+    * the device-side step (normalize → forward → CE → backward → wgrad
+      flush → pack → divergence probe [→ optimizer]) is captured in a
+      hipGraph once and replayed per step, with the per-step H2D batch copy
+      staged into static buffers outside the capture;
+    * world>1: the bf16 gradient all-reduce runs EAGERLY between two
+      captured graphs (fwd/bwd graph → RCCL all-reduce → optimizer graph).
+      Keeping the collective out of the capture makes rank-divergent
+      capture failures harmless (the collective sequence is identical in
+      graph and eager modes) and gives an honest ``comm_time`` CSV column —
+      the fully-captured single-graph variant (collective inside the graph,
+      bucket-overlapped) is bench.py's flagship configuration;
+    * loss/accuracy/divergence accumulate in static device scalars read
+      once per epoch (no per-step sync); CSV schema unchanged.
+
+    Capture runs 3 warmup steps on the first batch shape; parameter /
+    BN-buffer / optimizer state is snapshotted before and restored after,
+    so training still starts from the seeded init.  Ragged final batches
+    (shapes differing from the captured one) fall back to an eager step of
+    the same math.
+    """
+    from .. import ops as _ops
+    ext = _ops.extension()  # raises loudly if the HIP extension is missing
+    from ..data.cifar import normalize_uint8
+    from ..models._functional_gpu import cross_entropy
+    from .flat import FlatParamManager, HorizonAdam, HorizonSGD
+
+    rank, world = ctx.rank, ctx.world_size
+    dev = ctx.device
+    seed_everything(rank=0)  # identical replicas on every rank (no bcast)
+    loader, sampler = get_dataloader(rank, world, batch_size, sample_size,
+                                     strategy="dp", data_dir=data_dir,
+                                     synthetic=synthetic, raw=True)
+    model = build_model(model_name, num_classes=10).to(dev)
+    mgr = FlatParamManager(model, dev)
+    opt = (HorizonAdam(mgr, lr=lr) if optimizer_name == "adam"
+           else HorizonSGD(mgr, lr=lr))
+    import torch.distributed as dist
+    force_comm = (os.environ.get("HZ_FORCE_COMM") == "1" and world == 1
+                  and dist.is_initialized() and ctx.backend == "nccl")
+    use_comm = (world > 1 and ctx.backend == "nccl") or force_comm
+    comm_buf = (torch.zeros_like(mgr.grad, dtype=torch.bfloat16)
+                if use_comm else None)
+    inv_world = 1.0 / world
+
+    # static buffers + on-device accumulators
+    sx = torch.empty(batch_size, 3, 32, 32, dtype=torch.uint8, device=dev)
+    sy = torch.empty(batch_size, dtype=torch.int64, device=dev)
+    loss_acc = torch.zeros((), device=dev)
+    corr_acc = torch.zeros((), device=dev)
+    prev = torch.zeros_like(mgr.grad) if probe_divergence else None
+    sumsq = torch.zeros(1, device=dev) if probe_divergence else None
+    div_acc = torch.zeros(1, device=dev) if probe_divergence else None
+
+    def fb_step(x_u8, y):
+        """Capturable device-side step up to (not including) the optimizer:
+        returns loss; leaves complete grads in mgr.grad and, when
+        ``use_comm``, the bf16 pack in comm_buf."""
+        x = normalize_uint8(x_u8) \
+            .to(memory_format=torch.channels_last).to(torch.bfloat16)
+        logits = model(x)
+        loss = cross_entropy(logits, y)
+        loss.backward()
+        ext.flush_wgrad()  # batched deferred wgrads -> grads complete
+        if use_comm:
+            comm_buf.copy_(mgr.grad)  # pack f32 -> bf16 (half xGMI bytes)
+        if probe_divergence:
+            ext.grad_divergence(mgr.grad, prev, sumsq, div_acc, False)
+        loss_acc.add_(loss.detach() * y.shape[0])
+        corr_acc.add_((logits.detach().argmax(1) == y).sum())
+        return loss
+
+    def opt_step():
+        if use_comm:
+            opt.step(grad_bf16=comm_buf, grad_scale=inv_world)
+        else:
+            opt.step()
+
+    # ---- hipGraph capture (state snapshotted/restored around it) --------
+    graph_fb = graph_opt = None
+    if use_graph:
+        snap_master = mgr.master.detach().clone()
+        snap_bufs = [(b, b.detach().clone()) for b in model.buffers()]
+        g = torch.Generator(device="cpu").manual_seed(4321 + rank)
+        sx.copy_(torch.randint(0, 256, sx.shape, dtype=torch.uint8,
+                               generator=g))
+        sy.copy_(torch.randint(0, 10, sy.shape, generator=g))
+        try:
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                for _ in range(3):
+                    loss = fb_step(sx, sy)
+                    if use_comm:
+                        dist.all_reduce(comm_buf)
+                    opt_step()
+            torch.cuda.current_stream().wait_stream(side)
+            torch.cuda.synchronize()
+            del loss
+            graph_fb = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(graph_fb):
+                fb_step(sx, sy)
+            if use_comm:
+                graph_opt = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(graph_opt):
+                    opt_step()
+            else:  # world=1: one whole-step graph
+                graph_opt = None
+                graph_all = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(graph_all):
+                    fb_step(sx, sy)
+                    opt_step()
+                graph_fb = graph_all
+        except Exception as e:  # noqa: BLE001 — eager fallback is safe:
+            # the collective sequence is identical either way
+            print(f"[dp-flat r{rank}] graph capture failed ({e!r}); "
+                  "running eager flat steps", flush=True)
+            graph_fb = graph_opt = None
+        # restore seeded init state: capture warmups must not train
+        with torch.no_grad():
+            mgr.master.copy_(snap_master)
+            mgr.shadow.copy_(mgr.master)
+            for b, s in snap_bufs:
+                b.copy_(s)
+            mgr.grad.zero_()
+            mgr.stats_arena.zero_()
+            if isinstance(opt, HorizonAdam):
+                opt.m.zero_(), opt.v.zero_(), opt.step_t.zero_()
+            elif opt.mom is not None:
+                opt.mom.zero_()
+            loss_acc.zero_(), corr_acc.zero_()
+            if probe_divergence:
+                prev.zero_(), sumsq.zero_(), div_acc.zero_()
+        mgr.refresh_rsck()
+        del snap_master, snap_bufs
+        torch.cuda.synchronize()
+
+    prof = StepProfiler(dev)
+    writer = MetricsWriter(logs_dir, rank, sample_size,
+                           with_bandwidth=False, with_gpu=True)
+    import psutil
+    proc = psutil.Process()
+    proc.cpu_percent(interval=None)
+
+    count = 0
+    for epoch in range(epochs):
+        if sampler is not None:
+            sampler.set_epoch(epoch)
+        with prof.idle():
+            barrier(ctx)
+        epoch_start = time.time()
+        cpu_samples, mem_samples = [], []
+        n_steps = 0
+
+        for x, y in progress_iter(loader, f"dp-flat r{rank} e{epoch + 1}",
+                                  log_progress):
+            prof.step_begin()
+            cpu, mem = sample_host_resources(proc)
+            cpu_samples.append(cpu)
+            mem_samples.append(mem)
+            if graph_fb is not None and x.shape[0] == batch_size:
+                with prof.compute():
+                    sx.copy_(x, non_blocking=True)
+                    sy.copy_(y, non_blocking=True)
+                    graph_fb.replay()
+                if use_comm:
+                    with prof.comm():
+                        dist.all_reduce(comm_buf)
+                        prof.add_bytes(comm_buf.numel()
+                                       * comm_buf.element_size())
+                if graph_opt is not None:
+                    with prof.compute():
+                        graph_opt.replay()
+            else:  # eager step (capture failed, or ragged final batch)
+                xg = x.to(dev, non_blocking=True)
+                yg = y.to(dev, non_blocking=True)
+                with prof.compute():
+                    fb_step(xg, yg)
+                if use_comm:
+                    with prof.comm():
+                        dist.all_reduce(comm_buf)
+                        prof.add_bytes(comm_buf.numel()
+                                       * comm_buf.element_size())
+                with prof.compute():
+                    opt_step()
+            count += y.shape[0]
+            n_steps += 1
+            prof.step_end()
+
+        epoch_time = time.time() - epoch_start
+        loss_v = float(loss_acc.cpu()) / max(1, count)
+        acc_v = 100.0 * float(corr_acc.cpu()) / max(1, count)
+        div_v = (float(div_acc.cpu()) / max(1, n_steps)
+                 if probe_divergence else 0.0)
+        loss_acc.zero_(), corr_acc.zero_()
+        if probe_divergence:
+            div_acc.zero_()
+        count = 0
+        t = prof.epoch_end()
+        gmem, gutil = sample_gpu_resources(dev)
+        m = EpochMetrics(
+            epoch=epoch + 1, loss=loss_v, accuracy=acc_v,
+            epoch_time=epoch_time, avg_step_time=t["avg_step_time"],
+            compute_time=t["compute_time"], comm_time=t["comm_time"],
+            idle_time=t["idle_time"],
+            avg_cpu=sum(cpu_samples) / max(1, len(cpu_samples)),
+            avg_memory=sum(mem_samples) / max(1, len(mem_samples)),
+            grad_divergence=div_v, gpu_memory_mb=gmem, gpu_util=gutil)
+        writer.append(m)
+        if log_progress and rank == 0:
+            print(f"[dp-flat rank0] epoch {epoch + 1}/{epochs} "
+                  f"loss={loss_v:.4f} acc={acc_v:.2f}% "
+                  f"time={epoch_time:.2f}s", flush=True)
+        with prof.idle():
+            barrier(ctx)
+    return writer.path
 
 
 def train_dp(ctx: DistContext, epochs: int, sample_size: int, logs_dir: str,
@@ -140,16 +386,27 @@ def dp_worker(rank: int, world_size: int, epochs: int, sample_size: int,
               port: int, logs_dir: str, batch_size: int = 64,
               model_name: str = "resnet18", backend: Optional[str] = None,
               synthetic: Optional[bool] = None, lr: float = 1e-3,
-              optimizer_name: str = "adam"):
-    """Spawned worker entry (reference ``data_parallel_train.py:192-230``)."""
+              optimizer_name: str = "adam", engine: str = "auto"):
+    """Spawned worker entry (reference ``data_parallel_train.py:192-230``).
+
+    ``engine``: ``flat`` (default on GPU via ``auto``) runs the hipGraph /
+    fused-step fast path; ``eager`` the bucketed-DDP torch-optimizer loop."""
     ctx = setup_distributed(rank, world_size, port, backend=backend)
     try:
         if ctx.is_gpu and model_name.startswith("resnet"):
             # fail loudly if the native extension is missing on a GPU box
             from .. import ops as _ops
             _ops.extension()
-        train_dp(ctx, epochs, sample_size, logs_dir, batch_size=batch_size,
-                 model_name=model_name, synthetic=synthetic, lr=lr,
-                 optimizer_name=optimizer_name)
+        eng = select_engine(engine, ctx, model_name)
+        if eng == "flat":
+            train_dp_flat(ctx, epochs, sample_size, logs_dir,
+                          batch_size=batch_size, model_name=model_name,
+                          synthetic=synthetic, lr=lr,
+                          optimizer_name=optimizer_name)
+        else:
+            train_dp(ctx, epochs, sample_size, logs_dir,
+                     batch_size=batch_size, model_name=model_name,
+                     synthetic=synthetic, lr=lr,
+                     optimizer_name=optimizer_name)
     finally:
         teardown_distributed(ctx)

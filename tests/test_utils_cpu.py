@@ -67,6 +67,20 @@ def test_auto_backend_rules(monkeypatch, capsys):
     assert "falling back to gloo" in capsys.readouterr().out
 
 
+def test_select_engine_rules():
+    from horizonml_amd.engine.dp import select_engine
+    from horizonml_amd.runtime.distributed import DistContext
+    cpu = DistContext(0, 1, "gloo", None)
+    assert select_engine("auto", cpu, "resnet18") == "eager"
+    assert select_engine("eager", cpu, "resnet18") == "eager"
+    with pytest.raises(RuntimeError):
+        select_engine("flat", cpu, "resnet18")  # GPU-only path
+    gpu = DistContext(0, 1, "nccl", torch.device("cuda", 0))
+    assert select_engine("auto", gpu, "resnet18") == "flat"
+    assert select_engine("auto", gpu, "mobilenet_v2") == "flat"
+    assert select_engine("eager", gpu, "resnet18") == "eager"
+
+
 def test_shared_subset_deterministic_and_shared():
     a = shared_subset_indices(50000, 1000, seed=7)
     b = shared_subset_indices(50000, 1000, seed=7)
