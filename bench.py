@@ -22,12 +22,9 @@ standalone multi-GPU invocation re-execs torchrun itself.)
 from __future__ import annotations
 
 import argparse
-import json
 import os
 import sys
 import time
-
-BASELINE_IMAGES_PER_SEC = 1000.0 / 19.6  # DP, 1000 samples, 5 ep
 
 
 def parse_args():
@@ -189,10 +186,8 @@ def main():
             # all ranks must agree on the execution mode BEFORE any replay:
             # a rank-divergent capture failure would otherwise deadlock the
             # captured collectives (replay on one side only)
-            ok = torch.tensor([1.0 if graphs is not None else 0.0],
-                              device=dev)
-            dist.all_reduce(ok, op=dist.ReduceOp.MIN)
-            if float(ok) < 1.0:
+            from horizonml_amd.runtime.bench_protocol import agree_all_ranks
+            if not agree_all_ranks(graphs is not None, world, dev):
                 graphs = None
                 mode = "eager"
         if graphs is not None:
@@ -219,13 +214,10 @@ def main():
         dist.barrier(device_ids=[local_rank])
     t1 = time.perf_counter()
 
-    elapsed = torch.tensor([t1 - t0], device=dev)
-    if world > 1:
-        dist.all_reduce(elapsed, op=dist.ReduceOp.MAX)
-    elapsed_s = float(elapsed)
-    ms_per_step = elapsed_s / args.steps * 1000.0
-    global_batch = bs * world
-    ips = global_batch * args.steps / elapsed_s
+    from horizonml_amd.runtime.bench_protocol import (build_record,
+                                                      emit_record,
+                                                      max_elapsed_over_ranks)
+    elapsed_s = max_elapsed_over_ranks(t1 - t0, world, dev)
     if graphs is not None:
         final_loss = float(losses_static[-1].detach().float().cpu())
     else:
@@ -233,33 +225,14 @@ def main():
                            .detach().float().cpu())
 
     if rank == 0:
-        out = {
-            "metric": "images/sec",
-            "value": round(ips, 2),
-            "unit": "images/sec",
-            "n_gpus": world,
-            "steps": args.steps,
-            "warmup": args.warmup,
-            "ms_per_step": round(ms_per_step, 4),
-            "higher_is_better": True,
-            "scaling": "weak",
-            "vs_baseline": round(ips / BASELINE_IMAGES_PER_SEC, 2),
-            "dtype": "bf16",
-            "data": "synthetic",
-            "config": {
-                "model": args.model + ("_infer" if args.infer else
-                                       "_cifar10"),
-                "global_batch": global_batch,
-                "seq_len": None,
-                "image": "3x32x32",
-                "parallelism": f"dp{world}",
-                "optimizer": args.optimizer,
-                "exec": mode,
-                "epoch_time_s_1000_samples": round(1000.0 / ips, 6),
-                "final_loss": round(final_loss, 4),
-            },
-        }
-        print(json.dumps(out), flush=True)
+        out = build_record(
+            elapsed_s=elapsed_s, steps=args.steps, warmup=args.warmup,
+            world=world, batch_size=bs, model=args.model,
+            optimizer=args.optimizer, exec_mode=mode,
+            final_loss=final_loss, infer=args.infer,
+            comm_mode=(f"bucketed{n_buckets}" if sched is not None
+                       else ("flat1" if use_comm else "none")))
+        print(emit_record(out), flush=True)
     if world > 1 or force_comm:
         dist.destroy_process_group()
 
