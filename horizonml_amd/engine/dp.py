@@ -174,20 +174,21 @@ def train_dp_flat(ctx: DistContext, epochs: int, sample_size: int,
             torch.cuda.current_stream().wait_stream(side)
             torch.cuda.synchronize()
             del loss
-            graph_fb = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(graph_fb):
-                fb_step(sx, sy)
             if use_comm:
+                # collective stays OUTSIDE the captures: fwd/bwd graph,
+                # eager RCCL all-reduce, optimizer graph
+                graph_fb = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(graph_fb):
+                    fb_step(sx, sy)
                 graph_opt = torch.cuda.CUDAGraph()
                 with torch.cuda.graph(graph_opt):
                     opt_step()
             else:  # world=1: one whole-step graph
-                graph_opt = None
-                graph_all = torch.cuda.CUDAGraph()
-                with torch.cuda.graph(graph_all):
+                graph_fb = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(graph_fb):
                     fb_step(sx, sy)
                     opt_step()
-                graph_fb = graph_all
+                graph_opt = None
         except Exception as e:  # noqa: BLE001 — eager fallback is safe:
             # the collective sequence is identical either way
             print(f"[dp-flat r{rank}] graph capture failed ({e!r}); "
