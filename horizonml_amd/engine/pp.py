@@ -123,10 +123,14 @@ def train_pp(ctx: DistContext, epochs: int, sample_size: int, logs_dir: str,
             if optimizer is not None:
                 optimizer.zero_grad(set_to_none=True)
             with prof.compute():
+                # batch_hint: every rank holds the step's batch locally
+                # (same loader) -> static-shape negotiated relay, no
+                # per-hop header sync after the first step
                 total_loss, n, corr = stage.forward_backward(
                     x if stage.is_first else None,
                     y if stage.is_last else None,
-                    loss_fn=loss_fn, microbatches=microbatches)
+                    loss_fn=loss_fn, microbatches=microbatches,
+                    batch_hint=x.shape[0])
             if ddp is not None:
                 with prof.comm():
                     ddp.finalize_backward()
