@@ -60,101 +60,124 @@ def _epoch_rows(df, last_worker_only=False):
     return df
 
 
-def generate_comparison_graphs(results, output_dir, world_size, epochs):
-    import matplotlib
-    matplotlib.use("Agg")
-    import matplotlib.pyplot as plt
+CHART_FILES = ["accuracy_comparison.png", "loss_comparison.png",
+               "training_time_comparison.png",
+               "compute_vs_comm_comparison.png",
+               "cpu_utilization_comparison.png",
+               "memory_usage_comparison.png", "idle_time_comparison.png",
+               "overall_performance_comparison.png"]
 
-    sample_sizes = sorted({n for s in STRATEGIES for n in results[s]})
-    for n in sample_sizes:
-        out = os.path.join(output_dir, str(n),
-                           f"worker-{world_size}-epoch-{epochs}")
-        os.makedirs(out, exist_ok=True)
 
-        def save(fig, name):
-            fig.tight_layout()
-            fig.savefig(os.path.join(out, name), dpi=110)
-            plt.close(fig)
+def _emit_chart_set(results, sample_sizes, out, plt):
+    """Draw the reference's 8-figure inventory into ``out`` for the given
+    sample sizes (one line/bar group per strategy per size — the reference
+    draws EVERY size into one figure, ``main.py:69-117``)."""
+    os.makedirs(out, exist_ok=True)
 
-        # 1/2: accuracy + loss curves (MP: last worker only, like main.py:73)
-        for metric, fname in [("accuracy", "accuracy_comparison.png"),
-                              ("loss", "loss_comparison.png")]:
-            fig, ax = plt.subplots(figsize=(7, 4.5))
+    def save(fig, name):
+        fig.tight_layout()
+        fig.savefig(os.path.join(out, name), dpi=110)
+        plt.close(fig)
+
+    # 1/2: accuracy + loss curves, all sizes in one figure (MP: last
+    # worker only, like reference main.py:73)
+    for metric, fname in [("accuracy", "accuracy_comparison.png"),
+                          ("loss", "loss_comparison.png")]:
+        fig, ax = plt.subplots(figsize=(9, 6))
+        for n in sample_sizes:
             for s in STRATEGIES:
                 df = _epoch_rows(results[s].get(n),
                                  last_worker_only=(s == "model_parallel"))
                 if df is None:
                     continue
                 g = df.groupby("epoch")[metric].mean()
-                ax.plot(g.index, g.values, marker="o", label=LABELS[s],
-                        color=COLORS[s])
-            ax.set_xlabel("Epoch")
-            ax.set_ylabel(metric.capitalize())
-            ax.set_title(f"{metric.capitalize()} ({n} samples)")
-            ax.legend()
-            ax.grid(alpha=0.3)
-            save(fig, fname)
+                lbl = (LABELS[s] if len(sample_sizes) == 1
+                       else f"{LABELS[s]} ({n} samples)")
+                ax.plot(g.index, g.values, marker="o", label=lbl,
+                        color=COLORS[s],
+                        alpha=1.0 if n == sample_sizes[-1] else 0.55)
+        ax.set_xlabel("Epoch")
+        ax.set_ylabel("Accuracy (%)" if metric == "accuracy" else "Loss")
+        ax.set_title(f"{metric.capitalize()} Comparison: "
+                     "Data vs Model vs Tensor Parallel")
+        ax.legend()
+        ax.grid(True, alpha=0.3)
+        save(fig, fname)
 
-        # 3: avg epoch time bars
-        fig, ax = plt.subplots(figsize=(7, 4.5))
-        vals, names, cols = [], [], []
-        for s in STRATEGIES:
+    # 3: avg epoch time — grouped bars over sample sizes (main.py:119-149)
+    fig, ax = plt.subplots(figsize=(9, 6))
+    width = 0.25
+    x = np.arange(len(sample_sizes))
+    for i, s in enumerate(STRATEGIES):
+        vals = []
+        for n in sample_sizes:
             df = _epoch_rows(results[s].get(n))
-            if df is None:
-                continue
-            vals.append(df.groupby("epoch")["epoch_time"].mean().mean())
-            names.append(LABELS[s])
-            cols.append(COLORS[s])
-        ax.bar(names, vals, color=cols)
-        for i, v in enumerate(vals):
-            ax.text(i, v, f"{v:.2f}s", ha="center", va="bottom")
-        ax.set_ylabel("Avg epoch time (s)")
-        ax.set_title(f"Training time ({n} samples)")
-        save(fig, "training_time_comparison.png")
+            vals.append(0.0 if df is None
+                        else df.groupby("epoch")["epoch_time"].mean().mean())
+        ax.bar(x + (i - 1) * width, vals, width, label=LABELS[s],
+               color=COLORS[s])
+        for xi, v in zip(x + (i - 1) * width, vals):
+            if v > 0:
+                ax.text(xi, v, f"{v:.2f}s", ha="center", va="bottom",
+                        fontsize=8)
+    ax.set_xlabel("Sample Size")
+    ax.set_ylabel("Average Epoch Time (s)")
+    ax.set_title("Training Time Comparison")
+    ax.set_xticks(x, [str(n) for n in sample_sizes])
+    ax.legend()
+    save(fig, "training_time_comparison.png")
 
-        # 4-6: compute vs comm stacked, per strategy
-        for s in STRATEGIES:
+    # 4: compute vs comm — one stacked grouped figure (main.py:151-203)
+    fig, ax = plt.subplots(figsize=(12, 5))
+    for i, s in enumerate(STRATEGIES):
+        comp, comm = [], []
+        for n in sample_sizes:
             df = _epoch_rows(results[s].get(n))
-            if df is None:
-                continue
-            fig, ax = plt.subplots(figsize=(7, 4.5))
-            g = df.groupby("worker")[["compute_time", "comm_time"]].sum()
-            ax.bar(g.index.astype(str), g["compute_time"],
-                   label="Compute", color="#4c72b0")
-            ax.bar(g.index.astype(str), g["comm_time"],
-                   bottom=g["compute_time"], label="Comm", color="#dd8452")
-            ax.set_xlabel("Worker")
-            ax.set_ylabel("Time (s)")
-            ax.set_title(f"{LABELS[s]}: compute vs comm ({n} samples)")
-            ax.legend()
-            save(fig, f"compute_vs_comm_{s}.png")
+            comp.append(0.0 if df is None else df["compute_time"].sum())
+            comm.append(0.0 if df is None else df["comm_time"].sum())
+        xs = x + (i - 1) * width
+        ax.bar(xs, comp, width, label=f"{LABELS[s]} compute",
+               color=COLORS[s])
+        ax.bar(xs, comm, width, bottom=comp,
+               label=f"{LABELS[s]} comm", color=COLORS[s], alpha=0.45,
+               hatch="//")
+    ax.set_xlabel("Sample Size")
+    ax.set_ylabel("Cumulative Time (s)")
+    ax.set_title("Communication vs Computation Time")
+    ax.set_xticks(x, [str(n) for n in sample_sizes])
+    ax.legend(fontsize=8)
+    save(fig, "compute_vs_comm_comparison.png")
 
-        # 7: CPU utilization / 8: memory / 9: idle time
-        for metric, ylab, fname in [
-                ("avg_cpu", "CPU %", "cpu_utilization_comparison.png"),
-                ("avg_memory", "Memory (MB)", "memory_usage_comparison.png"),
-                ("idle_time", "Idle time (s)", "idle_time_comparison.png")]:
-            fig, ax = plt.subplots(figsize=(7, 4.5))
-            vals, names, cols = [], [], []
-            for s in STRATEGIES:
+    # 5-7: CPU utilization / memory / idle time (main.py:205-302)
+    for metric, ylab, fname in [
+            ("avg_cpu", "CPU Utilization (%)",
+             "cpu_utilization_comparison.png"),
+            ("avg_memory", "Memory Usage (MB)",
+             "memory_usage_comparison.png"),
+            ("idle_time", "Idle Time (s)", "idle_time_comparison.png")]:
+        fig, ax = plt.subplots(figsize=(9, 6))
+        for i, s in enumerate(STRATEGIES):
+            vals = []
+            for n in sample_sizes:
                 df = _epoch_rows(results[s].get(n))
-                if df is None:
-                    continue
-                vals.append(df[metric].mean())
-                names.append(LABELS[s])
-                cols.append(COLORS[s])
-            ax.bar(names, vals, color=cols)
-            ax.set_ylabel(ylab)
-            ax.set_title(f"{ylab} ({n} samples)")
-            save(fig, fname)
+                vals.append(0.0 if df is None else df[metric].mean())
+            ax.bar(x + (i - 1) * width, vals, width, label=LABELS[s],
+                   color=COLORS[s])
+        ax.set_xlabel("Sample Size")
+        ax.set_ylabel(ylab)
+        ax.set_title(f"{ylab} Comparison")
+        ax.set_xticks(x, [str(n) for n in sample_sizes])
+        ax.legend()
+        save(fig, fname)
 
-    # radar over 6 normalized metrics at the largest sample size (main.py:304)
+    # 8: overall performance radar, 6 normalized metrics at the largest
+    # size (main.py:304-388)
     n = sample_sizes[-1]
     metrics = ["accuracy", "epoch_time", "compute_time", "comm_time",
                "avg_memory", "idle_time"]
-    avail = [s for s in STRATEGIES if _epoch_rows(results[s].get(n)) is not None]
+    avail = [s for s in STRATEGIES
+             if _epoch_rows(results[s].get(n)) is not None]
     if len(avail) >= 2:
-        import matplotlib.pyplot as plt
         table = {}
         for s in avail:
             df = _epoch_rows(results[s][n],
@@ -164,8 +187,8 @@ def generate_comparison_graphs(results, output_dir, world_size, epochs):
                         df["compute_time"].sum(), df["comm_time"].sum(),
                         df["avg_memory"].mean(), df["idle_time"].sum()]
         arr = np.array([table[s] for s in avail], dtype=float)
-        # normalize each metric to [0,1]; time-like metrics inverted (lower
-        # is better)
+        # normalize each metric to [0,1]; time-like metrics inverted
+        # (lower is better)
         norm = np.zeros_like(arr)
         for j in range(arr.shape[1]):
             col = arr[:, j]
@@ -184,14 +207,43 @@ def generate_comparison_graphs(results, output_dir, world_size, epochs):
             ax.fill(angles, vals, alpha=0.12, color=COLORS[s])
         ax.set_xticks(angles[:-1])
         ax.set_xticklabels(metrics)
-        ax.set_title(f"Normalized comparison ({n} samples)")
+        ax.set_title(f"Overall Performance ({n} samples)")
         ax.legend(loc="lower right", bbox_to_anchor=(1.2, -0.1))
-        out = os.path.join(output_dir, str(n),
-                           f"worker-{world_size}-epoch-{epochs}")
-        os.makedirs(out, exist_ok=True)
-        fig.savefig(os.path.join(out, "radar_comparison.png"), dpi=110,
-                    bbox_inches="tight")
+        fig.savefig(os.path.join(out, "overall_performance_comparison.png"),
+                    dpi=110, bbox_inches="tight")
         plt.close(fig)
+
+
+def chart_set_name(results, n) -> str:
+    """Published-tree set level (``benchmark_results/{all|datavlayer}/``):
+    ``all`` when every strategy produced results, ``datavlayer`` when only
+    DP + layer-MP did (the reference's TP-failed runs)."""
+    if _epoch_rows(results["tensor_parallel"].get(n)) is None and \
+            _epoch_rows(results["data_parallel"].get(n)) is not None:
+        return "datavlayer"
+    return "all"
+
+
+def generate_comparison_graphs(results, output_dir, world_size, epochs):
+    """Reference figure inventory (``main.py:64-390``: 8 figures, every
+    sample size drawn into each) at the top of ``output_dir``, plus the
+    published per-run tree ``{output_dir}/{set}/{n}/worker-{ws}-epoch-{e}/``
+    with the same 8 figures per sample size
+    (``/root/reference/benchmark_results/``)."""
+    import matplotlib
+    matplotlib.use("Agg")
+    import matplotlib.pyplot as plt
+
+    sample_sizes = sorted({n for s in STRATEGIES for n in results[s]})
+    if not sample_sizes:
+        return
+    # top level: the exact output of the reference's flat generate call
+    _emit_chart_set(results, sample_sizes, output_dir, plt)
+    # published layout: one run dir per sample size under its set
+    for n in sample_sizes:
+        out = os.path.join(output_dir, chart_set_name(results, n), str(n),
+                           f"worker-{world_size}-epoch-{epochs}")
+        _emit_chart_set(results, [n], out, plt)
 
 
 def main():

@@ -25,17 +25,29 @@ def test_main_benchmark_and_charts(tmp_path, monkeypatch):
         assert "total_training_time" in df.columns
     out = tmp_path / "charts"
     generate_comparison_graphs(results, str(out), 2, 1)
-    chart_dir = out / "64" / "worker-2-epoch-1"
-    expected = ["accuracy_comparison.png", "loss_comparison.png",
-                "training_time_comparison.png",
-                "compute_vs_comm_data_parallel.png",
-                "compute_vs_comm_model_parallel.png",
-                "compute_vs_comm_tensor_parallel.png",
-                "cpu_utilization_comparison.png",
-                "memory_usage_comparison.png", "idle_time_comparison.png",
-                "radar_comparison.png"]
-    for name in expected:
+    from main import CHART_FILES
+    # reference inventory (main.py:64-390): 8 figures at the top level ...
+    for name in CHART_FILES:
+        assert (out / name).is_file(), f"missing top-level chart {name}"
+    # ... plus the published per-run tree
+    # benchmark_results/{set}/{n}/worker-{ws}-epoch-{e}/ with the same 8
+    chart_dir = out / "all" / "64" / "worker-2-epoch-1"
+    for name in CHART_FILES:
         assert (chart_dir / name).is_file(), f"missing chart {name}"
+
+
+def test_chart_set_name_datavlayer():
+    """TP-failed runs land under the datavlayer set, like the reference's
+    published tree."""
+    import pandas as pd
+
+    from main import chart_set_name
+    df = pd.DataFrame({"epoch": [1], "worker": [0], "accuracy": [1.0]})
+    results = {"data_parallel": {64: df}, "model_parallel": {64: df},
+               "tensor_parallel": {64: None}}
+    assert chart_set_name(results, 64) == "datavlayer"
+    results["tensor_parallel"][64] = df
+    assert chart_set_name(results, 64) == "all"
 
 
 @pytest.mark.timeout(300)
