@@ -135,7 +135,15 @@ class RawView(Dataset):
 
 def normalize_uint8(x: torch.Tensor) -> torch.Tensor:
     """The reference transform ((0.5,0.5,0.5) mean/std) for raw uint8
-    batches, applied on whatever device x lives on."""
+    batches, applied on whatever device x lives on.
+
+    GPU: ONE fused kernel producing the bf16 channels_last layout the
+    gfx950 conv kernels consume (the eager to(f32)/div/sub/div + permute +
+    cast chain costs ~6 dispatches per step); callers' subsequent
+    ``.to(channels_last).to(bf16)`` become no-ops.  CPU keeps f32."""
+    if x.is_cuda and x.dim() == 4 and x.dtype == torch.uint8:
+        from .. import ops as _ops
+        return _ops.extension().normalize_u8(x.contiguous(), _MEAN, _STD)
     return x.to(torch.float32).div_(255.0).sub_(_MEAN).div_(_STD)
 
 

@@ -80,6 +80,8 @@ void launch_permute_krsc_rsck(const void*, void*, const int*, int, int,
                               hipStream_t);
 void launch_grad_divergence(const float*, float*, float*, float*, long, int,
                             hipStream_t);
+void launch_normalize_u8(const void*, void*, long, int, long, float, float,
+                         hipStream_t);
 }
 
 namespace {
@@ -848,6 +850,24 @@ void grad_divergence(Tensor g, Tensor prev, Tensor sumsq, Tensor out,
                          g.numel(), skip_first ? 1 : 0, cur_stream());
 }
 
+// uint8 NCHW batch -> normalized bf16 channels_last, one kernel (the
+// engines' H2D preprocessing; reference transform (0.5,0.5,0.5) mean/std,
+// data_parallel_train.py:44-47).
+Tensor normalize_u8(Tensor x, double mean, double std) {
+  TORCH_CHECK(x.dim() == 4 && x.dtype() == torch::kUInt8 && x.is_cuda()
+                  && x.is_contiguous(),
+              "normalize_u8 expects contiguous NCHW uint8 on GPU");
+  const long N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  auto y = at::empty({N, C, H, W},
+                     x.options().dtype(torch::kBFloat16),
+                     at::MemoryFormat::ChannelsLast);
+  const float scale = 1.0f / (255.0f * (float)std);
+  const float shift = -(float)mean / (float)std;
+  launch_normalize_u8(x.data_ptr(), y.data_ptr(), N * C * H * W, (int)C,
+                      H * W, scale, shift, cur_stream());
+  return y;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -867,6 +887,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sgd_step", &sgd_step);
   m.def("permute_krsc_rsck", &permute_krsc_rsck);
   m.def("grad_divergence", &grad_divergence);
+  m.def("normalize_u8", &normalize_u8);
   m.def("set_deterministic", &set_deterministic);
   m.def("deterministic_enabled", &deterministic_enabled);
   m.def("set_wgrad_defer", &set_wgrad_defer);

@@ -11,8 +11,11 @@ counters" evidence.
 
 Usage: python scripts/gpu_mfma_bench.py [bs] [image] [steps] [model]
 """
+import os
 import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 

@@ -229,6 +229,33 @@ def test_grad_divergence_kernel():
     assert abs(out.item() - ref) / ref < 1e-4
 
 
+def test_grad_divergence_kernel_odd_size():
+    """n % 4 != 0 exercises the scalar tail of the float4 main body."""
+    n = 5003
+    g1 = torch.randn(n, device="cuda")
+    g2 = torch.randn(n, device="cuda")
+    prev = g1.clone()
+    sumsq = torch.zeros(1, device="cuda")
+    out = torch.zeros(1, device="cuda")
+    _C().grad_divergence(g2, prev, sumsq, out, False)
+    ref = (g2 - g1).norm().item()
+    assert abs(out.item() - ref) / ref < 1e-4
+    assert torch.equal(prev, g2), "prev must be updated to g"
+
+
+def test_normalize_u8_kernel():
+    """Fused u8 NCHW -> bf16 channels_last normalize vs the torch chain."""
+    torch.manual_seed(0)
+    for shape in [(4, 3, 32, 32), (2, 3, 17, 9), (1, 3, 224, 224)]:
+        x = torch.randint(0, 256, shape, dtype=torch.uint8, device="cuda")
+        y = _C().normalize_u8(x, 0.5, 0.5)
+        assert y.dtype == torch.bfloat16
+        assert y.is_contiguous(memory_format=torch.channels_last)
+        ref = (x.float() / 255.0 - 0.5) / 0.5
+        ref = ref.to(memory_format=torch.channels_last).to(torch.bfloat16)
+        assert torch.equal(y, ref), f"mismatch at {shape}"
+
+
 def test_permute_krsc_rsck():
     torch.manual_seed(0)
     K, R, S, C = 32, 3, 3, 16
