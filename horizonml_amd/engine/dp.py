@@ -129,6 +129,11 @@ def train_dp_flat(ctx: DistContext, epochs: int, sample_size: int,
     sumsq = torch.zeros(1, device=dev) if probe_divergence else None
     div_acc = torch.zeros(1, device=dev) if probe_divergence else None
 
+    # Adam: divergence probe fused into the optimizer kernel (it already
+    # streams the f32 grad); SGD keeps the standalone gdiv kernel.
+    fused_probe = probe_divergence and isinstance(opt, HorizonAdam)
+    probe_bufs = (prev, sumsq, div_acc) if fused_probe else None
+
     def fb_step(x_u8, y):
         """Capturable device-side step up to (not including) the optimizer:
         returns loss; leaves complete grads in mgr.grad and, when
@@ -141,7 +146,7 @@ def train_dp_flat(ctx: DistContext, epochs: int, sample_size: int,
         ext.flush_wgrad()  # batched deferred wgrads -> grads complete
         if use_comm:
             comm_buf.copy_(mgr.grad)  # pack f32 -> bf16 (half xGMI bytes)
-        if probe_divergence:
+        if probe_divergence and not fused_probe:
             ext.grad_divergence(mgr.grad, prev, sumsq, div_acc, False)
         loss_acc.add_(loss.detach() * y.shape[0])
         corr_acc.add_((logits.detach().argmax(1) == y).sum())
@@ -149,9 +154,10 @@ def train_dp_flat(ctx: DistContext, epochs: int, sample_size: int,
 
     def opt_step():
         if use_comm:
-            opt.step(grad_bf16=comm_buf, grad_scale=inv_world)
+            opt.step(grad_bf16=comm_buf, grad_scale=inv_world,
+                     probe=probe_bufs)
         else:
-            opt.step()
+            opt.step(probe=probe_bufs)
 
     # ---- hipGraph capture (state snapshotted/restored around it) --------
     graph_fb = graph_opt = None

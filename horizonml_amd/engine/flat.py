@@ -109,16 +109,20 @@ class HorizonAdam:
         self.step_t = torch.zeros(1, device=dev)
 
     def step(self, zero_grad: bool = True, grad_bf16=None,
-             grad_scale: float = 1.0):
+             grad_scale: float = 1.0, probe=None):
         """``grad_bf16``: consume an all-reduced bf16 gradient buffer
-        directly (× ``grad_scale``) — skips the DP unpack pass."""
+        directly (× ``grad_scale``) — skips the DP unpack pass.
+        ``probe``: optional ``(prev, sumsq, out)`` f32 tensors — the
+        grad-divergence probe fused into the Adam pass (the standalone
+        kernel costs a 3×|grad| extra sweep per step)."""
         _ops.extension().flush_wgrad()  # batched deferred weight grads
+        p0, p1, p2 = probe if probe is not None else (None, None, None)
         _ops.extension().adam_step(self.mgr.master, self.mgr.grad, self.m,
                                    self.v, self.mgr.shadow, self.step_t,
                                    self.lr, self.betas[0], self.betas[1],
                                    self.eps, self.wd, zero_grad,
                                    self.mgr.stats_arena, grad_bf16,
-                                   grad_scale)
+                                   grad_scale, p0, p1, p2)
         self.mgr.refresh_rsck()
 
     # -- checkpoint round-trip (utils/checkpoint.py) ----------------------
@@ -145,7 +149,10 @@ class HorizonSGD:
         self.mom = (torch.zeros_like(mgr.master) if momentum > 0 else None)
 
     def step(self, zero_grad: bool = True, grad_bf16=None,
-             grad_scale: float = 1.0):
+             grad_scale: float = 1.0, probe=None):
+        if probe is not None:
+            raise ValueError("fused divergence probe is Adam-only; use the "
+                             "standalone grad_divergence kernel with SGD")
         _ops.extension().flush_wgrad()  # batched deferred weight grads
         _ops.extension().sgd_step(self.mgr.master, self.mgr.grad, self.mom,
                                   self.mgr.shadow, self.lr, self.mu, self.wd,

@@ -73,7 +73,8 @@ void launch_ce_fwd_bwd(const float*, const long*, float*, float*, int, int,
                        hipStream_t);
 void launch_adam_step(float*, float*, float*, float*, void*, const float*,
                       long, float, float, float, float, float, int, float*,
-                      long, const void*, float, hipStream_t);
+                      long, const void*, float, float*, float*, float*,
+                      hipStream_t);
 void launch_sgd_step(float*, float*, float*, void*, long, float, float, float,
                      int, const void*, float, hipStream_t);
 void launch_permute_krsc_rsck(const void*, void*, const int*, int, int,
@@ -802,12 +803,20 @@ Tensor wgrad_only(Tensor x, Tensor dz, int64_t K, int64_t R, int64_t S,
 }
 
 // ------------------------------------------------------------- optimizers --
+// probe_prev/probe_sumsq/probe_out (all-or-none): fused grad-divergence
+// probe — Σ(g−prev)² accumulated in-kernel, sqrt appended to probe_out.
 void adam_step(Tensor master, Tensor grad, Tensor m, Tensor v,
                c10::optional<Tensor> shadow, Tensor step_t, double lr,
                double b1, double b2, double eps, double wd, bool zero_grad,
                c10::optional<Tensor> extra_zero,
-               c10::optional<Tensor> grad_bf16, double grad_scale) {
+               c10::optional<Tensor> grad_bf16, double grad_scale,
+               c10::optional<Tensor> probe_prev,
+               c10::optional<Tensor> probe_sumsq,
+               c10::optional<Tensor> probe_out) {
   check_f32(master, "master");
+  TORCH_CHECK(probe_prev.has_value() == probe_sumsq.has_value()
+                  && probe_prev.has_value() == probe_out.has_value(),
+              "probe tensors must be passed together");
   launch_adam_step(master.data_ptr<float>(), grad.data_ptr<float>(),
                    m.data_ptr<float>(), v.data_ptr<float>(),
                    shadow.has_value() ? shadow->data_ptr() : nullptr,
@@ -818,7 +827,14 @@ void adam_step(Tensor master, Tensor grad, Tensor m, Tensor v,
                                           : nullptr,
                    extra_zero.has_value() ? extra_zero->numel() : 0,
                    grad_bf16.has_value() ? grad_bf16->data_ptr() : nullptr,
-                   (float)grad_scale, cur_stream());
+                   (float)grad_scale,
+                   probe_prev.has_value() ? probe_prev->data_ptr<float>()
+                                          : nullptr,
+                   probe_sumsq.has_value() ? probe_sumsq->data_ptr<float>()
+                                           : nullptr,
+                   probe_out.has_value() ? probe_out->data_ptr<float>()
+                                         : nullptr,
+                   cur_stream());
 }
 
 void sgd_step(Tensor master, Tensor grad, c10::optional<Tensor> mom,
@@ -861,10 +877,8 @@ Tensor normalize_u8(Tensor x, double mean, double std) {
   auto y = at::empty({N, C, H, W},
                      x.options().dtype(torch::kBFloat16),
                      at::MemoryFormat::ChannelsLast);
-  const float scale = 1.0f / (255.0f * (float)std);
-  const float shift = -(float)mean / (float)std;
   launch_normalize_u8(x.data_ptr(), y.data_ptr(), N * C * H * W, (int)C,
-                      H * W, scale, shift, cur_stream());
+                      H * W, (float)mean, (float)std, cur_stream());
   return y;
 }
 

@@ -857,20 +857,34 @@ __global__ __launch_bounds__(256) void k_ce_fwd_bwd(
 // hipGraph-replayable (bias correction computed on device).
 // gsrc (optional): all-reduced bf16 gradient consumed directly with a
 // 1/world scale — skips the DP path's unpack copy+mul pass.
+// Optional fused divergence probe (prev/sumsq non-null): the kernel already
+// streams the f32 gradient, so Σ(g−g_prev)² + prev←g ride along for one
+// extra read+write of prev instead of a standalone 3×45 MB gdiv pass
+// (measured ~61 µs/step standalone vs ~13 µs marginal here; engine-flat
+// trace r02).  The probe always reads the LOCAL f32 grad (pre-average),
+// matching the reference's per-worker probe semantics.
 __global__ __launch_bounds__(256) void k_adam_step(
     float* __restrict__ master, float* __restrict__ grad,
     float* __restrict__ m, float* __restrict__ v, bf16* __restrict__ shadow,
     const float* __restrict__ step_t, long n, float lr, float b1, float b2,
     float eps, float wd, int zero_grad, float* __restrict__ extra_zero,
-    long n_extra, const bf16* __restrict__ gsrc, float gscale) {
+    long n_extra, const bf16* __restrict__ gsrc, float gscale,
+    float* __restrict__ prev, float* __restrict__ sumsq) {
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n_extra;
        i += (long)gridDim.x * blockDim.x)
     extra_zero[i] = 0.f;
   float t = step_t[0];
   float bc1 = 1.f - __powf(b1, t), bc2 = 1.f - __powf(b2, t);
+  float acc = 0.f;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (long)gridDim.x * blockDim.x) {
-    float g = gsrc != nullptr ? b2f(gsrc[i]) * gscale : grad[i];
+    float gl = grad[i];
+    float g = gsrc != nullptr ? b2f(gsrc[i]) * gscale : gl;
+    if (prev != nullptr) {
+      float d = gl - prev[i];
+      prev[i] = gl;
+      acc += d * d;
+    }
     float w = master[i];
     if (wd != 0.f) g += wd * w;
     float mi = b1 * m[i] + (1.f - b1) * g;
@@ -881,6 +895,14 @@ __global__ __launch_bounds__(256) void k_adam_step(
     master[i] = w;
     if (shadow != nullptr) shadow[i] = f2b(w);
     if (zero_grad) grad[i] = 0.f;
+  }
+  if (prev != nullptr) {
+    acc = wave_reduce_sum(acc);
+    __shared__ float ws[4];
+    if ((threadIdx.x & 63) == 0) ws[threadIdx.x >> 6] = acc;
+    __syncthreads();
+    if (threadIdx.x == 0)
+      atomicAdd(sumsq, ws[0] + ws[1] + ws[2] + ws[3]);
   }
 }
 
@@ -988,13 +1010,16 @@ __global__ __launch_bounds__(256) void k_gdiv_partial(
 // dispatches per step in the training engines' H2D path).
 __global__ __launch_bounds__(256) void k_normalize_u8(
     const unsigned char* __restrict__ x, bf16* __restrict__ y, long total,
-    int C, long HW, float scale, float shift) {
+    int C, long HW, float mean, float std) {
   for (long o = (long)blockIdx.x * blockDim.x + threadIdx.x; o < total;
        o += (long)gridDim.x * blockDim.x) {
     const int c = (int)(o % C);
     const long p = o / C;          // pixel index: n*HW + hw
     const long nidx = p / HW, hw = p % HW;
-    const float v = (float)x[(nidx * C + c) * HW + hw] * scale + shift;
+    // same op order as the reference transform (div, sub, div) so results
+    // match the torch chain to the final bf16 rounding
+    const float v = ((float)x[(nidx * C + c) * HW + hw] / 255.0f - mean)
+                    / std;
     y[o] = (bf16)v;
   }
 }
@@ -1213,12 +1238,16 @@ void launch_adam_step(float* master, float* grad, float* m, float* v,
                       void* shadow, const float* step_t, long n, float lr,
                       float b1, float b2, float eps, float wd, int zero_grad,
                       float* extra_zero, long n_extra, const void* gsrc,
-                      float gscale, hipStream_t st) {
+                      float gscale, float* prev, float* sumsq, float* divout,
+                      hipStream_t st) {
   k_inc_step<<<1, 1, 0, st>>>((float*)step_t);
   k_adam_step<<<gsz(n), 256, 0, st>>>(master, grad, m, v, (bf16*)shadow,
                                       step_t, n, lr, b1, b2, eps, wd,
                                       zero_grad, extra_zero, n_extra,
-                                      (const bf16*)gsrc, gscale);
+                                      (const bf16*)gsrc, gscale, prev,
+                                      sumsq);
+  if (prev != nullptr)
+    k_gdiv_finalize<<<1, 1, 0, st>>>(sumsq, divout, 0);
 }
 
 void launch_sgd_step(float* master, float* grad, float* mom, void* shadow,
@@ -1244,10 +1273,10 @@ void launch_grad_divergence(const float* g, float* prev, float* sumsq,
 }
 
 void launch_normalize_u8(const void* x, void* y, long total, int C, long HW,
-                         float scale, float shift, hipStream_t st) {
+                         float mean, float std, hipStream_t st) {
   k_normalize_u8<<<gsz(total), 256, 0, st>>>((const unsigned char*)x,
-                                             (bf16*)y, total, C, HW, scale,
-                                             shift);
+                                             (bf16*)y, total, C, HW, mean,
+                                             std);
 }
 
 }  // extern "C"

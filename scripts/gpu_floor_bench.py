@@ -11,7 +11,8 @@ m = torch.zeros_like(master); v = torch.zeros_like(master)
 def burst(n):
     for _ in range(n):
         C.adam_step(master, grad, m, v, None, t, 1e-3, 0.9, 0.999, 1e-8,
-                    0.0, False, None, None, 1.0)  # k_inc_step + k_adam (4096)
+                    0.0, False, None, None, 1.0,
+                    None, None, None)  # k_inc_step + k_adam (4096)
 
 for _ in range(3):
     burst(50)
