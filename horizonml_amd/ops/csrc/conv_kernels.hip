@@ -184,7 +184,18 @@ DEV V8 load8_b(const bf16* __restrict__ src, const ConvP& p,
 // SPLIT: each blockIdx.z writes its f32 partial into its own slab
 // ws_out[z][M][Ntot] (no atomics, no pre-zeroing — the consumer pass sums
 // the slabs); otherwise write bf16 into Y (+ optional stats).
-template <int MODE, bool VECA, bool VECB, bool STATS, bool SPLIT>
+//
+// TM×TN block tile (waves in a fixed 2×2 grid, wave tile (TM/2)×(TN/2)):
+//  * 64×64 (default) — the LATENCY tile for the CIFAR-shape grids: small
+//    LDS footprint, shortest prologue, best when the grid underfills the
+//    256 CUs and every block must hide HBM latency alone;
+//  * 128×64 / 128×128 — THROUGHPUT tiles for ImageNet-shaped work
+//    (selected by pick_tile when the tiled grid still fills the chip):
+//    each wave runs 8/16 MFMAs per K-step against the same 6/8 LDS
+//    fragment reads, lifting the MFMA:issue ratio that caps the 64×64
+//    tile at ~2% of the bf16 peak on ResNet50@224 (r50_224_pmc_mfma.txt).
+template <int MODE, bool VECA, bool VECB, bool STATS, bool SPLIT,
+          int TM = 64, int TN = 64>
 __global__ __launch_bounds__(256) void k_conv_mfma(
     const bf16* __restrict__ A, const bf16* __restrict__ Bw,
     bf16* __restrict__ Y, float* __restrict__ ws_out,
@@ -192,12 +203,16 @@ __global__ __launch_bounds__(256) void k_conv_mfma(
     int accum) {
   // Double-buffered LDS, ONE barrier per K-step, 2-deep register
   // prefetch: tile k+2's global loads are in flight while tile k computes,
-  // so the ~900-cycle HBM latency is fully hidden even at 1 block/CU
-  // (these conv grids are small — latency, not bandwidth, bounds them).
-  __shared__ bf16 As[2][64 * LDA];
-  __shared__ bf16 Bs[2][64 * LDA];
+  // so the ~900-cycle HBM latency hides even at 1 block/CU.
+  constexpr int AC = TM / 64;   // A stage chunks per thread (vec8 each)
+  constexpr int BC = TN / 64;
+  constexpr int MI = TM / 32;   // 16-row m fragments per wave
+  constexpr int NI = TN / 32;
+  constexpr int WM = TM / 2, WN = TN / 2;
+  __shared__ bf16 As[2][TM * LDA];
+  __shared__ bf16 Bs[2][TN * LDA];
 
-  const int m0 = blockIdx.y * 64, n0 = blockIdx.x * 64;
+  const int m0 = blockIdx.y * TM, n0 = blockIdx.x * TN;
   const int kbeg = SPLIT ? blockIdx.z * kchunk : 0;
   const int kend = SPLIT ? min(p.Kd, kbeg + kchunk) : p.Kd;
   const int tid = threadIdx.x;
@@ -206,19 +221,30 @@ __global__ __launch_bounds__(256) void k_conv_mfma(
   const int wr = wave >> 1, wc = wave & 1;
   const int fr = lane & 15, fk = lane >> 4;
 
-  f32x4 acc[2][2] = {};
+  f32x4 acc[MI][NI] = {};
 
   // prologue: tile 0 -> LDS[0]; tile 1 -> registers
-  {
-    V8 a0 = load8_a<MODE, VECA>(A, p, mg, m0 + srow, kbeg + scol);
-    V8 b0 = load8_b<MODE, VECB>(Bw, p, mg, n0 + srow, kbeg + scol, Ntot);
-    *(V8*)&As[0][srow * LDA + scol] = a0;
-    *(V8*)&Bs[0][srow * LDA + scol] = b0;
+#pragma unroll
+  for (int j = 0; j < AC; j++) {
+    V8 a0 = load8_a<MODE, VECA>(A, p, mg, m0 + j * 64 + srow, kbeg + scol);
+    *(V8*)&As[0][(j * 64 + srow) * LDA + scol] = a0;
   }
-  V8 a_nx, b_nx;
+#pragma unroll
+  for (int j = 0; j < BC; j++) {
+    V8 b0 = load8_b<MODE, VECB>(Bw, p, mg, n0 + j * 64 + srow, kbeg + scol,
+                                Ntot);
+    *(V8*)&Bs[0][(j * 64 + srow) * LDA + scol] = b0;
+  }
+  V8 a_nx[AC], b_nx[BC];
   if (kbeg + 32 < kend) {
-    a_nx = load8_a<MODE, VECA>(A, p, mg, m0 + srow, kbeg + 32 + scol);
-    b_nx = load8_b<MODE, VECB>(Bw, p, mg, n0 + srow, kbeg + 32 + scol, Ntot);
+#pragma unroll
+    for (int j = 0; j < AC; j++)
+      a_nx[j] = load8_a<MODE, VECA>(A, p, mg, m0 + j * 64 + srow,
+                                    kbeg + 32 + scol);
+#pragma unroll
+    for (int j = 0; j < BC; j++)
+      b_nx[j] = load8_b<MODE, VECB>(Bw, p, mg, n0 + j * 64 + srow,
+                                    kbeg + 32 + scol, Ntot);
   }
   __syncthreads();
 
@@ -227,27 +253,36 @@ __global__ __launch_bounds__(256) void k_conv_mfma(
     // stage tile k+1 from registers into the OTHER buffer (its readers
     // synchronized at the previous barrier), then issue tile k+2's loads
     if (k0 + 32 < kend) {
-      *(V8*)&As[buf ^ 1][srow * LDA + scol] = a_nx;
-      *(V8*)&Bs[buf ^ 1][srow * LDA + scol] = b_nx;
+#pragma unroll
+      for (int j = 0; j < AC; j++)
+        *(V8*)&As[buf ^ 1][(j * 64 + srow) * LDA + scol] = a_nx[j];
+#pragma unroll
+      for (int j = 0; j < BC; j++)
+        *(V8*)&Bs[buf ^ 1][(j * 64 + srow) * LDA + scol] = b_nx[j];
       if (k0 + 64 < kend) {
-        a_nx = load8_a<MODE, VECA>(A, p, mg, m0 + srow, k0 + 64 + scol);
-        b_nx = load8_b<MODE, VECB>(Bw, p, mg, n0 + srow, k0 + 64 + scol,
-                                   Ntot);
+#pragma unroll
+        for (int j = 0; j < AC; j++)
+          a_nx[j] = load8_a<MODE, VECA>(A, p, mg, m0 + j * 64 + srow,
+                                        k0 + 64 + scol);
+#pragma unroll
+        for (int j = 0; j < BC; j++)
+          b_nx[j] = load8_b<MODE, VECB>(Bw, p, mg, n0 + j * 64 + srow,
+                                        k0 + 64 + scol, Ntot);
       }
     }
-    bf16x8 af[2], bf[2];
+    bf16x8 af[MI], bf[NI];
 #pragma unroll
-    for (int mi = 0; mi < 2; mi++)
+    for (int mi = 0; mi < MI; mi++)
       af[mi] =
-          *(const bf16x8*)&As[buf][(wr * 32 + mi * 16 + fr) * LDA + fk * 8];
+          *(const bf16x8*)&As[buf][(wr * WM + mi * 16 + fr) * LDA + fk * 8];
 #pragma unroll
-    for (int ni = 0; ni < 2; ni++)
+    for (int ni = 0; ni < NI; ni++)
       bf[ni] =
-          *(const bf16x8*)&Bs[buf][(wc * 32 + ni * 16 + fr) * LDA + fk * 8];
+          *(const bf16x8*)&Bs[buf][(wc * WN + ni * 16 + fr) * LDA + fk * 8];
 #pragma unroll
-    for (int mi = 0; mi < 2; mi++)
+    for (int mi = 0; mi < MI; mi++)
 #pragma unroll
-      for (int ni = 0; ni < 2; ni++)
+      for (int ni = 0; ni < NI; ni++)
         acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
     buf ^= 1;
@@ -255,15 +290,17 @@ __global__ __launch_bounds__(256) void k_conv_mfma(
   }
 
   // Epilogue. D fragment: col = lane&15 (=fr), row = fk*4 + q.
-  float ssum[2] = {0.f, 0.f}, ssq[2] = {0.f, 0.f};
+  float ssum[NI], ssq[NI];
 #pragma unroll
-  for (int mi = 0; mi < 2; mi++) {
+  for (int ni = 0; ni < NI; ni++) ssum[ni] = ssq[ni] = 0.f;
 #pragma unroll
-    for (int ni = 0; ni < 2; ni++) {
-      int gn = n0 + wc * 32 + ni * 16 + fr;
+  for (int mi = 0; mi < MI; mi++) {
+#pragma unroll
+    for (int ni = 0; ni < NI; ni++) {
+      int gn = n0 + wc * WN + ni * 16 + fr;
 #pragma unroll
       for (int q = 0; q < 4; q++) {
-        int gm = m0 + wr * 32 + mi * 16 + fk * 4 + q;
+        int gm = m0 + wr * WM + mi * 16 + fk * 4 + q;
         if (gm < p.M && gn < Ntot) {
           float v = acc[mi][ni][q];
           if (SPLIT) {
@@ -283,12 +320,12 @@ __global__ __launch_bounds__(256) void k_conv_mfma(
   }
   if (STATS && !SPLIT && stats != nullptr) {
 #pragma unroll
-    for (int ni = 0; ni < 2; ni++) {
+    for (int ni = 0; ni < NI; ni++) {
       float s = ssum[ni] + __shfl_xor(ssum[ni], 16, 64);
       s += __shfl_xor(s, 32, 64);
       float s2 = ssq[ni] + __shfl_xor(ssq[ni], 16, 64);
       s2 += __shfl_xor(s2, 32, 64);
-      int gn = n0 + wc * 32 + ni * 16 + fr;
+      int gn = n0 + wc * WN + ni * 16 + fr;
       if (fk == 0 && gn < Ntot) {
         atomicAdd(&stats[gn], s);
         atomicAdd(&stats[Ntot + gn], s2);
@@ -599,24 +636,47 @@ extern "C" int conv_dgrad_splitk(ConvP p) {
   return sk;
 }
 
+// Throughput-tile selection: the biggest block tile whose grid still
+// fills the 256 CUs (HZ_TILE_FILL, default 192 blocks).  CIFAR-shape
+// grids stay on the latency-optimized 64×64 tile; ImageNet-shaped convs
+// (ResNet50@224 etc.) move to 128×64 / 128×128 where each wave issues
+// 2–4× the MFMAs per LDS fragment read.
+static inline int pick_tile(int M, int N) {
+  static int fill = [] {
+    const char* e = getenv("HZ_TILE_FILL");
+    return e ? atoi(e) : 192;
+  }();
+  if (N >= 128 && (long)cdiv_h(M, 128) * cdiv_h(N, 128) >= fill) return 2;
+  if ((long)cdiv_h(M, 128) * cdiv_h(N, 64) >= fill) return 1;
+  return 0;
+}
+
 // Non-split forward (bf16 out + fused stats).
 extern "C" void launch_conv_fwd(const void* x, const void* w, void* y,
                                 float* stats, ConvP p, hipStream_t st) {
-  dim3 grid(cdiv_h(p.K, 64), cdiv_h(p.M, 64));
   bool vec = (p.C % 8) == 0 && (p.Kd % 8) == 0;
   bool s = stats != nullptr;
   auto A = (const bf16*)x;
   auto B = (const bf16*)w;
   auto Y = (bf16*)y;
   MagicP mg = make_magic(p, 1);
-#define CASE(VA, VB, ST)                                     \
-  k_conv_mfma<1, VA, VB, ST, false><<<grid, 256, 0, st>>>(   \
+  int tile = pick_tile(p.M, p.K);
+#define CASE_T(VA, VB, ST, TM, TN)                                     \
+  k_conv_mfma<1, VA, VB, ST, false, TM, TN><<<grid, 256, 0, st>>>(     \
       A, B, Y, nullptr, stats, p, mg, p.K, 0, 0)
-  if (vec && s) CASE(true, true, true);
-  else if (vec) CASE(true, true, false);
-  else if (s) CASE(false, false, true);
-  else CASE(false, false, false);
-#undef CASE
+#define DISPATCH(TM, TN)                                               \
+  do {                                                                 \
+    dim3 grid(cdiv_h(p.K, TN), cdiv_h(p.M, TM));                       \
+    if (vec && s) CASE_T(true, true, true, TM, TN);                    \
+    else if (vec) CASE_T(true, true, false, TM, TN);                   \
+    else if (s) CASE_T(false, false, true, TM, TN);                    \
+    else CASE_T(false, false, false, TM, TN);                          \
+  } while (0)
+  if (tile == 2) DISPATCH(128, 128);
+  else if (tile == 1) DISPATCH(128, 64);
+  else DISPATCH(64, 64);
+#undef DISPATCH
+#undef CASE_T
 }
 
 // Split-K forward: f32 atomic partials into ws (pre-zeroed [M][K]).
@@ -659,15 +719,22 @@ extern "C" void launch_conv_dgrad(const void* dz, const void* w_rsck,
           <<<grid, 256, 0, st>>>(A, B, nullptr, ws, nullptr, p,
                                  make_magic(p, 2), p.C, kchunk, 0);
   } else {
-    dim3 grid(cdiv_h(p.C, 64), cdiv_h(p.M, 64));
-    if (vec)
-      k_conv_mfma<2, true, true, false, false>
-          <<<grid, 256, 0, st>>>(A, B, (bf16*)dx, nullptr, nullptr, p,
-                                 make_magic(p, 2), p.C, 0, accum);
-    else
-      k_conv_mfma<2, false, false, false, false>
-          <<<grid, 256, 0, st>>>(A, B, (bf16*)dx, nullptr, nullptr, p,
-                                 make_magic(p, 2), p.C, 0, accum);
+    MagicP mg = make_magic(p, 2);
+    int tile = pick_tile(p.M, p.C);
+#define DGRAD_T(VA, VB, TM, TN)                                         \
+  k_conv_mfma<2, VA, VB, false, false, TM, TN><<<grid, 256, 0, st>>>(   \
+      A, B, (bf16*)dx, nullptr, nullptr, p, mg, p.C, 0, accum)
+#define DISPATCH(TM, TN)                                                \
+  do {                                                                  \
+    dim3 grid(cdiv_h(p.C, TN), cdiv_h(p.M, TM));                        \
+    if (vec) DGRAD_T(true, true, TM, TN);                               \
+    else DGRAD_T(false, false, TM, TN);                                 \
+  } while (0)
+    if (tile == 2) DISPATCH(128, 128);
+    else if (tile == 1) DISPATCH(128, 64);
+    else DISPATCH(64, 64);
+#undef DISPATCH
+#undef DGRAD_T
   }
 }
 

@@ -77,6 +77,31 @@ def test_conv_bn_relu_fwd(cfg):
     assert rel(gpu.running_var, cpu.running_var) < 3e-2
 
 
+@pytest.mark.parametrize("cfg", [
+    (64, 64, 3, 1, 56),    # M=50k -> 128x64 throughput tile
+    (64, 128, 3, 1, 56),   # -> 128x128 tile
+    (64, 130, 3, 1, 56),   # N tail on the 128x128 tile
+])
+def test_conv_throughput_tiles_fwd_bwd(cfg):
+    """ImageNet-shaped convs route to the 128-wide throughput tiles
+    (pick_tile): numerics must match the CPU reference exactly like the
+    64x64 latency tile does."""
+    cin, cout, k, s, hw = cfg
+    cpu, gpu = _make_pair(cin, cout, k, s)
+    x = torch.randn(16, cin, hw, hw)
+    xg = to_gpu_cl(x).requires_grad_(True)
+    xc = x.clone().requires_grad_(True)
+    y_ref = cpu(xc)
+    y = gpu(xg)
+    assert rel(y, y_ref) < 3e-2, f"cfg={cfg} rel={rel(y, y_ref)}"
+    gy = torch.randn_like(y_ref)
+    y_ref.backward(gy)
+    y.backward(to_gpu_cl(gy))
+    assert rel(xg.grad, xc.grad) < 4e-2, f"dgrad cfg={cfg}"
+    assert rel(gpu.weight.grad, cpu.weight.grad) < 4e-2, f"wgrad cfg={cfg}"
+    assert rel(gpu.bn_weight.grad, cpu.bn_weight.grad) < 4e-2
+
+
 def test_conv_bn_eval_mode():
     cpu, gpu = _make_pair(64, 64, 3, 1)
     cpu.eval()
