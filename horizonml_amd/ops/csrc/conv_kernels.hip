@@ -646,6 +646,14 @@ static inline int pick_tile(int M, int N) {
     const char* e = getenv("HZ_TILE_FILL");
     return e ? atoi(e) : 192;
   }();
+  // measured (gpurun conv sweep, r50@224 bs=32): the 128-row tiles win
+  // from M≈25k up (+30-60% fwd) but lose ~20% at M≈6k even when the
+  // grid-fill bound is met — keep a hard M floor
+  static int m_min = [] {
+    const char* e = getenv("HZ_TILE_M_MIN");
+    return e ? atoi(e) : 16384;
+  }();
+  if (M < m_min) return 0;
   if (N >= 128 && (long)cdiv_h(M, 128) * cdiv_h(N, 128) >= fill) return 2;
   if ((long)cdiv_h(M, 128) * cdiv_h(N, 64) >= fill) return 1;
   return 0;
