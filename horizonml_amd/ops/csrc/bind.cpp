@@ -869,6 +869,18 @@ void grad_divergence(Tensor g, Tensor prev, Tensor sumsq, Tensor out,
 // uint8 NCHW batch -> normalized bf16 channels_last, one kernel (the
 // engines' H2D preprocessing; reference transform (0.5,0.5,0.5) mean/std,
 // data_parallel_train.py:44-47).
+// Debug/tuning entry: launch the BN-backward reduce directly (variant
+// pinned by HZ_BN_V8) for kernel-isolation timing sweeps.
+void bn_reduce_bench(Tensor dy, Tensor y, Tensor x, Tensor mean,
+                     Tensor invstd, Tensor gamma, Tensor beta, Tensor sdz,
+                     Tensor sdzx, int64_t M, int64_t C, int64_t mask) {
+  launch_bnact_bwd_reduce(dy.data_ptr(), y.data_ptr(), x.data_ptr(),
+                          mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                          gamma.data_ptr<float>(), beta.data_ptr<float>(),
+                          sdz.data_ptr<float>(), sdzx.data_ptr<float>(), M,
+                          (int)C, (int)mask, cur_stream());
+}
+
 Tensor normalize_u8(Tensor x, double mean, double std) {
   TORCH_CHECK(x.dim() == 4 && x.dtype() == torch::kUInt8 && x.is_cuda()
                   && x.is_contiguous(),
@@ -902,6 +914,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("permute_krsc_rsck", &permute_krsc_rsck);
   m.def("grad_divergence", &grad_divergence);
   m.def("normalize_u8", &normalize_u8);
+  m.def("bn_reduce_bench", &bn_reduce_bench);
   m.def("set_deterministic", &set_deterministic);
   m.def("deterministic_enabled", &deterministic_enabled);
   m.def("set_wgrad_defer", &set_wgrad_defer);
