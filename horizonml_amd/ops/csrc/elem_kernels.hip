@@ -371,21 +371,20 @@ __global__ __launch_bounds__(256) void k_bnact_bwd_reduce_v8(
     sdzx[tid][e] = adzx[e];
   }
   __syncthreads();
-  if (th_m == 0 && tid < active) {  // tid < lpr: one reducer per c-group
-    float rdz[8] = {}, rdzx[8] = {};
+  // wave-parallel channel reduction: flat LDS index j*C + c walks the
+  // [m-lane][channel] partials; one lane per channel, ONE coalesced
+  // atomic instruction per wave per array (the per-e unrolled epilogue
+  // serialized ~0.2 us of LDS latency + 16 atomic issues per block)
+  const float* S1 = &sdz[0][0];
+  const float* S2 = &sdzx[0][0];
+  for (int c = tid; c < C; c += 256) {
+    float r1 = 0.f, r2 = 0.f;
     for (int j = 0; j < mstep; j++) {
-      int s = j * lpr + tid;
-#pragma unroll
-      for (int e = 0; e < 8; e++) {
-        rdz[e] += sdz[s][e];
-        rdzx[e] += sdzx[s][e];
-      }
+      r1 += S1[j * C + c];
+      r2 += S2[j * C + c];
     }
-#pragma unroll
-    for (int e = 0; e < 8; e++) {
-      atomicAdd(&sum_dz[th_c + e], rdz[e]);
-      atomicAdd(&sum_dzx[th_c + e], rdzx[e]);
-    }
+    atomicAdd(&sum_dz[c], r1);
+    atomicAdd(&sum_dzx[c], r2);
   }
 }
 
@@ -469,21 +468,20 @@ __global__ __launch_bounds__(256) void k_cast_bnact_v8(
     sdzx[tid][e] = adzx[e];
   }
   __syncthreads();
-  if (th_m == 0 && tid < active) {
-    float rdz[8] = {}, rdzx[8] = {};
+  // wave-parallel channel reduction: flat LDS index j*C + c walks the
+  // [m-lane][channel] partials; one lane per channel, ONE coalesced
+  // atomic instruction per wave per array (the per-e unrolled epilogue
+  // serialized ~0.2 us of LDS latency + 16 atomic issues per block)
+  const float* S1 = &sdz[0][0];
+  const float* S2 = &sdzx[0][0];
+  for (int c = tid; c < C; c += 256) {
+    float r1 = 0.f, r2 = 0.f;
     for (int j = 0; j < mstep; j++) {
-      int s = j * lpr + tid;
-#pragma unroll
-      for (int e = 0; e < 8; e++) {
-        rdz[e] += sdz[s][e];
-        rdzx[e] += sdzx[s][e];
-      }
+      r1 += S1[j * C + c];
+      r2 += S2[j * C + c];
     }
-#pragma unroll
-    for (int e = 0; e < 8; e++) {
-      atomicAdd(&sum_dz[th_c + e], rdz[e]);
-      atomicAdd(&sum_dzx[th_c + e], rdzx[e]);
-    }
+    atomicAdd(&sum_dz[c], r1);
+    atomicAdd(&sum_dzx[c], r2);
   }
 }
 
