@@ -1281,13 +1281,25 @@ static int bn_v8_enabled() {
   return v;
 }
 
+// v8 wins from M >= ~65k rows (isolated sweep gpurun_out/bn_v8c.txt:
+// 3.0 TB/s vs 0.7 at M=401k/C=64); below that the scalar kernel's
+// per-channel-column layout is faster — per-block setup + atomic volume
+// dominate the short v8 m-loops.
+static long bn_v8_m_min() {
+  static long v = [] {
+    const char* e = getenv("HZ_BN_V8_MMIN");
+    return e ? atol(e) : 65536L;
+  }();
+  return v;
+}
+
 void launch_cast_bnact(const float* src, void* dst, long M, int C,
                        int nsplit, int accum, const void* x_up,
                        const void* y_up, const float* smean,
                        const float* sinvstd, const float* gamma,
                        const float* beta, float* sum_dz, float* sum_dzx,
                        int mask_mode, hipStream_t st) {
-  if ((C & 7) == 0 && C <= 2048 && bn_v8_enabled()) {
+  if ((C & 7) == 0 && C <= 2048 && bn_v8_enabled() && M >= bn_v8_m_min()) {
     // vectorized: whole C per block, m split across blockIdx.y.  msplit
     // sized so every thread has >=1 row and the grid reaches ~768 blocks
     // on big-M shapes (matching the scalar kernel's fill).
@@ -1328,7 +1340,7 @@ void launch_bnact_bwd_reduce(const void* dy, const void* yout, const void* x,
                              float* sum_dz, float* sum_dzx, long M, int C,
                              int mask_mode, hipStream_t st) {
   // sum_dz/sum_dzx must be pre-zeroed (they are accumulated atomically)
-  if ((C & 7) == 0 && C <= 2048 && bn_v8_enabled()) {
+  if ((C & 7) == 0 && C <= 2048 && bn_v8_enabled() && M >= bn_v8_m_min()) {
     int lpr = C >> 3, mstep = 256 / lpr;
     int msplit = (int)min((long)768, max((long)1, (M + mstep - 1) / mstep));
     if (g_det_kernels) msplit = 1;
