@@ -41,11 +41,11 @@ def main():
             .to(memory_format=torch.channels_last).to(torch.bfloat16)
         it = 30
         for _ in range(5):
-            C_.wgrad_only(x, dz, cout, k, k, s, k // 2, False)
+            C_.wgrad_only(x, dz, cout, k, k, s, k // 2, False, None)
         torch.cuda.synchronize()
         t0 = time.perf_counter()
         for _ in range(it):
-            C_.wgrad_only(x, dz, cout, k, k, s, k // 2, False)
+            C_.wgrad_only(x, dz, cout, k, k, s, k // 2, False, None)
         torch.cuda.synchronize()
         dt = (time.perf_counter() - t0) / it
         flops = 2.0 * M * cout * cin * k * k
