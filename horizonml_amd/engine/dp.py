@@ -271,6 +271,7 @@ def train_dp_flat(ctx: DistContext, epochs: int, sample_size: int,
             n_steps += 1
             prof.step_end()
 
+        t = prof.epoch_end()  # syncs: epoch_time includes the GPU tail
         epoch_time = time.time() - epoch_start
         loss_v = float(loss_acc.cpu()) / max(1, count)
         acc_v = 100.0 * float(corr_acc.cpu()) / max(1, count)
@@ -280,7 +281,6 @@ def train_dp_flat(ctx: DistContext, epochs: int, sample_size: int,
         if probe_divergence:
             div_acc.zero_()
         count = 0
-        t = prof.epoch_end()
         gmem, gutil = sample_gpu_resources(dev)
         m = EpochMetrics(
             epoch=epoch + 1, loss=loss_v, accuracy=acc_v,
@@ -366,9 +366,9 @@ def train_dp(ctx: DistContext, epochs: int, sample_size: int, logs_dir: str,
                 probe.step()
             prof.step_end()
 
+        t = prof.epoch_end()  # syncs: epoch_time includes the GPU tail
         epoch_time = time.time() - epoch_start
         loss_v, acc_v = meters.epoch_values()
-        t = prof.epoch_end()
         gmem, gutil = sample_gpu_resources(ctx.device if ctx.is_gpu else None)
         m = EpochMetrics(
             epoch=epoch + 1, loss=loss_v, accuracy=acc_v,
