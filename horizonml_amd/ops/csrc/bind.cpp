@@ -30,6 +30,7 @@ int wgrad_msplit(ConvP);
 void launch_wgrad(const void*, const void*, float*, float*, ConvP,
                   hipStream_t);
 void launch_wgrad_batched(const void*, int, hipStream_t);
+void launch_wgrad_batched_t128(const void*, int, hipStream_t);
 void launch_wgrad_reduce_batched(const void*, int, hipStream_t);
 void launch_bn_apply(const void*, const void*, void*, const float*,
                      const float*, const float*, float*, float*, float*,
@@ -263,29 +264,43 @@ void flush_one_group(int group_lo, int group_hi) {
                            pend[0].x.options().dtype(torch::kFloat32));
   float* ws_base = ws_total > 0 ? g_wgrad_ws.data_ptr<float>() : nullptr;
 
-  for (int lo = 0; lo < n; lo += WG_MAX_TASKS) {
-    WgradBatchArgs a{};
-    a.n = std::min(WG_MAX_TASKS, n - lo);
-    int blocks = 0;
-    for (int j = 0; j < a.n; j++) {
-      const PendingWgrad& pw = pend[lo + j];
-      WgradTask& t = a.t[j];
-      t.X = pw.x.data_ptr();
-      t.Dz = pw.dconv.data_ptr();
-      t.out = msplit[lo + j] > 1 ? ws_base + ws_off[lo + j]
-                                 : pw.dw.data_ptr<float>();
-      t.p = pw.p;
-      t.mg = make_magic_mode1(pw.p);
-      t.Ntot = pw.p.K;
-      t.mchunk = mchunk[lo + j];
-      t.msplit = msplit[lo + j];
-      t.tx = cdiv_i(pw.p.Kd, 64);
-      t.ty = cdiv_i(pw.p.K, 64);
-      t.base = blocks;
-      t.vec = (pw.p.C % 8) == 0;
-      blocks += t.tx * t.ty * t.msplit;
+  // two tile classes: 128-wide k3 tiles for big-Kd large-M convs (halves
+  // the Dz re-read and doubles MFMA per staging write), 64 otherwise —
+  // separate launches so each keeps its own LDS footprint/occupancy
+  for (int tk3 : {64, 128}) {
+    std::vector<int> idx;
+    for (int i = 0; i < n; i++) {
+      bool big = pend[i].p.Kd >= 512 && pend[i].p.M >= 8192;
+      if ((tk3 == 128) == big) idx.push_back(i);
     }
-    launch_wgrad_batched(&a, blocks, st);
+    for (size_t lo = 0; lo < idx.size(); lo += WG_MAX_TASKS) {
+      WgradBatchArgs a{};
+      a.n = (int)std::min((size_t)WG_MAX_TASKS, idx.size() - lo);
+      int blocks = 0;
+      for (int j = 0; j < a.n; j++) {
+        const int i = idx[lo + j];
+        const PendingWgrad& pw = pend[i];
+        WgradTask& t = a.t[j];
+        t.X = pw.x.data_ptr();
+        t.Dz = pw.dconv.data_ptr();
+        t.out = msplit[i] > 1 ? ws_base + ws_off[i]
+                              : pw.dw.data_ptr<float>();
+        t.p = pw.p;
+        t.mg = make_magic_mode1(pw.p);
+        t.Ntot = pw.p.K;
+        t.mchunk = mchunk[i];
+        t.msplit = msplit[i];
+        t.tx = cdiv_i(pw.p.Kd, tk3);
+        t.ty = cdiv_i(pw.p.K, 64);
+        t.base = blocks;
+        t.vec = (pw.p.C % 8) == 0;
+        blocks += t.tx * t.ty * t.msplit;
+      }
+      if (tk3 == 128)
+        launch_wgrad_batched_t128(&a, blocks, st);
+      else
+        launch_wgrad_batched(&a, blocks, st);
+    }
   }
 
   // batched slab reduce for the split tasks

@@ -357,22 +357,30 @@ __global__ __launch_bounds__(256) void k_conv_mfma(
 // replace ~40 (bind.cpp::flush_wgrad).
 #define LDW 40  // 32 m + 8 pad
 
-union WgradSmem {
+// TK3: k3-tile width (64 default; 128 for big-Kd large-M convs — halves
+// the Dz re-read traffic and doubles MFMA per staging write).  The out
+// image keeps an odd f32 row stride (65 / 131) for conflict-free stores.
+template <int TK3>
+union WgradSmemT {
   struct {
-    bf16 At[64 * LDW];  // [k3][m]
-    bf16 Dt[64 * LDW];  // [ko][m]
+    bf16 At[TK3 * LDW];  // [k3][m]
+    bf16 Dt[64 * LDW];   // [ko][m]
   } s;
-  float out[64][65];  // transposed epilogue staging (65: conflict-free)
+  float out[64][TK3 == 64 ? 65 : 131];  // transposed epilogue staging
 };
+using WgradSmem = WgradSmemT<64>;
 
-template <bool VECA>
+template <bool VECA, int TK3 = 64>
 DEV void wgrad_tile(const bf16* __restrict__ X, const bf16* __restrict__ Dz,
                     float* __restrict__ out, const ConvP& p,
                     const MagicP& mg, int Ntot, int mchunk, int tx, int ty,
-                    int z, bool split, WgradSmem& smem) {
+                    int z, bool split, WgradSmemT<TK3>& smem) {
+  constexpr int AC = TK3 / 64;   // A staging chunks per thread
+  constexpr int MI = TK3 / 32;   // k3 fragments per wave
+  constexpr int OSTR = TK3 == 64 ? 65 : 131;
   bf16* At = smem.s.At;   // [k3][m]
   bf16* Dt = smem.s.Dt;   // [ko][m]
-  const int k3_0 = tx * 64, n0 = ty * 64;
+  const int k3_0 = tx * TK3, n0 = ty * 64;
   const int mbeg = z * mchunk;
   const int mend = min(p.M, mbeg + mchunk);
   const int tid = threadIdx.x;
@@ -381,10 +389,12 @@ DEV void wgrad_tile(const bf16* __restrict__ X, const bf16* __restrict__ Dz,
   const int wr = wave >> 1, wc = wave & 1;
   const int fr = lane & 15, fk = lane >> 4;
 
-  f32x4 acc[2][2] = {};
+  f32x4 acc[MI][2] = {};
 
-  V8 a_nx = load8_a<1, VECA>(X, p, mg, mbeg + sm, k3_0 + sv);
-  V8 d_nx;
+  V8 a_nx[AC], d_nx;
+#pragma unroll
+  for (int j = 0; j < AC; j++)
+    a_nx[j] = load8_a<1, VECA>(X, p, mg, mbeg + sm, k3_0 + j * 64 + sv);
   {
     int m = mbeg + sm, n = n0 + sv;
     if (m < p.M && n < Ntot) d_nx.u = *(const uint4*)(Dz + (long)m * Ntot + n);
@@ -395,11 +405,13 @@ DEV void wgrad_tile(const bf16* __restrict__ X, const bf16* __restrict__ Dz,
     // transposed scatter into LDS: element (m=sm, k3=sv+e) -> At[sv+e][sm']
     // with the m 8-chunk XOR-swizzled by row&3 (8-way -> 2-way write banks)
 #pragma unroll
-    for (int e = 0; e < 8; e++) {
-      int row = sv + e;
-      int smx = smlo | (((smhi ^ row) & 3) << 3);
-      At[row * LDW + smx] = a_nx.e[e];
-    }
+    for (int j = 0; j < AC; j++)
+#pragma unroll
+      for (int e = 0; e < 8; e++) {
+        int row = j * 64 + sv + e;
+        int smx = smlo | (((smhi ^ row) & 3) << 3);
+        At[row * LDW + smx] = a_nx[j].e[e];
+      }
 #pragma unroll
     for (int e = 0; e < 8; e++) {
       int row = sv + e;
@@ -408,17 +420,20 @@ DEV void wgrad_tile(const bf16* __restrict__ X, const bf16* __restrict__ Dz,
     }
     __syncthreads();
     if (m0 + 32 < mend) {
-      a_nx = load8_a<1, VECA>(X, p, mg, m0 + 32 + sm, k3_0 + sv);
+#pragma unroll
+      for (int j = 0; j < AC; j++)
+        a_nx[j] = load8_a<1, VECA>(X, p, mg, m0 + 32 + sm,
+                                   k3_0 + j * 64 + sv);
       int m = m0 + 32 + sm, n = n0 + sv;
       if (m < p.M && n < Ntot)
         d_nx.u = *(const uint4*)(Dz + (long)m * Ntot + n);
       else
         d_nx.u = uint4{0, 0, 0, 0};
     }
-    bf16x8 af[2], bf[2];
+    bf16x8 af[MI], bf[2];
 #pragma unroll
-    for (int mi = 0; mi < 2; mi++) {
-      int row = wr * 32 + mi * 16 + fr;
+    for (int mi = 0; mi < MI; mi++) {
+      int row = wr * (TK3 / 2) + mi * 16 + fr;
       af[mi] = *(const bf16x8*)&At[row * LDW + ((fk ^ row) & 3) * 8];
     }
 #pragma unroll
@@ -427,7 +442,7 @@ DEV void wgrad_tile(const bf16* __restrict__ X, const bf16* __restrict__ Dz,
       bf[ni] = *(const bf16x8*)&Dt[row * LDW + ((fk ^ row) & 3) * 8];
     }
 #pragma unroll
-    for (int mi = 0; mi < 2; mi++)
+    for (int mi = 0; mi < MI; mi++)
 #pragma unroll
       for (int ni = 0; ni < 2; ni++)
         acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
@@ -437,23 +452,25 @@ DEV void wgrad_tile(const bf16* __restrict__ X, const bf16* __restrict__ Dz,
   // D: col (=ko) = fr, row (=k3) = fk*4+q.  Transpose through LDS.
   __syncthreads();  // LDS union: staging buffers are done
 #pragma unroll
-  for (int mi = 0; mi < 2; mi++)
+  for (int mi = 0; mi < MI; mi++)
 #pragma unroll
     for (int ni = 0; ni < 2; ni++) {
       int ko_l = wc * 32 + ni * 16 + fr;
 #pragma unroll
       for (int q = 0; q < 4; q++)
-        smem.out[ko_l][wr * 32 + mi * 16 + fk * 4 + q] = acc[mi][ni][q];
+        smem.out[ko_l][wr * (TK3 / 2) + mi * 16 + fk * 4 + q] =
+            acc[mi][ni][q];
     }
   __syncthreads();
   float* dst = out + (split ? (long)z * Ntot * (long)p.Kd : 0L);
-  const int ko_r = tid >> 2, cch = (tid & 3) * 16;
+  constexpr int CCH = TK3 / 4;  // k3 span per thread in the store pass
+  const int ko_r = tid >> 2, cch = (tid & 3) * CCH;
   const int gko = n0 + ko_r;
   if (gko < Ntot) {
     long base = (long)gko * p.Kd + k3_0 + cch;
-    if (k3_0 + cch + 16 <= p.Kd) {
+    if (k3_0 + cch + CCH <= p.Kd) {
 #pragma unroll
-      for (int e = 0; e < 16; e += 4) {
+      for (int e = 0; e < CCH; e += 4) {
         float4 v = *(const float4*)&smem.out[ko_r][cch + e];
         if (!split) {  // accumulate into .grad (single writer, RMW is safe)
           float4 d = *(const float4*)&dst[base + e];
@@ -462,7 +479,7 @@ DEV void wgrad_tile(const bf16* __restrict__ X, const bf16* __restrict__ Dz,
         *(float4*)&dst[base + e] = v;
       }
     } else {
-      for (int e = 0; e < 16 && k3_0 + cch + e < p.Kd; e++)
+      for (int e = 0; e < CCH && k3_0 + cch + e < p.Kd; e++)
         dst[base + e] = smem.out[ko_r][cch + e] + (split ? 0.f : dst[base + e]);
     }
   }
@@ -528,6 +545,7 @@ struct WgradBatchArgs {
   WgradTask t[WG_MAX_TASKS];
 };
 
+template <int TK3>
 __global__ __launch_bounds__(256) void k_wgrad_batched(WgradBatchArgs a) {
   int bid = blockIdx.x;
   int i = 0;
@@ -537,13 +555,13 @@ __global__ __launch_bounds__(256) void k_wgrad_batched(WgradBatchArgs a) {
   int nt = t.tx * t.ty;
   int z = local / nt, rem = local - z * nt;
   int ty = rem / t.tx, tx = rem - ty * t.tx;
-  __shared__ WgradSmem smem;
+  __shared__ WgradSmemT<TK3> smem;
   if (t.vec)
-    wgrad_tile<true>(t.X, t.Dz, t.out, t.p, t.mg, t.Ntot, t.mchunk, tx, ty,
-                     z, t.msplit > 1, smem);
+    wgrad_tile<true, TK3>(t.X, t.Dz, t.out, t.p, t.mg, t.Ntot, t.mchunk,
+                          tx, ty, z, t.msplit > 1, smem);
   else
-    wgrad_tile<false>(t.X, t.Dz, t.out, t.p, t.mg, t.Ntot, t.mchunk, tx,
-                      ty, z, t.msplit > 1, smem);
+    wgrad_tile<false, TK3>(t.X, t.Dz, t.out, t.p, t.mg, t.Ntot, t.mchunk,
+                           tx, ty, z, t.msplit > 1, smem);
 }
 
 struct WredTask {
@@ -767,7 +785,16 @@ extern "C" void launch_gemm_bf16(const void* a, const void* b, void* c, int M,
 extern "C" void launch_wgrad_batched(const void* args, int blocks,
                                      hipStream_t st) {
   if (blocks > 0)
-    k_wgrad_batched<<<blocks, 256, 0, st>>>(
+    k_wgrad_batched<64><<<blocks, 256, 0, st>>>(
+        *(const WgradBatchArgs*)args);
+}
+
+// 128-wide k3 tile class (tasks with Kd >= 512 and M >= 8192): tx counted
+// in 128-wide tiles by the caller.
+extern "C" void launch_wgrad_batched_t128(const void* args, int blocks,
+                                          hipStream_t st) {
+  if (blocks > 0)
+    k_wgrad_batched<128><<<blocks, 256, 0, st>>>(
         *(const WgradBatchArgs*)args);
 }
 

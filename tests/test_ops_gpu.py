@@ -276,6 +276,33 @@ def test_grad_divergence_kernel():
     assert abs(out.item() - ref) / ref < 1e-4
 
 
+def test_wgrad_batched_t128_matches_standalone():
+    """Deferred/batched wgrad routes Kd>=512 & M>=8192 tasks through the
+    128-wide k3 tile — must match the standalone 64-wide kernel."""
+    torch.manual_seed(4)
+    shapes = [(16, 64, 32, 64, 3, 1),    # M=16384, Kd=576 -> t128
+              (16, 128, 16, 128, 3, 1),  # M=4096 -> stays t64
+              (40, 72, 24, 128, 3, 1)]   # M=23040, Kd=648 (non-vec-ish C)
+    for bs, cin, hw, cout, k, s in shapes:
+        x = torch.randn(bs, cin, hw, hw, device="cuda") \
+            .to(memory_format=torch.channels_last).to(torch.bfloat16)
+        ho = (hw + 2 * (k // 2) - k) // s + 1
+        dz = torch.randn(bs, cout, ho, ho, device="cuda") \
+            .to(memory_format=torch.channels_last).to(torch.bfloat16)
+        ref = _C().wgrad_only(x, dz, cout, k, k, s, k // 2, False, None)
+        dw = torch.zeros_like(ref)
+        prev = _C().wgrad_defer_enabled()
+        _C().set_wgrad_defer(True)
+        try:
+            _C().wgrad_only(x, dz, cout, k, k, s, k // 2, True, dw)
+            assert _C().wgrad_pending() >= 1
+            _C().flush_wgrad()
+        finally:
+            _C().set_wgrad_defer(prev)
+        assert _C().wgrad_pending() == 0
+        assert rel(dw, ref) < 1e-4, f"shape {(bs, cin, hw, cout, k)}"
+
+
 def test_adam_fused_divergence_probe():
     """adam_step's fused probe must equal the standalone gdiv kernel."""
     torch.manual_seed(3)
