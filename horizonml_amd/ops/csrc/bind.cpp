@@ -248,7 +248,12 @@ void flush_one_group(int group_lo, int group_hi) {
   long ws_total = 0;
   for (int i = 0; i < n; i++) {
     const ConvP& p = pend[i].p;
-    int ms = std::max(1, cdiv_i(p.M, 512));
+    // large-M tasks take 2048-row chunks: the batched grid pools every
+    // task's blocks so per-task fill matters less than slab volume — at
+    // mchunk 512 a ResNet50 layer1 conv makes 196 f32 slabs whose
+    // write+reduce traffic dominated wgrad_reduce (r02 traces)
+    const int mc_base = p.M >= 32768 ? 2048 : 512;
+    int ms = std::max(1, cdiv_i(p.M, mc_base));
     int mc = cdiv_i(cdiv_i(p.M, ms), 32) * 32;
     ms = cdiv_i(p.M, mc);
     msplit[i] = ms;
