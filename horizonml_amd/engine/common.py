@@ -73,6 +73,54 @@ class GradDivergenceProbe:
         return v
 
 
+class FlatEagerOptimizer:
+    """The flat fast-path machinery (FlatParamManager: direct grads into
+    one flat buffer, batched deferred wgrads, fused Adam/SGD + bf16 shadow
+    refresh) behind a torch-optimizer interface, WITHOUT graph capture —
+    the eager PP/TP engines' GPU optimizer (r02).  Replaces torch Adam's
+    multi-tensor passes + per-conv wgrad launches + per-param zero fills.
+
+    Ordering contract: deferred weight-gradient GEMMs complete at
+    ``flush_wgrad()`` (or inside ``step()``); anything reading ``p.grad``
+    before the step — DP finalize packs, divergence probes — must flush
+    first.  Gradients are zeroed at ``zero_grad()`` (loop start), never by
+    the step, so probes may read them after stepping.
+    """
+
+    def __init__(self, module, name: str, lr: float, device):
+        from .flat import FlatParamManager, HorizonAdam, HorizonSGD
+        self.mgr = FlatParamManager(module, device)
+        self.opt = (HorizonAdam(self.mgr, lr=lr) if name == "adam"
+                    else HorizonSGD(self.mgr, lr=lr))
+
+    def zero_grad(self, set_to_none: bool = False):
+        # set_to_none would detach the flat views — always zero in place
+        self.mgr.zero_grad()
+
+    def flush_wgrad(self):
+        from .. import ops as _ops
+        _ops.extension().flush_wgrad()
+
+    def step(self):
+        self.opt.step(zero_grad=False)
+
+    def state_dict(self):
+        return self.opt.state_dict()
+
+    def load_state_dict(self, state):
+        self.opt.load_state_dict(state)
+
+
+def build_engine_optimizer(module, params, name: str, lr: float, device):
+    """Fused flat-eager optimizer on GPU (extension present), torch
+    optimizer otherwise."""
+    if device is not None and device.type == "cuda":
+        from .. import ops as _ops
+        if _ops.has_extension():
+            return FlatEagerOptimizer(module, name, lr, device)
+    return build_optimizer(params, name, lr=lr)
+
+
 def build_optimizer(params, name: str = "adam", lr: float = 1e-3,
                     momentum: float = 0.9, weight_decay: float = 0.0):
     """Reference default: Adam(lr=1e-3) (``data_parallel_train.py:205``).

@@ -16,7 +16,7 @@ import torch
 import torch.nn.functional as F
 
 from ..data import get_dataloader
-from ..engine.common import (GradDivergenceProbe, build_optimizer,
+from ..engine.common import (GradDivergenceProbe, build_engine_optimizer,
                              progress_iter)
 from ..models import build_model, partition_model
 from ..parallel.pipeline import PipelineStage
@@ -72,8 +72,11 @@ def train_pp(ctx: DistContext, epochs: int, sample_size: int, logs_dir: str,
                           group=group, profiler=prof)
     params = list(seg.parameters())
     has_params = len(params) > 0
-    optimizer = build_optimizer(params, optimizer_name, lr=lr) \
-        if has_params else None
+    # GPU: flat-eager fused optimizer (direct grads, batched wgrad, fused
+    # Adam); CPU: torch optimizer
+    optimizer = build_engine_optimizer(seg, params, optimizer_name, lr,
+                                       ctx.device) if has_params else None
+    flat_opt = optimizer is not None and hasattr(optimizer, "flush_wgrad")
     probe = (GradDivergenceProbe(params)
              if (probe_divergence and has_params) else None)
     # DP replica sync for hybrid DP×PP: bucketed all-reduce over dp_group
@@ -132,12 +135,15 @@ def train_pp(ctx: DistContext, epochs: int, sample_size: int, logs_dir: str,
                     loss_fn=loss_fn, microbatches=microbatches,
                     batch_hint=x.shape[0])
             if ddp is not None:
+                if flat_opt:  # deferred wgrads must land before the pack
+                    with prof.compute():
+                        optimizer.flush_wgrad()
                 with prof.comm():
                     ddp.finalize_backward()
             if optimizer is not None:
                 with prof.compute():
                     optimizer.step()
-                    if ctx.is_gpu:
+                    if ctx.is_gpu and not flat_opt:
                         from ..models import refresh_all_shadows
                         refresh_all_shadows(seg)
             if probe is not None:

@@ -16,7 +16,7 @@ import torch.nn.functional as F
 
 from ..data import get_dataloader
 from ..engine.common import (GradDivergenceProbe, Meters,
-                             build_optimizer, progress_iter)
+                             build_engine_optimizer, progress_iter)
 from ..parallel import BucketedDataParallel
 from ..parallel.tensor_parallel import replicated_parameters
 from ..parallel.tp_models import build_tp_resnet18
@@ -47,7 +47,12 @@ def train_tp(ctx: DistContext, epochs: int, sample_size: int, logs_dir: str,
     rep_params = replicated_parameters(model)
     ddp = BucketedDataParallel(model, profiler=prof, parameters=rep_params) \
         if rep_params else None
-    optimizer = build_optimizer(model.parameters(), optimizer_name, lr=lr)
+    # GPU: flat-eager fused optimizer (shard params simply stay local slices
+    # of the flat buffer — no cross-rank state); CPU: torch optimizer
+    optimizer = build_engine_optimizer(model, model.parameters(),
+                                       optimizer_name, lr,
+                                       ctx.device if ctx.is_gpu else None)
+    flat_opt = hasattr(optimizer, "flush_wgrad")
     probe = (GradDivergenceProbe(model.parameters())
              if probe_divergence else None)
 
@@ -86,11 +91,14 @@ def train_tp(ctx: DistContext, epochs: int, sample_size: int, logs_dir: str,
                     loss = F.cross_entropy(logits.float(), y)
                 loss.backward()
             if ddp is not None:
+                if flat_opt:  # deferred wgrads must land before the pack
+                    with prof.compute():
+                        optimizer.flush_wgrad()
                 with prof.comm():
                     ddp.finalize_backward()
             with prof.compute():
                 optimizer.step()
-                if ctx.is_gpu:
+                if ctx.is_gpu and not flat_opt:
                     from ..models import refresh_all_shadows
                     refresh_all_shadows(model)
             meters.update(loss, logits, y)
