@@ -57,6 +57,16 @@ def test_dp_flat_engine_learns_and_probes_gpu(tmp_path):
 
 
 @pytest.mark.timeout(300)
+def test_dp_flat_mobilenet_gpu(tmp_path):
+    """auto routes mobilenet_v2 to the flat engine as well: depthwise
+    kernels + inverted residuals must survive capture and train."""
+    from data_parallel_train import run_data_parallel
+    df = run_data_parallel(1, 2, 128, str(tmp_path / "mbn"), batch_size=32,
+                           synthetic=True, model_name="mobilenet_v2")
+    _check(df, 2)
+
+
+@pytest.mark.timeout(300)
 def test_pp_engine_gpu(tmp_path):
     from layer_model_parallel_train import run_model_parallel
     df = run_model_parallel(1, 4, 128, str(tmp_path / "pp"), batch_size=32,
