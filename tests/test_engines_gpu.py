@@ -57,6 +57,25 @@ def test_dp_flat_engine_learns_and_probes_gpu(tmp_path):
 
 
 @pytest.mark.timeout(300)
+def test_dp_flat_forced_comm_gpu(tmp_path, monkeypatch):
+    """HZ_FORCE_COMM=1 exercises the flat engine's world>1 code path on a
+    1-rank RCCL communicator: bf16 pack, eager all-reduce between the
+    fwd/bwd and optimizer graphs, grad_scale consumption.  The loss
+    trajectory must match the no-comm path (sum/world of one rank is
+    identity)."""
+    from data_parallel_train import run_data_parallel
+    monkeypatch.setenv("HZ_FORCE_COMM", "1")
+    df = run_data_parallel(1, 3, 128, str(tmp_path / "fc"), batch_size=32,
+                           synthetic=True, engine="flat", backend="nccl")
+    monkeypatch.delenv("HZ_FORCE_COMM")
+    _check(df, 3)
+    assert (df["comm_time"] > 0).all(), "all-reduce not timed as comm"
+    by_epoch = df.groupby("epoch")["loss"].mean()
+    assert by_epoch.iloc[-1] < by_epoch.iloc[0] * 0.9, \
+        f"forced-comm flat engine not learning: {by_epoch.tolist()}"
+
+
+@pytest.mark.timeout(300)
 def test_dp_flat_mobilenet_gpu(tmp_path):
     """auto routes mobilenet_v2 to the flat engine as well: depthwise
     kernels + inverted residuals must survive capture and train."""
