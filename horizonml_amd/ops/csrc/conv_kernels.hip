@@ -24,6 +24,8 @@
 // statistics (Σy, Σy²) via wave shuffle reduction + one atomicAdd per
 // 16-lane group, so a training conv+BN block is 2 kernels total
 // (north-star fused conv+BN+ReLU).
+#include <cstdlib>
+
 #include "common.h"
 
 struct ConvP {
@@ -792,8 +794,12 @@ __global__ __launch_bounds__(256) void k_stem_conv(
 }
 
 static inline bool stem_eligible(const ConvP& p) {
-  return p.C == 3 && p.R == 7 && p.S == 7 && p.str == 2 && p.pad == 3 &&
-         p.K <= 64 && p.Wo <= 128;
+  static int on = [] {
+    const char* e = getenv("HZ_STEM_DIRECT");
+    return e ? atoi(e) : 1;
+  }();
+  return on && p.C == 3 && p.R == 7 && p.S == 7 && p.str == 2 &&
+         p.pad == 3 && p.K <= 64 && p.Wo <= 128;
 }
 
 static void launch_stem_conv(const bf16* x, const bf16* w, bf16* y,
