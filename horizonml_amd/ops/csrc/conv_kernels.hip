@@ -794,9 +794,14 @@ __global__ __launch_bounds__(256) void k_stem_conv(
 }
 
 static inline bool stem_eligible(const ConvP& p) {
+  // measured SLOWER than the generic t128x64 im2col path everywhere
+  // (CIFAR 31.5 vs 30.5 us incl. BN; @224 238 vs 206): per-block W LDS
+  // staging (64x232 padded image re-built 3.5k times) eats the gather
+  // savings.  Kept as an opt-in experiment + measured negative
+  // (docs/STATUS.md); default OFF.
   static int on = [] {
     const char* e = getenv("HZ_STEM_DIRECT");
-    return e ? atoi(e) : 1;
+    return e ? atoi(e) : 0;
   }();
   return on && p.C == 3 && p.R == 7 && p.S == 7 && p.str == 2 &&
          p.pad == 3 && p.K <= 64 && p.Wo <= 128;
