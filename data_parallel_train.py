@@ -20,13 +20,15 @@ def run_data_parallel(world_size: int, epochs: int, sample_size: int,
                       logs_dir: str = "data_parallel_logs",
                       batch_size: int = 64, model_name: str = "resnet18",
                       backend=None, synthetic=None, lr: float = 1e-3,
-                      optimizer_name: str = "adam", engine: str = "auto"):
+                      optimizer_name: str = "adam", engine: str = "auto",
+                      checkpoint_path=None):
     """Launcher parity with reference ``run_data_parallel``
     (``data_parallel_train.py:233-291``). Returns the combined DataFrame."""
     return run_workers(dp_worker, world_size, epochs, sample_size, logs_dir,
                        timeout_base=120,
                        extra_args=(batch_size, model_name, backend, synthetic,
-                                   lr, optimizer_name, engine))
+                                   lr, optimizer_name, engine,
+                                   checkpoint_path))
 
 
 def main():
@@ -51,6 +53,9 @@ def main():
                     choices=["auto", "eager", "flat"],
                     help="flat = hipGraph/fused fast path (GPU default); "
                          "eager = bucketed-DDP torch-optimizer loop")
+    ap.add_argument("--checkpoint", type=str, default=None,
+                    help="checkpoint file: saved per epoch (rank 0), "
+                         "resumed from when it exists")
     args = ap.parse_args()
     if args.deterministic:
         import os
@@ -58,7 +63,7 @@ def main():
     df = run_data_parallel(args.world_size, args.epochs, args.sample_size,
                            args.logs_dir, args.batch_size, args.model,
                            args.backend, args.synthetic, args.lr,
-                           args.optimizer, args.engine)
+                           args.optimizer, args.engine, args.checkpoint)
     if df is not None:
         print(df.tail(args.world_size).to_string(index=False))
 

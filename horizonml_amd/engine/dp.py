@@ -69,7 +69,8 @@ def train_dp_flat(ctx: DistContext, epochs: int, sample_size: int,
                   optimizer_name: str = "adam",
                   synthetic: Optional[bool] = None, data_dir: str = "./data",
                   probe_divergence: bool = True, log_progress: bool = True,
-                  use_graph: bool = True):
+                  use_graph: bool = True,
+                  checkpoint_path: Optional[str] = None):
     """DP training on the flat fast path — the entrypoint-facing version of
     the bench.py flagship step (the parity entrypoint
     should run the best path we have).
@@ -219,6 +220,17 @@ def train_dp_flat(ctx: DistContext, epochs: int, sample_size: int,
         del snap_master, snap_bufs
         torch.cuda.synchronize()
 
+    # checkpoint/resume (addition over the reference):
+    # restore AFTER capture so warmup/restore cannot clobber resumed state
+    start_epoch = 0
+    if checkpoint_path is not None and os.path.isfile(checkpoint_path):
+        from ..utils.checkpoint import load_checkpoint
+        state = load_checkpoint(checkpoint_path, model, opt, mgr=mgr)
+        start_epoch = int(state.get("epoch", 0))
+        if log_progress and rank == 0:
+            print(f"[dp-flat] resumed from {checkpoint_path} "
+                  f"(epoch {start_epoch})", flush=True)
+
     prof = StepProfiler(dev)
     writer = MetricsWriter(logs_dir, rank, sample_size,
                            with_bandwidth=False, with_gpu=True)
@@ -227,7 +239,7 @@ def train_dp_flat(ctx: DistContext, epochs: int, sample_size: int,
     proc.cpu_percent(interval=None)
 
     count = 0
-    for epoch in range(epochs):
+    for epoch in range(start_epoch, epochs):
         if sampler is not None:
             sampler.set_epoch(epoch)
         with prof.idle():
@@ -295,6 +307,10 @@ def train_dp_flat(ctx: DistContext, epochs: int, sample_size: int,
             print(f"[dp-flat rank0] epoch {epoch + 1}/{epochs} "
                   f"loss={loss_v:.4f} acc={acc_v:.2f}% "
                   f"time={epoch_time:.2f}s", flush=True)
+        if checkpoint_path is not None and rank == 0:
+            from ..utils.checkpoint import save_checkpoint
+            save_checkpoint(checkpoint_path, model, opt, epoch=epoch + 1,
+                            mgr=mgr)
         with prof.idle():
             barrier(ctx)
     return writer.path
@@ -304,7 +320,8 @@ def train_dp(ctx: DistContext, epochs: int, sample_size: int, logs_dir: str,
              batch_size: int = 64, model_name: str = "resnet18",
              lr: float = 1e-3, optimizer_name: str = "adam",
              synthetic: Optional[bool] = None, data_dir: str = "./data",
-             probe_divergence: bool = True, log_progress: bool = True):
+             probe_divergence: bool = True, log_progress: bool = True,
+             checkpoint_path: Optional[str] = None):
     """Run the DP training loop for this rank; writes the per-worker CSV."""
     rank, world = ctx.rank, ctx.world_size
     seed_everything(rank=rank)
@@ -329,7 +346,16 @@ def train_dp(ctx: DistContext, epochs: int, sample_size: int, logs_dir: str,
     proc = psutil.Process()
     proc.cpu_percent(interval=None)  # prime
 
-    for epoch in range(epochs):
+    start_epoch = 0
+    if checkpoint_path is not None and os.path.isfile(checkpoint_path):
+        from ..utils.checkpoint import load_checkpoint
+        state = load_checkpoint(checkpoint_path, model, optimizer)
+        start_epoch = int(state.get("epoch", 0))
+        if log_progress and rank == 0:
+            print(f"[dp] resumed from {checkpoint_path} "
+                  f"(epoch {start_epoch})", flush=True)
+
+    for epoch in range(start_epoch, epochs):
         if sampler is not None:
             sampler.set_epoch(epoch)
         with prof.idle():
@@ -384,6 +410,10 @@ def train_dp(ctx: DistContext, epochs: int, sample_size: int, logs_dir: str,
             print(f"[dp rank0] epoch {epoch + 1}/{epochs} "
                   f"loss={loss_v:.4f} acc={acc_v:.2f}% "
                   f"time={epoch_time:.2f}s", flush=True)
+        if checkpoint_path is not None and rank == 0:
+            from ..utils.checkpoint import save_checkpoint
+            save_checkpoint(checkpoint_path, model, optimizer,
+                            epoch=epoch + 1)
         with prof.idle():
             barrier(ctx)
     return writer.path
@@ -393,11 +423,14 @@ def dp_worker(rank: int, world_size: int, epochs: int, sample_size: int,
               port: int, logs_dir: str, batch_size: int = 64,
               model_name: str = "resnet18", backend: Optional[str] = None,
               synthetic: Optional[bool] = None, lr: float = 1e-3,
-              optimizer_name: str = "adam", engine: str = "auto"):
+              optimizer_name: str = "adam", engine: str = "auto",
+              checkpoint_path: Optional[str] = None):
     """Spawned worker entry (reference ``data_parallel_train.py:192-230``).
 
     ``engine``: ``flat`` (default on GPU via ``auto``) runs the hipGraph /
-    fused-step fast path; ``eager`` the bucketed-DDP torch-optimizer loop."""
+    fused-step fast path; ``eager`` the bucketed-DDP torch-optimizer loop.
+    ``checkpoint_path``: save per epoch (rank 0) and resume when the file
+    exists — an extension over the reference."""
     ctx = setup_distributed(rank, world_size, port, backend=backend)
     try:
         if ctx.is_gpu and model_name.startswith("resnet"):
@@ -409,11 +442,13 @@ def dp_worker(rank: int, world_size: int, epochs: int, sample_size: int,
             train_dp_flat(ctx, epochs, sample_size, logs_dir,
                           batch_size=batch_size, model_name=model_name,
                           synthetic=synthetic, lr=lr,
-                          optimizer_name=optimizer_name)
+                          optimizer_name=optimizer_name,
+                          checkpoint_path=checkpoint_path)
         else:
             train_dp(ctx, epochs, sample_size, logs_dir,
                      batch_size=batch_size, model_name=model_name,
                      synthetic=synthetic, lr=lr,
-                     optimizer_name=optimizer_name)
+                     optimizer_name=optimizer_name,
+                     checkpoint_path=checkpoint_path)
     finally:
         teardown_distributed(ctx)
