@@ -57,6 +57,28 @@ def test_dp_flat_engine_learns_and_probes_gpu(tmp_path):
 
 
 @pytest.mark.timeout(300)
+def test_dp_flat_checkpoint_resume_gpu(tmp_path):
+    """Flat engine checkpoint: master/shadow/optimizer state round-trips
+    through in-place buffer copies AFTER graph capture — replays must use
+    the resumed weights."""
+    import pandas as pd
+
+    from data_parallel_train import run_data_parallel
+    logs = str(tmp_path / "l1")
+    ckpt = str(tmp_path / "flat.ckpt")
+    run_data_parallel(1, 2, 128, logs, batch_size=32, synthetic=True,
+                      engine="flat", checkpoint_path=ckpt)
+    df1 = pd.read_csv(f"{logs}/worker_0_samples_128.csv")
+    logs2 = str(tmp_path / "l2")
+    run_data_parallel(1, 4, 128, logs2, batch_size=32, synthetic=True,
+                      engine="flat", checkpoint_path=ckpt)
+    df2 = pd.read_csv(f"{logs2}/worker_0_samples_128.csv")
+    assert list(df2["epoch"]) == [3, 4]
+    assert df2["loss"].iloc[-1] < df1["loss"].iloc[0], \
+        "resumed run did not continue from trained weights"
+
+
+@pytest.mark.timeout(300)
 def test_dp_flat_forced_comm_gpu(tmp_path, monkeypatch):
     """HZ_FORCE_COMM=1 exercises the flat engine's world>1 code path on a
     1-rank RCCL communicator: bf16 pack, eager all-reduce between the
