@@ -9,6 +9,7 @@ backward relay trains *every* stage (each rank owns an optimizer).
 """
 from __future__ import annotations
 
+import os
 import time
 from typing import Optional
 
@@ -36,7 +37,8 @@ def train_pp(ctx: DistContext, epochs: int, sample_size: int, logs_dir: str,
              stage_idx: Optional[int] = None, dp_group=None,
              data_rank: Optional[int] = None, data_world: Optional[int] = None,
              num_classes: int = 10, image_size: int = 32,
-             log_progress: bool = True, probe_divergence: bool = True):
+             log_progress: bool = True, probe_divergence: bool = True,
+             checkpoint_path: Optional[str] = None):
     rank, world = ctx.rank, ctx.world_size
     n_stages = n_stages or world
     stage_idx = stage_idx if stage_idx is not None else rank
@@ -97,13 +99,37 @@ def train_pp(ctx: DistContext, epochs: int, sample_size: int, logs_dir: str,
     proc = psutil.Process()
     proc.cpu_percent(interval=None)
 
+    # per-RANK checkpoints (each stage owns distinct parameters); ranks
+    # agree on MIN(resumed epoch) so a kill between per-rank saves cannot
+    # desynchronize the relay
+    start_epoch = 0
+    ckpt_file = (f"{checkpoint_path}.rank{rank}"
+                 if checkpoint_path is not None else None)
+    if ckpt_file is not None:
+        if os.path.isfile(ckpt_file) and optimizer is not None:
+            from ..utils.checkpoint import load_checkpoint
+            state = load_checkpoint(ckpt_file, seg, optimizer)
+            start_epoch = int(state.get("epoch", 0))
+        elif os.path.isfile(ckpt_file):
+            from ..utils.checkpoint import load_checkpoint
+            state = load_checkpoint(ckpt_file, seg)
+            start_epoch = int(state.get("epoch", 0))
+        import torch.distributed as dist
+        if dist.is_available() and dist.is_initialized():
+            t = torch.tensor([float(start_epoch)],
+                             device=ctx.device if ctx.is_gpu else "cpu")
+            dist.all_reduce(t, op=dist.ReduceOp.MIN)
+            start_epoch = int(t.item())
+        if log_progress and start_epoch > 0 and rank == 0:
+            print(f"[pp] resumed at epoch {start_epoch}", flush=True)
+
     def loss_fn(logits, y):
         if logits.is_cuda:
             from ..models._functional_gpu import cross_entropy
             return cross_entropy(logits, y)
         return F.cross_entropy(logits.float(), y)
 
-    for epoch in range(epochs):
+    for epoch in range(start_epoch, epochs):
         with prof.idle():
             barrier(ctx)
         epoch_start = time.time()
@@ -174,6 +200,9 @@ def train_pp(ctx: DistContext, epochs: int, sample_size: int, logs_dir: str,
         if log_progress and stage.is_last:
             print(f"[pp stage{stage_idx}] epoch {epoch + 1}/{epochs} "
                   f"loss={loss_v:.4f} time={epoch_time:.2f}s", flush=True)
+        if ckpt_file is not None:
+            from ..utils.checkpoint import save_checkpoint
+            save_checkpoint(ckpt_file, seg, optimizer, epoch=epoch + 1)
         barrier(ctx)
     train_pp.last_segment = seg  # exposed for tests (DP-sync verification)
     return writer.path
@@ -213,7 +242,8 @@ def pp_worker(rank: int, world_size: int, epochs: int, sample_size: int,
               port: int, logs_dir: str, batch_size: int = 64,
               model_name: str = "resnet18", backend: Optional[str] = None,
               synthetic: Optional[bool] = None, lr: float = 1e-3,
-              optimizer_name: str = "adam", microbatches: int = 1):
+              optimizer_name: str = "adam", microbatches: int = 1,
+              checkpoint_path: Optional[str] = None):
     ctx = setup_distributed(rank, world_size, port, backend=backend)
     try:
         if ctx.is_gpu and model_name.startswith("resnet"):
@@ -221,6 +251,7 @@ def pp_worker(rank: int, world_size: int, epochs: int, sample_size: int,
             _ops.extension()
         train_pp(ctx, epochs, sample_size, logs_dir, batch_size=batch_size,
                  model_name=model_name, synthetic=synthetic, lr=lr,
-                 optimizer_name=optimizer_name, microbatches=microbatches)
+                 optimizer_name=optimizer_name, microbatches=microbatches,
+                 checkpoint_path=checkpoint_path)
     finally:
         teardown_distributed(ctx)

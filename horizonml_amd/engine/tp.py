@@ -32,7 +32,8 @@ def train_tp(ctx: DistContext, epochs: int, sample_size: int, logs_dir: str,
              batch_size: int = 64, lr: float = 1e-3,
              optimizer_name: str = "adam", synthetic: Optional[bool] = None,
              data_dir: str = "./data", tp_mode: str = "fc",
-             probe_divergence: bool = True, log_progress: bool = True):
+             probe_divergence: bool = True, log_progress: bool = True,
+             checkpoint_path: Optional[str] = None):
     rank, world = ctx.rank, ctx.world_size
     seed_everything(rank=0)   # replicated params identical across ranks
     torch.manual_seed(1234 + rank)  # shard params differ per rank by design
@@ -63,7 +64,25 @@ def train_tp(ctx: DistContext, epochs: int, sample_size: int, logs_dir: str,
     proc = psutil.Process()
     proc.cpu_percent(interval=None)
 
-    for epoch in range(epochs):
+    # per-RANK checkpoints (shard parameters differ per rank); MIN-epoch
+    # agreement keeps the collective schedule aligned after a partial save
+    import os as _os
+    start_epoch = 0
+    ckpt_file = (f"{checkpoint_path}.rank{rank}"
+                 if checkpoint_path is not None else None)
+    if ckpt_file is not None and _os.path.isfile(ckpt_file):
+        from ..utils.checkpoint import load_checkpoint
+        state = load_checkpoint(ckpt_file, model, optimizer)
+        start_epoch = int(state.get("epoch", 0))
+    if ckpt_file is not None:
+        import torch.distributed as dist
+        if dist.is_available() and dist.is_initialized():
+            t = torch.tensor([float(start_epoch)],
+                             device=ctx.device if ctx.is_gpu else "cpu")
+            dist.all_reduce(t, op=dist.ReduceOp.MIN)
+            start_epoch = int(t.item())
+
+    for epoch in range(start_epoch, epochs):
         with prof.idle():
             barrier(ctx)
         epoch_start = time.time()
@@ -123,6 +142,9 @@ def train_tp(ctx: DistContext, epochs: int, sample_size: int, logs_dir: str,
         if log_progress and rank == 0:
             print(f"[tp rank0] epoch {epoch + 1}/{epochs} loss={loss_v:.4f} "
                   f"acc={acc_v:.2f}% time={epoch_time:.2f}s", flush=True)
+        if ckpt_file is not None:
+            from ..utils.checkpoint import save_checkpoint
+            save_checkpoint(ckpt_file, model, optimizer, epoch=epoch + 1)
         barrier(ctx)
     return writer.path
 
@@ -131,7 +153,8 @@ def tp_worker(rank: int, world_size: int, epochs: int, sample_size: int,
               port: int, logs_dir: str, batch_size: int = 64,
               backend: Optional[str] = None,
               synthetic: Optional[bool] = None, lr: float = 1e-3,
-              optimizer_name: str = "adam", tp_mode: str = "fc"):
+              optimizer_name: str = "adam", tp_mode: str = "fc",
+              checkpoint_path: Optional[str] = None):
     ctx = setup_distributed(rank, world_size, port, backend=backend)
     try:
         if ctx.is_gpu:
@@ -139,6 +162,6 @@ def tp_worker(rank: int, world_size: int, epochs: int, sample_size: int,
             _ops.extension()
         train_tp(ctx, epochs, sample_size, logs_dir, batch_size=batch_size,
                  synthetic=synthetic, lr=lr, optimizer_name=optimizer_name,
-                 tp_mode=tp_mode)
+                 tp_mode=tp_mode, checkpoint_path=checkpoint_path)
     finally:
         teardown_distributed(ctx)

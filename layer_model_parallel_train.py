@@ -22,13 +22,15 @@ def run_model_parallel(world_size: int, epochs: int, sample_size: int,
                        logs_dir: str = "model_parallel_logs",
                        batch_size: int = 64, model_name: str = "resnet18",
                        backend=None, synthetic=None, lr: float = 1e-3,
-                       optimizer_name: str = "adam", microbatches: int = 1):
+                       optimizer_name: str = "adam", microbatches: int = 1,
+                       checkpoint_path=None):
     """Launcher parity with reference ``run_model_parallel``
     (``layer_model_parallel_train.py:365-423``)."""
     return run_workers(pp_worker, world_size, epochs, sample_size, logs_dir,
                        timeout_base=120,
                        extra_args=(batch_size, model_name, backend, synthetic,
-                                   lr, optimizer_name, microbatches))
+                                   lr, optimizer_name, microbatches,
+                                   checkpoint_path))
 
 
 def main():
@@ -48,6 +50,9 @@ def main():
                          "remain reproducible only to bf16/atomic rounding)")
     ap.add_argument("--optimizer", type=str, default="adam",
                     choices=["adam", "sgd"])
+    ap.add_argument("--checkpoint", type=str, default=None,
+                    help="checkpoint base path: per-rank files saved per "
+                         "epoch, resumed when present")
     ap.add_argument("--microbatches", type=int, default=1,
                     help="pipeline microbatches per step (bubble reduction)")
     args = ap.parse_args()
@@ -57,7 +62,8 @@ def main():
     df = run_model_parallel(args.world_size, args.epochs, args.sample_size,
                             args.logs_dir, args.batch_size, args.model,
                             args.backend, args.synthetic, args.lr,
-                            args.optimizer, args.microbatches)
+                            args.optimizer, args.microbatches,
+                            args.checkpoint)
     if df is not None:
         print(df.tail(args.world_size).to_string(index=False))
 

@@ -50,6 +50,9 @@ def main():
                     choices=["fc", "full"],
                     help="fc: reference-parity classifier shard; "
                          "full: sharded conv2d + all-gather")
+    ap.add_argument("--checkpoint", type=str, default=None,
+                    help="checkpoint base path: per-rank files saved per "
+                         "epoch, resumed when present")
     args = ap.parse_args()
     if args.deterministic:
         import os
@@ -57,7 +60,7 @@ def main():
     df = run_tensor_parallel(args.world_size, args.epochs, args.sample_size,
                              args.logs_dir, args.batch_size, args.backend,
                              args.synthetic, args.lr, args.optimizer,
-                             args.tp_mode)
+                             args.tp_mode, args.checkpoint)
     if df is not None:
         print(df.tail(args.world_size).to_string(index=False))
 

@@ -67,6 +67,31 @@ def test_dp_entrypoint_checkpoint_resume(tmp_path):
     assert df2["loss"].iloc[-1] < df1["loss"].iloc[0]
 
 
+def test_pp_entrypoint_checkpoint_resume(tmp_path):
+    """Pipeline checkpoint: per-RANK files (each stage owns distinct
+    params) with MIN-epoch agreement; a relaunch resumes every stage at
+    the same epoch."""
+    import pandas as pd
+
+    from layer_model_parallel_train import run_model_parallel
+
+    logs = str(tmp_path / "logs")
+    ckpt = str(tmp_path / "pp.ckpt")
+    run_model_parallel(world_size=2, epochs=1, sample_size=32,
+                       logs_dir=logs, batch_size=16, backend="gloo",
+                       synthetic=True, checkpoint_path=ckpt)
+    import os
+    assert os.path.isfile(ckpt + ".rank0") and os.path.isfile(
+        ckpt + ".rank1"), "per-rank checkpoint files missing"
+    logs2 = str(tmp_path / "logs2")
+    run_model_parallel(world_size=2, epochs=3, sample_size=32,
+                       logs_dir=logs2, batch_size=16, backend="gloo",
+                       synthetic=True, checkpoint_path=ckpt)
+    df = pd.read_csv(f"{logs2}/worker_1_samples_32.csv")
+    assert list(df["epoch"]) == [2, 3], \
+        f"pipeline resume wrong epochs: {list(df['epoch'])}"
+
+
 def test_fused_optimizer_state_roundtrip(tmp_path):
     """HorizonAdam/HorizonSGD moments survive save/load (they
     were silently dropped).  Uses a stand-in manager so the state-dict path
