@@ -123,5 +123,6 @@ def test_bench_protocol_two_ranks():
     rec = json.loads(out[0][3])
     assert rec["n_gpus"] == 2
     assert rec["config"]["global_batch"] == 128
-    assert rec["value"] == pytest.approx(128 * 5 / out[0][2], rel=1e-6)
+    # record value is rounded to 2 decimals by the contract
+    assert rec["value"] == pytest.approx(128 * 5 / out[0][2], abs=0.006)
     assert rec["config"]["comm"] == "bucketed4"
