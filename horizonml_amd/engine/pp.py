@@ -215,7 +215,8 @@ def hybrid_worker(rank: int, world_size: int, epochs: int, sample_size: int,
                   synthetic: Optional[bool] = None, lr: float = 1e-3,
                   optimizer_name: str = "adam", microbatches: int = 1,
                   dp_size: int = 2, pp_size: int = 4,
-                  num_classes: int = 10, image_size: int = 32):
+                  num_classes: int = 10, image_size: int = 32,
+                  checkpoint_path: Optional[str] = None):
     """Hybrid DP×PP worker (BASELINE.json config #5: ResNet50, 2×4 on 8
     GPUs).  Rank layout ``rank = dp_rank * pp_size + pp_stage`` keeps each
     pipeline chain on contiguous (xGMI-adjacent) GPUs; the DP replica sync
@@ -233,7 +234,8 @@ def hybrid_worker(rank: int, world_size: int, epochs: int, sample_size: int,
                  optimizer_name=optimizer_name, microbatches=microbatches,
                  group=pp_group, n_stages=pp_size, stage_idx=pp_stage,
                  dp_group=dp_group, data_rank=dp_rank, data_world=dp_size,
-                 num_classes=num_classes, image_size=image_size)
+                 num_classes=num_classes, image_size=image_size,
+                 checkpoint_path=checkpoint_path)
     finally:
         teardown_distributed(ctx)
 

@@ -1358,7 +1358,16 @@ __global__ void k_gdiv_finalize(float* __restrict__ sumsq,
 }
 
 // ------------------------------------------------------------- launchers --
-static inline int gsz(long total, int block = 256, int cap = 4096) {
+static inline int gsz_cap() {
+  static int cap = [] {
+    const char* e = getenv("HZ_GSZ_CAP");
+    return e ? atoi(e) : 4096;
+  }();
+  return cap;
+}
+
+static inline int gsz(long total, int block = 256, int cap = 0) {
+  if (cap == 0) cap = gsz_cap();
   long g = (total + block - 1) / block;
   if (g > cap) g = cap;
   if (g < 1) g = 1;

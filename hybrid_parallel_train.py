@@ -30,13 +30,15 @@ def run_hybrid_parallel(dp_size: int, pp_size: int, epochs: int,
                         batch_size: int = 64, model_name: str = "resnet50",
                         backend=None, synthetic=None, lr: float = 1e-3,
                         optimizer_name: str = "adam", microbatches: int = 1,
-                        num_classes: int = 10, image_size: int = 32):
+                        num_classes: int = 10, image_size: int = 32,
+                        checkpoint_path=None):
     world_size = dp_size * pp_size
     return run_workers(hybrid_worker, world_size, epochs, sample_size,
                        logs_dir, timeout_base=240,
                        extra_args=(batch_size, model_name, backend, synthetic,
                                    lr, optimizer_name, microbatches,
-                                   dp_size, pp_size, num_classes, image_size))
+                                   dp_size, pp_size, num_classes, image_size,
+                                   checkpoint_path))
 
 
 def main():
@@ -61,6 +63,8 @@ def main():
     ap.add_argument("--num_classes", type=int, default=10)
     ap.add_argument("--image_size", type=int, default=32,
                     help="224 for the ImageNet-shaped hybrid config")
+    ap.add_argument("--checkpoint", type=str, default=None,
+                    help="checkpoint base path (per-rank files)")
     args = ap.parse_args()
     if args.deterministic:
         import os
@@ -70,7 +74,7 @@ def main():
                              args.batch_size, args.model, args.backend,
                              args.synthetic, args.lr, args.optimizer,
                              args.microbatches, args.num_classes,
-                             args.image_size)
+                             args.image_size, args.checkpoint)
     if df is not None:
         print(df.tail(args.dp_size * args.pp_size).to_string(index=False))
 
