@@ -21,14 +21,14 @@ def run_data_parallel(world_size: int, epochs: int, sample_size: int,
                       batch_size: int = 64, model_name: str = "resnet18",
                       backend=None, synthetic=None, lr: float = 1e-3,
                       optimizer_name: str = "adam", engine: str = "auto",
-                      checkpoint_path=None):
+                      checkpoint_path=None, per_step_barrier: bool = False):
     """Launcher parity with reference ``run_data_parallel``
     (``data_parallel_train.py:233-291``). Returns the combined DataFrame."""
     return run_workers(dp_worker, world_size, epochs, sample_size, logs_dir,
                        timeout_base=120,
                        extra_args=(batch_size, model_name, backend, synthetic,
                                    lr, optimizer_name, engine,
-                                   checkpoint_path))
+                                   checkpoint_path, per_step_barrier))
 
 
 def main():
@@ -56,6 +56,10 @@ def main():
     ap.add_argument("--checkpoint", type=str, default=None,
                     help="checkpoint file: saved per epoch (rank 0), "
                          "resumed from when it exists")
+    ap.add_argument("--per_step_barrier", action="store_true",
+                    help="restore the reference's full-world barrier after "
+                         "every step (exact idle_time semantics; the "
+                         "default barriers per epoch)")
     args = ap.parse_args()
     if args.deterministic:
         import os
@@ -63,7 +67,8 @@ def main():
     df = run_data_parallel(args.world_size, args.epochs, args.sample_size,
                            args.logs_dir, args.batch_size, args.model,
                            args.backend, args.synthetic, args.lr,
-                           args.optimizer, args.engine, args.checkpoint)
+                           args.optimizer, args.engine, args.checkpoint,
+                           args.per_step_barrier)
     if df is not None:
         print(df.tail(args.world_size).to_string(index=False))
 

@@ -70,7 +70,8 @@ def train_dp_flat(ctx: DistContext, epochs: int, sample_size: int,
                   synthetic: Optional[bool] = None, data_dir: str = "./data",
                   probe_divergence: bool = True, log_progress: bool = True,
                   use_graph: bool = True,
-                  checkpoint_path: Optional[str] = None):
+                  checkpoint_path: Optional[str] = None,
+                  per_step_barrier: bool = False):
     """DP training on the flat fast path — the entrypoint-facing version of
     the bench.py flagship step (the parity entrypoint
     should run the best path we have).
@@ -279,6 +280,9 @@ def train_dp_flat(ctx: DistContext, epochs: int, sample_size: int,
                                        * comm_buf.element_size())
                 with prof.compute():
                     opt_step()
+            if per_step_barrier:  # reference data_parallel_train.py:150-152
+                with prof.idle():
+                    barrier(ctx)
             count += y.shape[0]
             n_steps += 1
             prof.step_end()
@@ -321,8 +325,15 @@ def train_dp(ctx: DistContext, epochs: int, sample_size: int, logs_dir: str,
              lr: float = 1e-3, optimizer_name: str = "adam",
              synthetic: Optional[bool] = None, data_dir: str = "./data",
              probe_divergence: bool = True, log_progress: bool = True,
-             checkpoint_path: Optional[str] = None):
-    """Run the DP training loop for this rank; writes the per-worker CSV."""
+             checkpoint_path: Optional[str] = None,
+             per_step_barrier: bool = False):
+    """Run the DP training loop for this rank; writes the per-worker CSV.
+
+    ``per_step_barrier=True`` restores the reference's full-world barrier
+    after EVERY step (``data_parallel_train.py:150-152``) so ``idle_time``
+    measures per-step peer skew exactly as the reference's does; the
+    default barriers per epoch (a per-step barrier would dominate ~1 ms
+    GPU steps — documented deviation)."""
     rank, world = ctx.rank, ctx.world_size
     seed_everything(rank=rank)
     loader, sampler = get_dataloader(rank, world, batch_size, sample_size,
@@ -390,6 +401,9 @@ def train_dp(ctx: DistContext, epochs: int, sample_size: int, logs_dir: str,
             meters.update(loss, logits, y)
             if probe is not None:
                 probe.step()
+            if per_step_barrier:  # reference data_parallel_train.py:150-152
+                with prof.idle():
+                    barrier(ctx)
             prof.step_end()
 
         t = prof.epoch_end()  # syncs: epoch_time includes the GPU tail
@@ -424,7 +438,8 @@ def dp_worker(rank: int, world_size: int, epochs: int, sample_size: int,
               model_name: str = "resnet18", backend: Optional[str] = None,
               synthetic: Optional[bool] = None, lr: float = 1e-3,
               optimizer_name: str = "adam", engine: str = "auto",
-              checkpoint_path: Optional[str] = None):
+              checkpoint_path: Optional[str] = None,
+              per_step_barrier: bool = False):
     """Spawned worker entry (reference ``data_parallel_train.py:192-230``).
 
     ``engine``: ``flat`` (default on GPU via ``auto``) runs the hipGraph /
@@ -443,12 +458,14 @@ def dp_worker(rank: int, world_size: int, epochs: int, sample_size: int,
                           batch_size=batch_size, model_name=model_name,
                           synthetic=synthetic, lr=lr,
                           optimizer_name=optimizer_name,
-                          checkpoint_path=checkpoint_path)
+                          checkpoint_path=checkpoint_path,
+                          per_step_barrier=per_step_barrier)
         else:
             train_dp(ctx, epochs, sample_size, logs_dir,
                      batch_size=batch_size, model_name=model_name,
                      synthetic=synthetic, lr=lr,
                      optimizer_name=optimizer_name,
-                     checkpoint_path=checkpoint_path)
+                     checkpoint_path=checkpoint_path,
+                     per_step_barrier=per_step_barrier)
     finally:
         teardown_distributed(ctx)

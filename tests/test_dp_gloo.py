@@ -125,6 +125,22 @@ def test_deferred_reduction_microbatch_grads():
 
 
 @pytest.mark.timeout(300)
+def test_dp_per_step_barrier_flag(tmp_path):
+    """--per_step_barrier restores the reference's per-step full-world
+    barrier: 2-rank gloo run must complete and account barrier waits in
+    idle_time."""
+    import pandas as pd
+
+    from data_parallel_train import run_data_parallel
+    df = run_data_parallel(world_size=2, epochs=1, sample_size=64,
+                           logs_dir=str(tmp_path), batch_size=32,
+                           backend="gloo", synthetic=True,
+                           per_step_barrier=True)
+    assert df is not None and len(df) == 2
+    assert (df["idle_time"] > 0).all()
+
+
+@pytest.mark.timeout(300)
 def test_dp_entrypoint_end_to_end(tmp_path):
     from data_parallel_train import run_data_parallel
     df = run_data_parallel(world_size=2, epochs=1, sample_size=64,
