@@ -162,6 +162,37 @@ def test_column_parallel_linear_correct():
                               b.grad[r * 2:(r + 1) * 2], atol=1e-5)
 
 
+def _rs_worker(rank, world, port, q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    ctx = setup_distributed(rank, world, port, backend="gloo")
+    from horizonml_amd.parallel.tensor_parallel import (
+        reduce_scatter_to_parallel)
+    g = torch.Generator().manual_seed(40 + rank)
+    x = torch.randn(4, 8, generator=g).requires_grad_(True)
+    y = reduce_scatter_to_parallel(x, dim=-1)   # rank's 4-wide shard of sum
+    gy = torch.full((4, 4), float(rank + 1))
+    y.backward(gy)
+    q.put((rank, (y.detach().tolist(), x.grad.tolist())))
+    teardown_distributed(ctx)
+
+
+def test_reduce_scatter_autograd():
+    """reduce-scatter forward = sum-then-shard; backward = all-gather of
+    the shard grads (the north-star C5/C6 collective pairing)."""
+    out = _spawn(_rs_worker, 2)
+    xs = [torch.randn(4, 8, generator=torch.Generator().manual_seed(40 + r))
+          for r in (0, 1)]
+    total = xs[0] + xs[1]
+    for r in (0, 1):
+        got_y = torch.tensor(out[r][0])
+        assert torch.allclose(got_y, total[:, r * 4:(r + 1) * 4], atol=1e-6)
+        # backward: grad of x = concat of both ranks' shard grads (1s, 2s)
+        got_gx = torch.tensor(out[r][1])
+        expect = torch.cat([torch.full((4, 4), 1.0),
+                            torch.full((4, 4), 2.0)], dim=1)
+        assert torch.equal(got_gx, expect)
+
+
 def _tp_resnet_worker(rank, world, port, q):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     ctx = setup_distributed(rank, world, port, backend="gloo")
