@@ -252,6 +252,23 @@ def test_pipeline_uneven_microbatches():
     assert out[1][1] == 10, "sample count wrong across uneven microbatches"
 
 
+def test_tp_full_ws4_end_to_end(tmp_path):
+    """BASELINE config #4 shape: tensor-parallel world_size=4 with sharded
+    conv2d (--tp_mode full) — all ranks train, identical data, identical
+    loss trajectory across ranks (gathered activations)."""
+    import pandas as pd
+
+    from tensor_parallel_train import run_tensor_parallel
+    df = run_tensor_parallel(world_size=4, epochs=1, sample_size=32,
+                             logs_dir=str(tmp_path), batch_size=16,
+                             backend="gloo", synthetic=True, tp_mode="full")
+    assert df is not None
+    assert sorted(df["worker"].unique().tolist()) == list(range(4))
+    # identical inputs + gathered activations => same loss on every rank
+    losses = df.groupby("worker")["loss"].last()
+    assert losses.max() - losses.min() < 1e-4, losses.tolist()
+
+
 def test_pipeline_8_stages_end_to_end(tmp_path):
     """BASELINE config #3 shape: ResNet18 split into 8 pipeline stages
     (block granularity — the reference's 5-group cap lifted), full
