@@ -29,8 +29,7 @@ void launch_gemm_bf16(const void*, const void*, void*, int, int, int,
 int wgrad_msplit(ConvP);
 void launch_wgrad(const void*, const void*, float*, float*, ConvP,
                   hipStream_t);
-void launch_wgrad_batched(const void*, int, hipStream_t);
-void launch_wgrad_batched_t128(const void*, int, hipStream_t);
+void launch_wgrad_batched_t(const void*, int, int, int, hipStream_t);
 void launch_wgrad_reduce_batched(const void*, int, hipStream_t);
 void launch_bn_apply(const void*, const void*, void*, const float*,
                      const float*, const float*, float*, float*, float*,
@@ -269,15 +268,34 @@ void flush_one_group(int group_lo, int group_hi) {
                            pend[0].x.options().dtype(torch::kFloat32));
   float* ws_base = ws_total > 0 ? g_wgrad_ws.data_ptr<float>() : nullptr;
 
-  // two tile classes: 128-wide k3 tiles for big-Kd large-M convs (halves
-  // the Dz re-read and doubles MFMA per staging write), 64 otherwise —
-  // separate launches so each keeps its own LDS footprint/occupancy
-  for (int tk3 : {64, 128}) {
+  // four tile classes (tk3, tko ∈ {64,128}): 128-wide k3 for big-Kd
+  // large-M tasks halves the Dz re-read; 128-wide ko for K>=128 tasks
+  // halves the X re-read — separate launches keep per-class LDS/occupancy
+  auto tk3_of = [](const ConvP& p) {
+    static int kd_min = [] {
+      const char* e = getenv("HZ_WG_TK3_KD");
+      return e ? atoi(e) : 512;
+    }();
+    static int m_min = [] {
+      const char* e = getenv("HZ_WG_TK3_M");
+      return e ? atoi(e) : 8192;
+    }();
+    return (p.Kd >= kd_min && p.M >= m_min) ? 128 : 64;
+  };
+  auto tko_of = [](const ConvP& p) {
+    static int k_min = [] {
+      const char* e = getenv("HZ_WG_TKO_K");
+      return e ? atoi(e) : 128;
+    }();
+    return p.K >= k_min ? 128 : 64;
+  };
+  for (int cls = 0; cls < 4; cls++) {
+    const int tk3 = (cls & 1) ? 128 : 64;
+    const int tko = (cls & 2) ? 128 : 64;
     std::vector<int> idx;
-    for (int i = 0; i < n; i++) {
-      bool big = pend[i].p.Kd >= 512 && pend[i].p.M >= 8192;
-      if ((tk3 == 128) == big) idx.push_back(i);
-    }
+    for (int i = 0; i < n; i++)
+      if (tk3_of(pend[i].p) == tk3 && tko_of(pend[i].p) == tko)
+        idx.push_back(i);
     for (size_t lo = 0; lo < idx.size(); lo += WG_MAX_TASKS) {
       WgradBatchArgs a{};
       a.n = (int)std::min((size_t)WG_MAX_TASKS, idx.size() - lo);
@@ -296,15 +314,12 @@ void flush_one_group(int group_lo, int group_hi) {
         t.mchunk = mchunk[i];
         t.msplit = msplit[i];
         t.tx = cdiv_i(pw.p.Kd, tk3);
-        t.ty = cdiv_i(pw.p.K, 64);
+        t.ty = cdiv_i(pw.p.K, tko);
         t.base = blocks;
         t.vec = (pw.p.C % 8) == 0;
         blocks += t.tx * t.ty * t.msplit;
       }
-      if (tk3 == 128)
-        launch_wgrad_batched_t128(&a, blocks, st);
-      else
-        launch_wgrad_batched(&a, blocks, st);
+      launch_wgrad_batched_t(&a, blocks, tk3, tko, st);
     }
   }
 

@@ -362,27 +362,28 @@ __global__ __launch_bounds__(256) void k_conv_mfma(
 // TK3: k3-tile width (64 default; 128 for big-Kd large-M convs — halves
 // the Dz re-read traffic and doubles MFMA per staging write).  The out
 // image keeps an odd f32 row stride (65 / 131) for conflict-free stores.
-template <int TK3>
+template <int TK3, int TKO = 64>
 union WgradSmemT {
   struct {
     bf16 At[TK3 * LDW];  // [k3][m]
-    bf16 Dt[64 * LDW];   // [ko][m]
+    bf16 Dt[TKO * LDW];  // [ko][m]
   } s;
   float out[64][TK3 == 64 ? 65 : 131];  // transposed epilogue staging
 };
-using WgradSmem = WgradSmemT<64>;
+using WgradSmem = WgradSmemT<64, 64>;
 
-template <bool VECA, int TK3 = 64>
+template <bool VECA, int TK3 = 64, int TKO = 64>
 DEV void wgrad_tile(const bf16* __restrict__ X, const bf16* __restrict__ Dz,
                     float* __restrict__ out, const ConvP& p,
                     const MagicP& mg, int Ntot, int mchunk, int tx, int ty,
-                    int z, bool split, WgradSmemT<TK3>& smem) {
+                    int z, bool split, WgradSmemT<TK3, TKO>& smem) {
   constexpr int AC = TK3 / 64;   // A staging chunks per thread
+  constexpr int DC = TKO / 64;   // Dz staging chunks per thread
   constexpr int MI = TK3 / 32;   // k3 fragments per wave
-  constexpr int OSTR = TK3 == 64 ? 65 : 131;
+  constexpr int NI = TKO / 32;   // ko fragments per wave
   bf16* At = smem.s.At;   // [k3][m]
   bf16* Dt = smem.s.Dt;   // [ko][m]
-  const int k3_0 = tx * TK3, n0 = ty * 64;
+  const int k3_0 = tx * TK3, n0 = ty * TKO;
   const int mbeg = z * mchunk;
   const int mend = min(p.M, mbeg + mchunk);
   const int tid = threadIdx.x;
@@ -391,16 +392,19 @@ DEV void wgrad_tile(const bf16* __restrict__ X, const bf16* __restrict__ Dz,
   const int wr = wave >> 1, wc = wave & 1;
   const int fr = lane & 15, fk = lane >> 4;
 
-  f32x4 acc[MI][2] = {};
+  f32x4 acc[MI][NI] = {};
 
-  V8 a_nx[AC], d_nx;
+  V8 a_nx[AC], d_nx[DC];
 #pragma unroll
   for (int j = 0; j < AC; j++)
     a_nx[j] = load8_a<1, VECA>(X, p, mg, mbeg + sm, k3_0 + j * 64 + sv);
-  {
-    int m = mbeg + sm, n = n0 + sv;
-    if (m < p.M && n < Ntot) d_nx.u = *(const uint4*)(Dz + (long)m * Ntot + n);
-    else d_nx.u = uint4{0, 0, 0, 0};
+#pragma unroll
+  for (int j = 0; j < DC; j++) {
+    int m = mbeg + sm, n = n0 + j * 64 + sv;
+    if (m < p.M && n < Ntot)
+      d_nx[j].u = *(const uint4*)(Dz + (long)m * Ntot + n);
+    else
+      d_nx[j].u = uint4{0, 0, 0, 0};
   }
   const int smlo = sm & 7, smhi = sm >> 3;
   for (int m0 = mbeg; m0 < mend; m0 += 32) {
@@ -415,74 +419,87 @@ DEV void wgrad_tile(const bf16* __restrict__ X, const bf16* __restrict__ Dz,
         At[row * LDW + smx] = a_nx[j].e[e];
       }
 #pragma unroll
-    for (int e = 0; e < 8; e++) {
-      int row = sv + e;
-      int smx = smlo | (((smhi ^ row) & 3) << 3);
-      Dt[row * LDW + smx] = d_nx.e[e];
-    }
+    for (int j = 0; j < DC; j++)
+#pragma unroll
+      for (int e = 0; e < 8; e++) {
+        int row = j * 64 + sv + e;
+        int smx = smlo | (((smhi ^ row) & 3) << 3);
+        Dt[row * LDW + smx] = d_nx[j].e[e];
+      }
     __syncthreads();
     if (m0 + 32 < mend) {
 #pragma unroll
       for (int j = 0; j < AC; j++)
         a_nx[j] = load8_a<1, VECA>(X, p, mg, m0 + 32 + sm,
                                    k3_0 + j * 64 + sv);
-      int m = m0 + 32 + sm, n = n0 + sv;
-      if (m < p.M && n < Ntot)
-        d_nx.u = *(const uint4*)(Dz + (long)m * Ntot + n);
-      else
-        d_nx.u = uint4{0, 0, 0, 0};
+#pragma unroll
+      for (int j = 0; j < DC; j++) {
+        int m = m0 + 32 + sm, n = n0 + j * 64 + sv;
+        if (m < p.M && n < Ntot)
+          d_nx[j].u = *(const uint4*)(Dz + (long)m * Ntot + n);
+        else
+          d_nx[j].u = uint4{0, 0, 0, 0};
+      }
     }
-    bf16x8 af[MI], bf[2];
+    bf16x8 af[MI], bf[NI];
 #pragma unroll
     for (int mi = 0; mi < MI; mi++) {
       int row = wr * (TK3 / 2) + mi * 16 + fr;
       af[mi] = *(const bf16x8*)&At[row * LDW + ((fk ^ row) & 3) * 8];
     }
 #pragma unroll
-    for (int ni = 0; ni < 2; ni++) {
-      int row = wc * 32 + ni * 16 + fr;
+    for (int ni = 0; ni < NI; ni++) {
+      int row = wc * (TKO / 2) + ni * 16 + fr;
       bf[ni] = *(const bf16x8*)&Dt[row * LDW + ((fk ^ row) & 3) * 8];
     }
 #pragma unroll
     for (int mi = 0; mi < MI; mi++)
 #pragma unroll
-      for (int ni = 0; ni < 2; ni++)
+      for (int ni = 0; ni < NI; ni++)
         acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
     __syncthreads();
   }
-  // D: col (=ko) = fr, row (=k3) = fk*4+q.  Transpose through LDS.
-  __syncthreads();  // LDS union: staging buffers are done
-#pragma unroll
-  for (int mi = 0; mi < MI; mi++)
-#pragma unroll
-    for (int ni = 0; ni < 2; ni++) {
-      int ko_l = wc * 32 + ni * 16 + fr;
-#pragma unroll
-      for (int q = 0; q < 4; q++)
-        smem.out[ko_l][wr * (TK3 / 2) + mi * 16 + fk * 4 + q] =
-            acc[mi][ni][q];
-    }
-  __syncthreads();
+  // D: col (=ko) = fr, row (=k3) = fk*4+q.  Transpose through the 64-ko
+  // LDS image, one pass per 64-ko slab (TKO=128: pass p uses only the
+  // wc==p waves' accumulators; every thread joins the global store).
   float* dst = out + (split ? (long)z * Ntot * (long)p.Kd : 0L);
   constexpr int CCH = TK3 / 4;  // k3 span per thread in the store pass
   const int ko_r = tid >> 2, cch = (tid & 3) * CCH;
-  const int gko = n0 + ko_r;
-  if (gko < Ntot) {
-    long base = (long)gko * p.Kd + k3_0 + cch;
-    if (k3_0 + cch + CCH <= p.Kd) {
 #pragma unroll
-      for (int e = 0; e < CCH; e += 4) {
-        float4 v = *(const float4*)&smem.out[ko_r][cch + e];
-        if (!split) {  // accumulate into .grad (single writer, RMW is safe)
-          float4 d = *(const float4*)&dst[base + e];
-          v.x += d.x; v.y += d.y; v.z += d.z; v.w += d.w;
+  for (int pass = 0; pass < TKO / 64; pass++) {
+    __syncthreads();  // LDS union / previous pass done
+    if (TKO == 64 || wc == pass) {
+#pragma unroll
+      for (int mi = 0; mi < MI; mi++)
+#pragma unroll
+        for (int ni = 0; ni < NI; ni++) {
+          int ko_l = wc * (TKO / 2) + ni * 16 + fr - pass * 64;
+#pragma unroll
+          for (int q = 0; q < 4; q++)
+            smem.out[ko_l][wr * (TK3 / 2) + mi * 16 + fk * 4 + q] =
+                acc[mi][ni][q];
         }
-        *(float4*)&dst[base + e] = v;
+    }
+    __syncthreads();
+    const int gko = n0 + pass * 64 + ko_r;
+    if (gko < Ntot) {
+      long base = (long)gko * p.Kd + k3_0 + cch;
+      if (k3_0 + cch + CCH <= p.Kd) {
+#pragma unroll
+        for (int e = 0; e < CCH; e += 4) {
+          float4 v = *(const float4*)&smem.out[ko_r][cch + e];
+          if (!split) {  // accumulate into .grad (single writer; RMW safe)
+            float4 d = *(const float4*)&dst[base + e];
+            v.x += d.x; v.y += d.y; v.z += d.z; v.w += d.w;
+          }
+          *(float4*)&dst[base + e] = v;
+        }
+      } else {
+        for (int e = 0; e < CCH && k3_0 + cch + e < p.Kd; e++)
+          dst[base + e] =
+              smem.out[ko_r][cch + e] + (split ? 0.f : dst[base + e]);
       }
-    } else {
-      for (int e = 0; e < CCH && k3_0 + cch + e < p.Kd; e++)
-        dst[base + e] = smem.out[ko_r][cch + e] + (split ? 0.f : dst[base + e]);
     }
   }
 }
@@ -492,8 +509,8 @@ __global__ __launch_bounds__(256) void k_wgrad(
     const bf16* __restrict__ X, const bf16* __restrict__ Dz,
     float* __restrict__ dW, ConvP p, MagicP mg, int Ntot, int mchunk) {
   __shared__ WgradSmem smem;
-  wgrad_tile<VECA>(X, Dz, dW, p, mg, Ntot, mchunk, blockIdx.x, blockIdx.y,
-                   blockIdx.z, gridDim.z > 1, smem);
+  wgrad_tile<VECA, 64, 64>(X, Dz, dW, p, mg, Ntot, mchunk, blockIdx.x,
+                           blockIdx.y, blockIdx.z, gridDim.z > 1, smem);
 }
 
 // Σ over msplit wgrad slabs [z][n] -> dW[n] (+=; dW pre-zeroed or fresh).
@@ -547,7 +564,7 @@ struct WgradBatchArgs {
   WgradTask t[WG_MAX_TASKS];
 };
 
-template <int TK3>
+template <int TK3, int TKO>
 __global__ __launch_bounds__(256) void k_wgrad_batched(WgradBatchArgs a) {
   int bid = blockIdx.x;
   int i = 0;
@@ -557,13 +574,13 @@ __global__ __launch_bounds__(256) void k_wgrad_batched(WgradBatchArgs a) {
   int nt = t.tx * t.ty;
   int z = local / nt, rem = local - z * nt;
   int ty = rem / t.tx, tx = rem - ty * t.tx;
-  __shared__ WgradSmemT<TK3> smem;
+  __shared__ WgradSmemT<TK3, TKO> smem;
   if (t.vec)
-    wgrad_tile<true, TK3>(t.X, t.Dz, t.out, t.p, t.mg, t.Ntot, t.mchunk,
-                          tx, ty, z, t.msplit > 1, smem);
+    wgrad_tile<true, TK3, TKO>(t.X, t.Dz, t.out, t.p, t.mg, t.Ntot,
+                               t.mchunk, tx, ty, z, t.msplit > 1, smem);
   else
-    wgrad_tile<false, TK3>(t.X, t.Dz, t.out, t.p, t.mg, t.Ntot, t.mchunk,
-                           tx, ty, z, t.msplit > 1, smem);
+    wgrad_tile<false, TK3, TKO>(t.X, t.Dz, t.out, t.p, t.mg, t.Ntot,
+                                t.mchunk, tx, ty, z, t.msplit > 1, smem);
 }
 
 struct WredTask {
@@ -961,20 +978,21 @@ extern "C" void launch_gemm_bf16(const void* a, const void* b, void* c, int M,
                                nullptr, nullptr, p, mg, N, 0, 0);
 }
 
-extern "C" void launch_wgrad_batched(const void* args, int blocks,
-                                     hipStream_t st) {
-  if (blocks > 0)
-    k_wgrad_batched<64><<<blocks, 256, 0, st>>>(
-        *(const WgradBatchArgs*)args);
-}
-
-// 128-wide k3 tile class (tasks with Kd >= 512 and M >= 8192): tx counted
-// in 128-wide tiles by the caller.
-extern "C" void launch_wgrad_batched_t128(const void* args, int blocks,
-                                          hipStream_t st) {
-  if (blocks > 0)
-    k_wgrad_batched<128><<<blocks, 256, 0, st>>>(
-        *(const WgradBatchArgs*)args);
+// tile class per task batch: tx/ty counted in (tk3, tko)-wide tiles by
+// the caller.  128-wide k3 halves Dz re-reads; 128-wide ko halves X
+// re-reads; both double the MFMA per staging write.
+extern "C" void launch_wgrad_batched_t(const void* args, int blocks,
+                                       int tk3, int tko, hipStream_t st) {
+  if (blocks <= 0) return;
+  const WgradBatchArgs& a = *(const WgradBatchArgs*)args;
+  if (tk3 == 128 && tko == 128)
+    k_wgrad_batched<128, 128><<<blocks, 256, 0, st>>>(a);
+  else if (tk3 == 128)
+    k_wgrad_batched<128, 64><<<blocks, 256, 0, st>>>(a);
+  else if (tko == 128)
+    k_wgrad_batched<64, 128><<<blocks, 256, 0, st>>>(a);
+  else
+    k_wgrad_batched<64, 64><<<blocks, 256, 0, st>>>(a);
 }
 
 extern "C" void launch_wgrad_reduce_batched(const void* args, int blocks,

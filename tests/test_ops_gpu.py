@@ -280,8 +280,9 @@ def test_wgrad_batched_t128_matches_standalone():
     """Deferred/batched wgrad routes Kd>=512 & M>=8192 tasks through the
     128-wide k3 tile — must match the standalone 64-wide kernel."""
     torch.manual_seed(4)
-    shapes = [(16, 64, 32, 64, 3, 1),    # M=16384, Kd=576 -> t128
-              (16, 128, 16, 128, 3, 1),  # M=4096 -> stays t64
+    shapes = [(16, 64, 32, 64, 3, 1),    # M=16384, Kd=576 -> tk3=128
+              (16, 128, 16, 128, 3, 1),  # M=4096, K=128 -> tko=128 only
+              (16, 64, 32, 128, 3, 1),   # M=16384, K=128 -> 128x128
               (40, 72, 24, 128, 3, 1)]   # M=23040, Kd=648 (non-vec-ish C)
     for bs, cin, hw, cout, k, s in shapes:
         x = torch.randn(bs, cin, hw, hw, device="cuda") \
