@@ -274,11 +274,11 @@ void flush_one_group(int group_lo, int group_hi) {
   auto tk3_of = [](const ConvP& p) {
     static int kd_min = [] {
       const char* e = getenv("HZ_WG_TK3_KD");
-      return e ? atoi(e) : 512;
+      return e ? atoi(e) : 256;  // swept on MI355X (r50 224 + bs1024)
     }();
     static int m_min = [] {
       const char* e = getenv("HZ_WG_TK3_M");
-      return e ? atoi(e) : 8192;
+      return e ? atoi(e) : 2048;  // swept on MI355X
     }();
     return (p.Kd >= kd_min && p.M >= m_min) ? 128 : 64;
   };
