@@ -862,7 +862,14 @@ static inline int pick_tile(int M, int N) {
     const char* e = getenv("HZ_TILE_M_MIN");
     return e ? atoi(e) : 16384;
   }();
+  static int m_min256 = [] {
+    const char* e = getenv("HZ_TILE_M256");
+    return e ? atoi(e) : 65536;
+  }();
   if (M < m_min) return 0;
+  if (M >= m_min256 && N < 128
+      && (long)cdiv_h(M, 256) * cdiv_h(N, 64) >= fill)
+    return 3;  // 256x64: very-large-M thin-N shapes (stem/layer1 @224)
   if (N >= 128 && (long)cdiv_h(M, 128) * cdiv_h(N, 128) >= fill) return 2;
   if ((long)cdiv_h(M, 128) * cdiv_h(N, 64) >= fill) return 1;
   return 0;
@@ -894,7 +901,8 @@ extern "C" void launch_conv_fwd(const void* x, const void* w, void* y,
     else if (s) CASE_T(false, false, true, TM, TN);                    \
     else CASE_T(false, false, false, TM, TN);                          \
   } while (0)
-  if (tile == 2) DISPATCH(128, 128);
+  if (tile == 3) DISPATCH(256, 64);
+  else if (tile == 2) DISPATCH(128, 128);
   else if (tile == 1) DISPATCH(128, 64);
   else DISPATCH(64, 64);
 #undef DISPATCH
@@ -952,7 +960,8 @@ extern "C" void launch_conv_dgrad(const void* dz, const void* w_rsck,
     if (vec) DGRAD_T(true, true, TM, TN);                               \
     else DGRAD_T(false, false, TM, TN);                                 \
   } while (0)
-    if (tile == 2) DISPATCH(128, 128);
+    if (tile == 3) DISPATCH(256, 64);
+    else if (tile == 2) DISPATCH(128, 128);
     else if (tile == 1) DISPATCH(128, 64);
     else DISPATCH(64, 64);
 #undef DISPATCH
