@@ -1445,6 +1445,17 @@ static long bn_v8_m_min() {
   return v;
 }
 
+// m-iterations per thread target for the v8 kernels (controls msplit =
+// blocks): more iterations -> fewer blocks -> less per-block epilogue +
+// atomic volume, at the cost of m-parallelism.
+static int bn_v8_iters() {
+  static int v = [] {
+    const char* e = getenv("HZ_BN_V8_ITERS");
+    return e ? atoi(e) : 1;
+  }();
+  return v;
+}
+
 void launch_cast_bnact(const float* src, void* dst, long M, int C,
                        int nsplit, int accum, const void* x_up,
                        const void* y_up, const float* smean,
@@ -1456,7 +1467,8 @@ void launch_cast_bnact(const float* src, void* dst, long M, int C,
     // sized so every thread has >=1 row and the grid reaches ~768 blocks
     // on big-M shapes (matching the scalar kernel's fill).
     int lpr = C >> 3, mstep = 256 / lpr;
-    int msplit = (int)min((long)768, max((long)1, (M + mstep - 1) / mstep));
+    int msplit = (int)min((long)768, max((long)1, (M + (long)mstep *
+        bn_v8_iters() - 1) / ((long)mstep * bn_v8_iters())));
     if (g_det_kernels) msplit = 1;
     long mchunk = (M + msplit - 1) / msplit;
     msplit = (int)((M + mchunk - 1) / mchunk);
@@ -1494,7 +1506,8 @@ void launch_bnact_bwd_reduce(const void* dy, const void* yout, const void* x,
   // sum_dz/sum_dzx must be pre-zeroed (they are accumulated atomically)
   if ((C & 7) == 0 && C <= 2048 && bn_v8_enabled() && M >= bn_v8_m_min()) {
     int lpr = C >> 3, mstep = 256 / lpr;
-    int msplit = (int)min((long)768, max((long)1, (M + mstep - 1) / mstep));
+    int msplit = (int)min((long)768, max((long)1, (M + (long)mstep *
+        bn_v8_iters() - 1) / ((long)mstep * bn_v8_iters())));
     if (g_det_kernels) msplit = 1;
     long mchunk = (M + msplit - 1) / msplit;
     msplit = (int)((M + mchunk - 1) / mchunk);
