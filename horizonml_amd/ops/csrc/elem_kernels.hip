@@ -1451,9 +1451,19 @@ static long bn_v8_m_min() {
 static int bn_v8_iters() {
   static int v = [] {
     const char* e = getenv("HZ_BN_V8_ITERS");
-    return e ? atoi(e) : 1;
+    return e ? atoi(e) : 8;  // swept: big-M flat, mid-M 1.5-2.5x faster
   }();
   return v;
+}
+
+// dispatch rule (isolated sweeps, HZ_BN_V8_ITERS grid): v8 wins at
+// M >= 65k for any C, and at mid M (>= ~6k) for C <= 512; the scalar
+// per-channel-column kernel keeps big-C small-M shapes (little
+// m-parallelism per v8 block: lpr eats the 256 threads).
+static bool bn_v8_pick(long M, int C) {
+  if ((C & 7) != 0 || C > 2048 || !bn_v8_enabled()) return false;
+  if (M >= bn_v8_m_min()) return true;
+  return C <= 512 && M >= 6000;
 }
 
 void launch_cast_bnact(const float* src, void* dst, long M, int C,
@@ -1462,7 +1472,7 @@ void launch_cast_bnact(const float* src, void* dst, long M, int C,
                        const float* sinvstd, const float* gamma,
                        const float* beta, float* sum_dz, float* sum_dzx,
                        int mask_mode, hipStream_t st) {
-  if ((C & 7) == 0 && C <= 2048 && bn_v8_enabled() && M >= bn_v8_m_min()) {
+  if (bn_v8_pick(M, C)) {
     // vectorized: whole C per block, m split across blockIdx.y.  msplit
     // sized so every thread has >=1 row and the grid reaches ~768 blocks
     // on big-M shapes (matching the scalar kernel's fill).
@@ -1504,7 +1514,7 @@ void launch_bnact_bwd_reduce(const void* dy, const void* yout, const void* x,
                              float* sum_dz, float* sum_dzx, long M, int C,
                              int mask_mode, hipStream_t st) {
   // sum_dz/sum_dzx must be pre-zeroed (they are accumulated atomically)
-  if ((C & 7) == 0 && C <= 2048 && bn_v8_enabled() && M >= bn_v8_m_min()) {
+  if (bn_v8_pick(M, C)) {
     int lpr = C >> 3, mstep = 256 / lpr;
     int msplit = (int)min((long)768, max((long)1, (M + (long)mstep *
         bn_v8_iters() - 1) / ((long)mstep * bn_v8_iters())));
