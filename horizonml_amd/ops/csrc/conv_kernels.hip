@@ -850,6 +850,35 @@ static void launch_stem_conv(const bf16* x, const bf16* w, bf16* y,
 // grids stay on the latency-optimized 64×64 tile; ImageNet-shaped convs
 // (ResNet50@224 etc.) move to 128×64 / 128×128 where each wave issues
 // 2–4× the MFMAs per LDS fragment read.
+static inline int pick_tile_env(int M, int N, int fill, int m_min,
+                                int m_min256) {
+  if (M < m_min) return 0;
+  if (M >= m_min256 && N < 128
+      && (long)cdiv_h(M, 256) * cdiv_h(N, 64) >= fill)
+    return 3;
+  if (N >= 128 && (long)cdiv_h(M, 128) * cdiv_h(N, 128) >= fill) return 2;
+  if ((long)cdiv_h(M, 128) * cdiv_h(N, 64) >= fill) return 1;
+  return 0;
+}
+
+// dgrad has its own thresholds: no stats epilogue and an RSCK B-gather,
+// so the fwd-measured floors need not match (HZ_TILE_*_DG)
+static inline int pick_tile_dgrad(int M, int N) {
+  static int fill = [] {
+    const char* e = getenv("HZ_TILE_FILL_DG");
+    return e ? atoi(e) : 192;
+  }();
+  static int m_min = [] {
+    const char* e = getenv("HZ_TILE_M_MIN_DG");
+    return e ? atoi(e) : 16384;
+  }();
+  static int m256 = [] {
+    const char* e = getenv("HZ_TILE_M256_DG");
+    return e ? atoi(e) : 65536;
+  }();
+  return pick_tile_env(M, N, fill, m_min, m256);
+}
+
 static inline int pick_tile(int M, int N) {
   static int fill = [] {
     const char* e = getenv("HZ_TILE_FILL");
@@ -950,7 +979,7 @@ extern "C" void launch_conv_dgrad(const void* dz, const void* w_rsck,
                                  make_magic(p, 2), p.C, kchunk, 0);
   } else {
     MagicP mg = make_magic(p, 2);
-    int tile = pick_tile(p.M, p.C);
+    int tile = pick_tile_dgrad(p.M, p.C);
 #define DGRAD_T(VA, VB, TM, TN)                                         \
   k_conv_mfma<2, VA, VB, false, false, TM, TN><<<grid, 256, 0, st>>>(   \
       A, B, (bf16*)dx, nullptr, nullptr, p, mg, p.C, 0, accum)
