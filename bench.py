@@ -36,6 +36,8 @@ def parse_args():
     ap.add_argument("--warmup", type=int, default=20)
     ap.add_argument("--batch-size", type=int, default=64,
                     help="per-GPU batch (reference parity: 64)")
+    ap.add_argument("--image-size", type=int, default=32,
+                    help="input H=W (224 for the ImageNet-shaped configs)")
     ap.add_argument("--model", type=str, default="resnet18")
     ap.add_argument("--optimizer", type=str, default="adam")
     ap.add_argument("--no-graph", action="store_true",
@@ -98,7 +100,8 @@ def main():
     bs = args.batch_size
     torch.manual_seed(1234 + rank)  # different data per rank (DP semantics)
     pool_n = 8
-    pool_x = [torch.randn(bs, 3, 32, 32, device=dev)
+    pool_x = [torch.randn(bs, 3, args.image_size, args.image_size,
+                          device=dev)
               .to(memory_format=torch.channels_last).to(torch.bfloat16)
               for _ in range(pool_n)]
     pool_y = [torch.randint(0, 10, (bs,), device=dev) for _ in range(pool_n)]
@@ -230,6 +233,7 @@ def main():
             world=world, batch_size=bs, model=args.model,
             optimizer=args.optimizer, exec_mode=mode,
             final_loss=final_loss, infer=args.infer,
+            image_size=args.image_size,
             comm_mode=(f"bucketed{n_buckets}" if sched is not None
                        else ("flat1" if use_comm else "none")))
         print(emit_record(out), flush=True)

@@ -56,7 +56,7 @@ def max_elapsed_over_ranks(elapsed_s: float, world: int,
 def build_record(*, elapsed_s: float, steps: int, warmup: int, world: int,
                  batch_size: int, model: str, optimizer: str, exec_mode: str,
                  final_loss: float, infer: bool = False,
-                 comm_mode: str = "none") -> dict:
+                 comm_mode: str = "none", image_size: int = 32) -> dict:
     """Assemble the driver-contract JSON record (one line, rank 0 only).
 
     ``value`` is the WHOLE-JOB aggregate images/sec over all ranks (weak
@@ -82,10 +82,12 @@ def build_record(*, elapsed_s: float, steps: int, warmup: int, world: int,
         "dtype": "bf16",
         "data": "synthetic",
         "config": {
-            "model": model + ("_infer" if infer else "_cifar10"),
+            "model": model + ("_infer" if infer
+                              else ("_cifar10" if image_size == 32
+                                    else f"_synthetic{image_size}")),
             "global_batch": global_batch,
             "seq_len": None,
-            "image": "3x32x32",
+            "image": f"3x{image_size}x{image_size}",
             "parallelism": f"dp{world}",
             "optimizer": optimizer,
             "exec": exec_mode,
