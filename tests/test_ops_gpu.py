@@ -81,6 +81,7 @@ def test_conv_bn_relu_fwd(cfg):
     (64, 64, 3, 1, 56),    # M=50k -> 128x64 throughput tile
     (64, 128, 3, 1, 56),   # -> 128x128 tile
     (64, 130, 3, 1, 56),   # N tail on the 128x128 tile
+    (64, 64, 3, 1, 72),    # M=83k -> 256x64 tile (fwd + dgrad)
 ])
 def test_conv_throughput_tiles_fwd_bwd(cfg):
     """ImageNet-shaped convs route to the 128-wide throughput tiles
