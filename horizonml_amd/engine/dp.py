@@ -131,9 +131,9 @@ def train_dp_flat(ctx: DistContext, epochs: int, sample_size: int,
     sumsq = torch.zeros(1, device=dev) if probe_divergence else None
     div_acc = torch.zeros(1, device=dev) if probe_divergence else None
 
-    # Adam: divergence probe fused into the optimizer kernel (it already
-    # streams the f32 grad); SGD keeps the standalone gdiv kernel.
-    fused_probe = probe_divergence and isinstance(opt, HorizonAdam)
+    # divergence probe fused into the optimizer kernel (Adam and SGD both
+    # stream the f32 grad anyway — the standalone pass cost 61 us/step)
+    fused_probe = probe_divergence
     probe_bufs = (prev, sumsq, div_acc) if fused_probe else None
 
     def fb_step(x_u8, y):

@@ -150,13 +150,14 @@ class HorizonSGD:
 
     def step(self, zero_grad: bool = True, grad_bf16=None,
              grad_scale: float = 1.0, probe=None):
-        if probe is not None:
-            raise ValueError("fused divergence probe is Adam-only; use the "
-                             "standalone grad_divergence kernel with SGD")
+        """``probe``: optional ``(prev, sumsq, out)`` — the divergence
+        probe fused into the SGD pass, like HorizonAdam's."""
         _ops.extension().flush_wgrad()  # batched deferred weight grads
+        p0, p1, p2 = probe if probe is not None else (None, None, None)
         _ops.extension().sgd_step(self.mgr.master, self.mgr.grad, self.mom,
                                   self.mgr.shadow, self.lr, self.mu, self.wd,
-                                  zero_grad, grad_bf16, grad_scale)
+                                  zero_grad, grad_bf16, grad_scale,
+                                  p0, p1, p2)
         self.mgr.stats_arena.zero_()
         self.mgr.refresh_rsck()
 

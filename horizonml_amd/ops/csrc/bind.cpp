@@ -76,7 +76,8 @@ void launch_adam_step(float*, float*, float*, float*, void*, const float*,
                       long, const void*, float, float*, float*, float*,
                       hipStream_t);
 void launch_sgd_step(float*, float*, float*, void*, long, float, float, float,
-                     int, const void*, float, hipStream_t);
+                     int, const void*, float, float*, float*, float*,
+                     hipStream_t);
 void launch_permute_krsc_rsck(const void*, void*, const int*, int, int,
                               hipStream_t);
 void launch_grad_divergence(const float*, float*, float*, float*, long, int,
@@ -877,15 +878,27 @@ void adam_step(Tensor master, Tensor grad, Tensor m, Tensor v,
 void sgd_step(Tensor master, Tensor grad, c10::optional<Tensor> mom,
               c10::optional<Tensor> shadow, double lr, double mu, double wd,
               bool zero_grad, c10::optional<Tensor> grad_bf16,
-              double grad_scale) {
+              double grad_scale, c10::optional<Tensor> probe_prev,
+              c10::optional<Tensor> probe_sumsq,
+              c10::optional<Tensor> probe_out) {
   check_f32(master, "master");
+  TORCH_CHECK(probe_prev.has_value() == probe_sumsq.has_value()
+                  && probe_prev.has_value() == probe_out.has_value(),
+              "probe tensors must be passed together");
   launch_sgd_step(master.data_ptr<float>(), grad.data_ptr<float>(),
                   mom.has_value() ? mom->data_ptr<float>() : nullptr,
                   shadow.has_value() ? shadow->data_ptr() : nullptr,
                   master.numel(), (float)lr, (float)mu, (float)wd,
                   zero_grad ? 1 : 0,
                   grad_bf16.has_value() ? grad_bf16->data_ptr() : nullptr,
-                  (float)grad_scale, cur_stream());
+                  (float)grad_scale,
+                  probe_prev.has_value() ? probe_prev->data_ptr<float>()
+                                         : nullptr,
+                  probe_sumsq.has_value() ? probe_sumsq->data_ptr<float>()
+                                          : nullptr,
+                  probe_out.has_value() ? probe_out->data_ptr<float>()
+                                        : nullptr,
+                  cur_stream());
 }
 
 void permute_krsc_rsck(Tensor src, Tensor dst, Tensor meta,

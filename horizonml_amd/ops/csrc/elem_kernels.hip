@@ -1237,10 +1237,17 @@ __global__ __launch_bounds__(256) void k_sgd_step(
     float* __restrict__ master, float* __restrict__ grad,
     float* __restrict__ mom, bf16* __restrict__ shadow, long n, float lr,
     float mu, float wd, int zero_grad, const bf16* __restrict__ gsrc,
-    float gscale) {
+    float gscale, float* __restrict__ prev, float* __restrict__ sumsq) {
+  float acc = 0.f;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (long)gridDim.x * blockDim.x) {
-    float g = gsrc != nullptr ? b2f(gsrc[i]) * gscale : grad[i];
+    float gl = grad[i];
+    float g = gsrc != nullptr ? b2f(gsrc[i]) * gscale : gl;
+    if (prev != nullptr) {  // fused divergence probe (see k_adam_step)
+      float d = gl - prev[i];
+      prev[i] = gl;
+      acc += d * d;
+    }
     float w = master[i];
     if (wd != 0.f) g += wd * w;
     float u = (mom != nullptr) ? (mu * mom[i] + g) : g;
@@ -1249,6 +1256,14 @@ __global__ __launch_bounds__(256) void k_sgd_step(
     master[i] = w;
     if (shadow != nullptr) shadow[i] = f2b(w);
     if (zero_grad) grad[i] = 0.f;
+  }
+  if (prev != nullptr) {
+    acc = wave_reduce_sum(acc);
+    __shared__ float ws[4];
+    if ((threadIdx.x & 63) == 0) ws[threadIdx.x >> 6] = acc;
+    __syncthreads();
+    if (threadIdx.x == 0)
+      atomicAdd(sumsq, ws[0] + ws[1] + ws[2] + ws[3]);
   }
 }
 
@@ -1691,10 +1706,13 @@ void launch_adam_step(float* master, float* grad, float* m, float* v,
 
 void launch_sgd_step(float* master, float* grad, float* mom, void* shadow,
                      long n, float lr, float mu, float wd, int zero_grad,
-                     const void* gsrc, float gscale, hipStream_t st) {
+                     const void* gsrc, float gscale, float* prev,
+                     float* sumsq, float* divout, hipStream_t st) {
   k_sgd_step<<<gsz(n), 256, 0, st>>>(master, grad, mom, (bf16*)shadow, n, lr,
                                      mu, wd, zero_grad, (const bf16*)gsrc,
-                                     gscale);
+                                     gscale, prev, sumsq);
+  if (prev != nullptr)
+    k_gdiv_finalize<<<1, 1, 0, st>>>(sumsq, divout, 0);
 }
 
 void launch_permute_krsc_rsck(const void* src, void* dst, const int* meta,

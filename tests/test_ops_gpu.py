@@ -261,7 +261,8 @@ def test_fused_sgd_matches_torch():
     mom = torch.zeros(n, device="cuda")
     for _ in range(3):
         opt.step()
-        _C().sgd_step(mg, gg, mom, None, 0.1, 0.9, 0.0, False, None, 1.0)
+        _C().sgd_step(mg, gg, mom, None, 0.1, 0.9, 0.0, False, None, 1.0,
+                      None, None, None)
     assert rel(mg, p.detach()) < 1e-5
 
 
