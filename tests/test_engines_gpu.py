@@ -129,6 +129,23 @@ def test_tp_engine_gpu(tmp_path):
 
 
 @pytest.mark.timeout(300)
+def test_pp_checkpoint_flat_eager_gpu(tmp_path):
+    """PP checkpoint on GPU round-trips the flat-eager fused optimizer's
+    state (HorizonAdam m/v/step) through the per-rank files."""
+    import pandas as pd
+
+    from layer_model_parallel_train import run_model_parallel
+    ckpt = str(tmp_path / "pp.ckpt")
+    run_model_parallel(1, 1, 128, str(tmp_path / "l1"), batch_size=32,
+                       synthetic=True, checkpoint_path=ckpt)
+    run_model_parallel(1, 3, 128, str(tmp_path / "l2"), batch_size=32,
+                       synthetic=True, checkpoint_path=ckpt)
+    df = pd.read_csv(f"{tmp_path}/l2/worker_0_samples_128.csv")
+    assert list(df["epoch"]) == [2, 3]
+    assert df["loss"].iloc[-1] < df["loss"].iloc[0]
+
+
+@pytest.mark.timeout(300)
 def test_pp_engine_microbatches_gpu(tmp_path):
     """Pipeline with microbatching (bubble reduction) through the fused
     blocks — exercises gradient accumulation across microbatch backwards
