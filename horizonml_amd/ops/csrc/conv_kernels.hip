@@ -663,9 +663,13 @@ extern "C" int conv_dgrad_splitk(ConvP p) {
     const char* e = getenv("HZ_SK_KD_DG");
     return e ? atoi(e) : 512;
   }();
+  static int tile_min_dg = [] {
+    const char* e = getenv("HZ_SK_TILES_DG");
+    return e ? atoi(e) : 96;
+  }();
   int tiles = cdiv_h(p.C, 64) * cdiv_h(p.Nb * p.H * p.W, 64);
   int Kd = p.R * p.S * p.K;
-  if (tiles >= 96 || Kd < kd_min_dg) return 1;
+  if (tiles >= tile_min_dg || Kd < kd_min_dg) return 1;
   int sk = cdiv_h(256, tiles);
   int maxsk = cdiv_h(Kd, 32);
   if (sk > maxsk) sk = maxsk;
