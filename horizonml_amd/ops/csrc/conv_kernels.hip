@@ -665,7 +665,7 @@ extern "C" int conv_dgrad_splitk(ConvP p) {
   }();
   static int tile_min_dg = [] {
     const char* e = getenv("HZ_SK_TILES_DG");
-    return e ? atoi(e) : 96;
+    return e ? atoi(e) : 256;  // swept: r50@224 9.05 -> 8.89 ms; CIFAR flat
   }();
   int tiles = cdiv_h(p.C, 64) * cdiv_h(p.Nb * p.H * p.W, 64);
   int Kd = p.R * p.S * p.K;
