@@ -311,22 +311,28 @@ __global__ __launch_bounds__(256) void k_cast_bnact(
 // channels, every load is a 16-byte dwordx4, and the whole C extent fits
 // one block (C ≤ 2048), partial sums LDS-reduced across the block's
 // m-lanes before one atomicAdd per channel.
+// cslab: channels handled per block (blockIdx.x indexes slabs of the C
+// extent).  cslab == C reproduces the original single-slab kernel; for
+// C > 512 the launcher slices C into <=512-channel slabs so each block
+// keeps >=4 m-rows in flight (full-C blocks at C=2048 had mstep=1 and
+// lost to the scalar kernel at small M — dispatch-rule comment below).
 template <int MASK>
 __global__ __launch_bounds__(256) void k_bnact_bwd_reduce_v8(
     const bf16* __restrict__ dy, const bf16* __restrict__ yout,
     const bf16* __restrict__ x, const float* __restrict__ save_mean,
     const float* __restrict__ save_invstd, const float* __restrict__ gamma,
     const float* __restrict__ beta, float* __restrict__ sum_dz,
-    float* __restrict__ sum_dzx, long M, int C, long mchunk) {
+    float* __restrict__ sum_dzx, long M, int C, long mchunk, int cslab) {
   // MASK compile-time: runtime mask branches in this 8-wide unrolled body
   // compiled to ~630 instructions per iteration (measured 45 GB/s)
   __shared__ float sdz[256][8];
   __shared__ float sdzx[256][8];
-  const int lpr = C >> 3;            // lanes per m-row
+  const int cbeg = blockIdx.x * cslab;
+  const int lpr = cslab >> 3;        // lanes per m-row
   const int mstep = 256 / lpr;       // m rows in flight per block
   const int active = mstep * lpr;
   const int tid = threadIdx.x;
-  const int th_c = (tid % lpr) * 8;
+  const int th_c = cbeg + (tid % lpr) * 8;
   const int th_m = tid / lpr;
   const long mbeg = (long)blockIdx.y * mchunk;
   const long mend = min(M, mbeg + mchunk);
@@ -377,14 +383,14 @@ __global__ __launch_bounds__(256) void k_bnact_bwd_reduce_v8(
   // serialized ~0.2 us of LDS latency + 16 atomic issues per block)
   const float* S1 = &sdz[0][0];
   const float* S2 = &sdzx[0][0];
-  for (int c = tid; c < C; c += 256) {
+  for (int c = tid; c < cslab; c += 256) {
     float r1 = 0.f, r2 = 0.f;
     for (int j = 0; j < mstep; j++) {
-      r1 += S1[j * C + c];
-      r2 += S2[j * C + c];
+      r1 += S1[j * cslab + c];
+      r2 += S2[j * cslab + c];
     }
-    atomicAdd(&sum_dz[c], r1);
-    atomicAdd(&sum_dzx[c], r2);
+    atomicAdd(&sum_dz[cbeg + c], r1);
+    atomicAdd(&sum_dzx[cbeg + c], r2);
   }
 }
 
@@ -397,15 +403,16 @@ __global__ __launch_bounds__(256) void k_cast_bnact_v8(
     const bf16* __restrict__ y_up, const float* __restrict__ save_mean,
     const float* __restrict__ save_invstd, const float* __restrict__ gamma,
     const float* __restrict__ beta, float* __restrict__ sum_dz,
-    float* __restrict__ sum_dzx, long mchunk) {
+    float* __restrict__ sum_dzx, long mchunk, int cslab) {
   __shared__ float sdz[256][8];
   __shared__ float sdzx[256][8];
   const long n = M * (long)C;
-  const int lpr = C >> 3;
+  const int cbeg = blockIdx.x * cslab;
+  const int lpr = cslab >> 3;
   const int mstep = 256 / lpr;
   const int active = mstep * lpr;
   const int tid = threadIdx.x;
-  const int th_c = (tid % lpr) * 8;
+  const int th_c = cbeg + (tid % lpr) * 8;
   const int th_m = tid / lpr;
   const long mbeg = (long)blockIdx.y * mchunk;
   const long mend = min(M, mbeg + mchunk);
@@ -474,14 +481,14 @@ __global__ __launch_bounds__(256) void k_cast_bnact_v8(
   // serialized ~0.2 us of LDS latency + 16 atomic issues per block)
   const float* S1 = &sdz[0][0];
   const float* S2 = &sdzx[0][0];
-  for (int c = tid; c < C; c += 256) {
+  for (int c = tid; c < cslab; c += 256) {
     float r1 = 0.f, r2 = 0.f;
     for (int j = 0; j < mstep; j++) {
-      r1 += S1[j * C + c];
-      r2 += S2[j * C + c];
+      r1 += S1[j * cslab + c];
+      r2 += S2[j * cslab + c];
     }
-    atomicAdd(&sum_dz[c], r1);
-    atomicAdd(&sum_dzx[c], r2);
+    atomicAdd(&sum_dz[cbeg + c], r1);
+    atomicAdd(&sum_dzx[cbeg + c], r2);
   }
 }
 
@@ -1471,14 +1478,55 @@ static int bn_v8_iters() {
   return v;
 }
 
+// mid-M floor for the C <= 512 full-slab rule (swept: scalar kernel wins
+// below ~6k rows there).
+static long bn_v8_mid_m() {
+  static long v = [] {
+    const char* e = getenv("HZ_BN_V8_MIDM");
+    return e ? atol(e) : 6000L;
+  }();
+  return v;
+}
+
+// channel-slab admission: C > 512 shapes run v8 with blockIdx.x slicing C
+// into <=512-channel slabs (keeps >=4 m-rows per block).  Floor swept on
+// the r50@224 leftovers (M=1568..6272, C=1024/2048).
+static int bn_v8_slab_enabled() {
+  static int v = [] {
+    const char* e = getenv("HZ_BN_V8_SLAB");
+    return e ? atoi(e) : 1;
+  }();
+  return v;
+}
+
+static long bn_v8_slab_m_min() {
+  static long v = [] {
+    const char* e = getenv("HZ_BN_V8_SLAB_MMIN");
+    return e ? atol(e) : 1024L;
+  }();
+  return v;
+}
+
+// largest slab width <= 512 that divides C and keeps 8-channel lanes;
+// C itself if no such divisor (then only the full-C rules admit v8).
+static int bn_v8_cslab(int C) {
+  if (C <= 512) return C;
+  for (int ns = (C + 511) / 512; ns <= 64; ns++)
+    if (C % ns == 0 && ((C / ns) & 7) == 0) return C / ns;
+  return C;
+}
+
 // dispatch rule (isolated sweeps, HZ_BN_V8_ITERS grid): v8 wins at
-// M >= 65k for any C, and at mid M (>= ~6k) for C <= 512; the scalar
-// per-channel-column kernel keeps big-C small-M shapes (little
-// m-parallelism per v8 block: lpr eats the 256 threads).
+// M >= 65k for any C, at mid M (>= ~6k) for C <= 512, and — via the
+// channel-slab grid — down to M >= ~1k for C > 512 (full-C blocks there
+// had mstep=1, the slab form keeps the scalar kernel's m-parallelism with
+// dwordx4 loads).
 static bool bn_v8_pick(long M, int C) {
   if ((C & 7) != 0 || C > 2048 || !bn_v8_enabled()) return false;
   if (M >= bn_v8_m_min()) return true;
-  return C <= 512 && M >= 6000;
+  if (C <= 512) return M >= bn_v8_mid_m();
+  return bn_v8_slab_enabled() && M >= bn_v8_slab_m_min() &&
+         bn_v8_cslab(C) < C;
 }
 
 void launch_cast_bnact(const float* src, void* dst, long M, int C,
@@ -1488,19 +1536,23 @@ void launch_cast_bnact(const float* src, void* dst, long M, int C,
                        const float* beta, float* sum_dz, float* sum_dzx,
                        int mask_mode, hipStream_t st) {
   if (bn_v8_pick(M, C)) {
-    // vectorized: whole C per block, m split across blockIdx.y.  msplit
-    // sized so every thread has >=1 row and the grid reaches ~768 blocks
-    // on big-M shapes (matching the scalar kernel's fill).
-    int lpr = C >> 3, mstep = 256 / lpr;
-    int msplit = (int)min((long)768, max((long)1, (M + (long)mstep *
-        bn_v8_iters() - 1) / ((long)mstep * bn_v8_iters())));
+    // vectorized: one <=512-channel slab per block (blockIdx.x), m split
+    // across blockIdx.y.  msplit sized so every thread has >=1 row and
+    // the grid reaches ~768 blocks on big-M shapes (matching the scalar
+    // kernel's fill).
+    int cslab = bn_v8_cslab(C), nslab = C / cslab;
+    int lpr = cslab >> 3, mstep = 256 / lpr;
+    int msplit = (int)min((long)max(1, 768 / nslab), max((long)1,
+        (M + (long)mstep * bn_v8_iters() - 1) /
+        ((long)mstep * bn_v8_iters())));
     if (g_det_kernels) msplit = 1;
     long mchunk = (M + msplit - 1) / msplit;
     msplit = (int)((M + mchunk - 1) / mchunk);
-    dim3 grid(1, msplit);
+    dim3 grid(nslab, msplit);
 #define LC(MK) k_cast_bnact_v8<MK><<<grid, 256, 0, st>>>( \
     src, (bf16*)dst, M, C, nsplit, accum, (const bf16*)x_up, \
-    (const bf16*)y_up, smean, sinvstd, gamma, beta, sum_dz, sum_dzx, mchunk)
+    (const bf16*)y_up, smean, sinvstd, gamma, beta, sum_dz, sum_dzx, \
+    mchunk, cslab)
     switch (mask_mode) {
       case 1: LC(1); break;
       case 2: LC(2); break;
@@ -1530,16 +1582,18 @@ void launch_bnact_bwd_reduce(const void* dy, const void* yout, const void* x,
                              int mask_mode, hipStream_t st) {
   // sum_dz/sum_dzx must be pre-zeroed (they are accumulated atomically)
   if (bn_v8_pick(M, C)) {
-    int lpr = C >> 3, mstep = 256 / lpr;
-    int msplit = (int)min((long)768, max((long)1, (M + (long)mstep *
-        bn_v8_iters() - 1) / ((long)mstep * bn_v8_iters())));
+    int cslab = bn_v8_cslab(C), nslab = C / cslab;
+    int lpr = cslab >> 3, mstep = 256 / lpr;
+    int msplit = (int)min((long)max(1, 768 / nslab), max((long)1,
+        (M + (long)mstep * bn_v8_iters() - 1) /
+        ((long)mstep * bn_v8_iters())));
     if (g_det_kernels) msplit = 1;
     long mchunk = (M + msplit - 1) / msplit;
     msplit = (int)((M + mchunk - 1) / mchunk);
-    dim3 grid(1, msplit);
+    dim3 grid(nslab, msplit);
 #define LB(MK) k_bnact_bwd_reduce_v8<MK><<<grid, 256, 0, st>>>( \
     (const bf16*)dy, (const bf16*)yout, (const bf16*)x, smean, sinvstd, \
-    gamma, beta, sum_dz, sum_dzx, M, C, mchunk)
+    gamma, beta, sum_dz, sum_dzx, M, C, mchunk, cslab)
     switch (mask_mode) {
       case 1: LB(1); break;
       case 2: LB(2); break;

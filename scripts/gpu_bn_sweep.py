@@ -27,6 +27,13 @@ SHAPES = [
     # CIFAR bs=64
     (64 * 8 * 8, 64),
     (64 * 1 * 1, 512),
+    # slab-rule shapes: mid/small M with big C (r50@224 layer3/4 leftovers,
+    # mobilenet head)
+    (32 * 14 * 14, 512),
+    (32 * 7 * 7, 1024),
+    (64 * 4 * 4, 256),
+    (64 * 2 * 2, 512),
+    (49 * 8, 1280),
 ]
 
 
@@ -54,8 +61,19 @@ def main():
                                sdzx, M, C, 2)
         torch.cuda.synchronize()
         dt = (time.perf_counter() - t0) / it
+        # correctness vs fp32 torch reference (mask 2: bn(x) > 0 gate)
+        sdz.zero_(); sdzx.zero_()
+        C_.bn_reduce_bench(dy, y, x, mean, invstd, gamma, beta, sdz,
+                           sdzx, M, C, 2)
+        g = dy.float() * ((gamma * invstd * x.float()
+                           + (beta - mean * gamma * invstd)) > 0)
+        r1 = g.sum(0)
+        r2 = (g * (x.float() - mean) * invstd).sum(0)
+        e1 = (sdz - r1).abs().max() / (r1.abs().max() + 1e-6)
+        e2 = (sdzx - r2).abs().max() / (r2.abs().max() + 1e-6)
+        ok = "ok" if max(e1.item(), e2.item()) < 2e-2 else "MISMATCH"
         gb = 2 * M * C * 2 / 1e9  # dy + x read, bf16
-        print(f"M={M:>9} C={C:>5}: {dt * 1e6:8.1f} us  {gb / dt:7.1f} GB/s")
+        print(f"M={M:>9} C={C:>5}: {dt * 1e6:8.1f} us  {gb / dt:7.1f} GB/s  {ok}")
 
 
 if __name__ == "__main__":

@@ -831,3 +831,37 @@ def test_deterministic_mode_bitwise_reproducible():
         assert torch.equal(outs[0][1], outs[1][1]), "masters differ"
     finally:
         _C().set_deterministic(False)
+
+
+@pytest.mark.parametrize("M,C,mask", [
+    (1568, 2048, 2),   # r50@224 layer4 expand: channel-slab v8 (4 slabs)
+    (6272, 1024, 1),   # r50@224 layer3 expand: 2 slabs, y-mask
+    (1568, 512, 0),    # mid-M full-slab boundary, no activation
+    (392, 1280, 2),    # mobilenet head: non-pow2 slab (1280 = 4x320)
+])
+def test_bn_reduce_slab_dispatch_matches_ref(M, C, mask):
+    """BN-backward reduce through the public launcher at shapes the
+    channel-slab v8 rule admits (C > 512 sliced into <=512-channel slabs,
+    elem_kernels.hip bn_v8_pick) vs a plain fp32 reference."""
+    torch.manual_seed(M + C)
+    dy = torch.randn(M, C, device="cuda").bfloat16()
+    y = torch.randn(M, C, device="cuda").bfloat16()
+    x = torch.randn(M, C, device="cuda").bfloat16()
+    mean = torch.randn(C, device="cuda")
+    invstd = torch.rand(C, device="cuda") + 0.5
+    gamma = torch.randn(C, device="cuda")
+    beta = torch.randn(C, device="cuda")
+    sdz = torch.zeros(C, device="cuda")
+    sdzx = torch.zeros(C, device="cuda")
+    _C().bn_reduce_bench(dy, y, x, mean, invstd, gamma, beta, sdz, sdzx,
+                         M, C, mask)
+    g = dy.float()
+    if mask == 1:
+        g = g * (y.float() > 0)
+    elif mask == 2:
+        g = g * ((gamma * invstd * x.float()
+                  + (beta - mean * gamma * invstd)) > 0)
+    r1 = g.sum(0)
+    r2 = (g * (x.float() - mean) * invstd).sum(0)
+    assert rel(sdz, r1) < 1e-2, f"sum_dz rel={rel(sdz, r1)}"
+    assert rel(sdzx, r2) < 1e-2, f"sum_dzx rel={rel(sdzx, r2)}"
