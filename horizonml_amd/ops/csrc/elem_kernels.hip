@@ -551,14 +551,17 @@ __global__ __launch_bounds__(256) void k_bnact_bwd_reduce(
 }
 
 // Pass 2: dconv = gamma·invstd·(dz − Σdz/M − xhat·Σdzx/M); optional dres =
-// dz.  V8 per thread (C % 8 == 0).
+// dz.  V8 per thread (C % 8 == 0).  MASK compile-time like the reduce
+// kernels: with runtime mask_mode all five paths landed in ONE 694-instr
+// main loop (instruction-bound, ~1.6x off the HBM roofline at r50@224).
+template <int MASK>
 __global__ __launch_bounds__(256) void k_bn_bwd_apply(
     const bf16* __restrict__ dy, const bf16* __restrict__ yout,
     const bf16* __restrict__ x, const float* __restrict__ save_mean,
     const float* __restrict__ save_invstd, const float* __restrict__ gamma,
     const float* __restrict__ beta, const float* __restrict__ sum_dz,
     const float* __restrict__ sum_dzx, bf16* __restrict__ dconv,
-    bf16* __restrict__ dres, long M, int C, int mask_mode) {
+    bf16* __restrict__ dres, long M, int C) {
   const float invM = 1.f / (float)M;
   if (C % 8 != 0) {  // scalar fallback
     long total = M * (long)C;
@@ -568,15 +571,15 @@ __global__ __launch_bounds__(256) void k_bn_bwd_apply(
       float mean = save_mean[c], invstd = save_invstd[c];
       float g = b2f(dy[i]);
       float xv = b2f(x[i]);
-      if (mask_mode == 1) {
+      if (MASK == 1) {
         if (b2f(yout[i]) <= 0.f) g = 0.f;
-      } else if (mask_mode == 2) {
+      } else if (MASK == 2) {
         float ga = gamma[c] * invstd;
         if (fmaf(ga, xv, beta[c] - mean * ga) <= 0.f) g = 0.f;
-      } else if (mask_mode == 3) {
+      } else if (MASK == 3) {
         float yv = b2f(yout[i]);
         if (yv <= 0.f || yv >= 6.f) g = 0.f;
-      } else if (mask_mode == 4) {
+      } else if (MASK == 4) {
         float ga = gamma[c] * invstd;
         float bn = fmaf(ga, xv, beta[c] - mean * ga);
         if (bn <= 0.f || bn >= 6.f) g = 0.f;
@@ -590,33 +593,36 @@ __global__ __launch_bounds__(256) void k_bn_bwd_apply(
     return;
   }
   long total8 = M * (long)C / 8;
-  for (long t = (long)blockIdx.x * blockDim.x + threadIdx.x; t < total8;
-       t += (long)gridDim.x * blockDim.x) {
-    long i = t * 8;
-    int c0 = (int)(i % C);
+  // channel index kept incrementally (conditional subtract) — the 64-bit
+  // modulo per tile fed the critical path of every loop iteration
+  const long stride = (long)gridDim.x * blockDim.x;
+  long i = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  int c0 = (int)(i % C);
+  const int sd = (int)((stride * 8) % C);
+  for (long t = i / 8; t < total8; t += stride, i += stride * 8) {
     F8 mean8 = load_f8(save_mean + c0), istd8 = load_f8(save_invstd + c0);
     F8 g8 = load_f8(gamma + c0);
     F8 sdz8 = load_f8(sum_dz + c0), sdzx8 = load_f8(sum_dzx + c0);
     V8 dy8, x8, y8;
     dy8.u = *(const uint4*)(dy + i);
     x8.u = *(const uint4*)(x + i);
-    if (mask_mode == 1 || mask_mode == 3) y8.u = *(const uint4*)(yout + i);
+    if (MASK == 1 || MASK == 3) y8.u = *(const uint4*)(yout + i);
     F8 b8;
-    if (mask_mode == 2 || mask_mode == 4) b8 = load_f8(beta + c0);
+    if (MASK == 2 || MASK == 4) b8 = load_f8(beta + c0);
     V8 dc8, dr8;
 #pragma unroll
     for (int e = 0; e < 8; e++) {
       float g = b2f(dy8.e[e]);
       float xv = b2f(x8.e[e]);
-      if (mask_mode == 1) {
+      if (MASK == 1) {
         if (b2f(y8.e[e]) <= 0.f) g = 0.f;
-      } else if (mask_mode == 2) {
+      } else if (MASK == 2) {
         float ga = g8.f[e] * istd8.f[e];
         if (fmaf(ga, xv, b8.f[e] - mean8.f[e] * ga) <= 0.f) g = 0.f;
-      } else if (mask_mode == 3) {
+      } else if (MASK == 3) {
         float yv = b2f(y8.e[e]);
         if (yv <= 0.f || yv >= 6.f) g = 0.f;
-      } else if (mask_mode == 4) {
+      } else if (MASK == 4) {
         float ga = g8.f[e] * istd8.f[e];
         float bn = fmaf(ga, xv, b8.f[e] - mean8.f[e] * ga);
         if (bn <= 0.f || bn >= 6.f) g = 0.f;
@@ -629,6 +635,8 @@ __global__ __launch_bounds__(256) void k_bn_bwd_apply(
     }
     if (dres != nullptr) *(uint4*)(dres + i) = dr8.u;
     *(uint4*)(dconv + i) = dc8.u;
+    c0 += sd;
+    if (c0 >= C) c0 -= C;
   }
 }
 
@@ -1621,10 +1629,17 @@ void launch_bn_bwd_apply(const void* dy, const void* yout, const void* x,
                          const float* sum_dz, const float* sum_dzx,
                          void* dconv, void* dres, long M, int C,
                          int mask_mode, hipStream_t st) {
-  k_bn_bwd_apply<<<gsz(M * (long)C / 8 + 1), 256, 0, st>>>(
-      (const bf16*)dy, (const bf16*)yout, (const bf16*)x, smean, sinvstd,
-      gamma, beta, sum_dz, sum_dzx, (bf16*)dconv, (bf16*)dres, M, C,
-      mask_mode);
+#define LA(MK) k_bn_bwd_apply<MK><<<gsz(M * (long)C / 8 + 1), 256, 0, \
+    st>>>((const bf16*)dy, (const bf16*)yout, (const bf16*)x, smean, \
+    sinvstd, gamma, beta, sum_dz, sum_dzx, (bf16*)dconv, (bf16*)dres, M, C)
+  switch (mask_mode) {
+    case 1: LA(1); break;
+    case 2: LA(2); break;
+    case 3: LA(3); break;
+    case 4: LA(4); break;
+    default: LA(0); break;
+  }
+#undef LA
 }
 
 void launch_dw_fwd(const void* x, const void* w, void* y, float* stats,
