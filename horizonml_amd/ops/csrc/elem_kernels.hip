@@ -28,7 +28,10 @@ DEV F8 load_f8(const float* p) {
   return v;
 }
 
-template <bool F32SRC>
+// ACT/TRAIN compile-time (mirrors k_bn_bwd_apply's MASK): the runtime
+// branches kept ~390-instruction specialized loops alive; templated the
+// V8 loop is lean and the eval path drops out of training kernels.
+template <bool F32SRC, int ACT, bool TRAIN>
 __global__ __launch_bounds__(256) void k_bn_apply(
     const void* __restrict__ xv, const bf16* __restrict__ res,
     bf16* __restrict__ y, bf16* __restrict__ convout,
@@ -36,13 +39,12 @@ __global__ __launch_bounds__(256) void k_bn_apply(
     const float* __restrict__ gamma, const float* __restrict__ beta,
     float* __restrict__ running_mean, float* __restrict__ running_var,
     float* __restrict__ save_mean, float* __restrict__ save_invstd,
-    long M, int C, float momentum, float eps, int training, int act,
-    int nsplit) {
+    long M, int C, float momentum, float eps, int nsplit) {
   const long slab = M * (long)C;
   const bf16* xb = (const bf16*)xv;
   const float* xf = (const float*)xv;
   const float invM = 1.f / (float)M;
-  if (training && blockIdx.x == 0) {
+  if (TRAIN && blockIdx.x == 0) {
     for (int c = threadIdx.x; c < C; c += blockDim.x) {
       float mean = stats[c] * invM;
       float var = fmaxf(stats[C + c] * invM - mean * mean, 0.f);
@@ -60,7 +62,7 @@ __global__ __launch_bounds__(256) void k_bn_apply(
          i += (long)gridDim.x * blockDim.x) {
       int c = (int)(i % C);
       float mean, invstd;
-      if (training) {
+      if (TRAIN) {
         mean = stats[c] * invM;
         float var = fmaxf(stats[C + c] * invM - mean * mean, 0.f);
         invstd = rsqrtf(var + eps);
@@ -78,8 +80,8 @@ __global__ __launch_bounds__(256) void k_bn_apply(
       if (F32SRC && convout != nullptr) convout[i] = f2b(xi);
       float v = (xi - mean) * invstd * gamma[c] + beta[c];
       if (res != nullptr) v += b2f(res[i]);
-      if (act == 1) v = fmaxf(v, 0.f);
-      else if (act == 2) v = fminf(fmaxf(v, 0.f), 6.f);
+      if (ACT == 1) v = fmaxf(v, 0.f);
+      else if (ACT == 2) v = fminf(fmaxf(v, 0.f), 6.f);
       y[i] = f2b(v);
     }
     return;
@@ -111,7 +113,7 @@ __global__ __launch_bounds__(256) void k_bn_apply(
     }
     F8 g8 = load_f8(gamma + c0), b8 = load_f8(beta + c0);
     F8 mean8, istd8;
-    if (training) {
+    if (TRAIN) {
       F8 s1 = load_f8(stats + c0), s2 = load_f8(stats + C + c0);
 #pragma unroll
       for (int e = 0; e < 8; e++) {
@@ -134,9 +136,104 @@ __global__ __launch_bounds__(256) void k_bn_apply(
     for (int e = 0; e < 8; e++) {
       float v = (xi.f[e] - mean8.f[e]) * istd8.f[e] * g8.f[e] + b8.f[e];
       if (res != nullptr) v += b2f(r8.e[e]);
-      if (act == 1) v = fmaxf(v, 0.f);
-      else if (act == 2) v = fminf(fmaxf(v, 0.f), 6.f);  // ReLU6
+      if (ACT == 1) v = fmaxf(v, 0.f);
+      else if (ACT == 2) v = fminf(fmaxf(v, 0.f), 6.f);  // ReLU6
       out.e[e] = f2b(v);
+    }
+    *(uint4*)(y + i) = out.u;
+  }
+}
+
+// c-blocked BN forward apply (the reduce kernels' slab geometry): each
+// block owns a <=512-channel slab x m-chunk, so per-channel mean/invstd/
+// gamma/beta fold into registers ONCE per block and the m-loop body is
+// load -> 8 fma -> store.  The flat elementwise k_bn_apply recomputed the
+// channel stats (incl. 8 precise rsqrtf) for EVERY 8-element tile —
+// ~350-instruction loops, ~1.6x off the HBM roofline at r50@224.
+template <bool F32SRC, int ACT, bool TRAIN>
+__global__ __launch_bounds__(256) void k_bn_apply_v8(
+    const void* __restrict__ xv, const bf16* __restrict__ res,
+    bf16* __restrict__ y, bf16* __restrict__ convout,
+    const float* __restrict__ stats,
+    const float* __restrict__ gamma, const float* __restrict__ beta,
+    float* __restrict__ running_mean, float* __restrict__ running_var,
+    float* __restrict__ save_mean, float* __restrict__ save_invstd,
+    long M, int C, float momentum, float eps, int nsplit, long mchunk,
+    int cslab) {
+  const long slab = M * (long)C;
+  const bf16* xb = (const bf16*)xv;
+  const float* xf = (const float*)xv;
+  const float invM = 1.f / (float)M;
+  if (TRAIN && blockIdx.x == 0 && blockIdx.y == 0) {
+    for (int c = threadIdx.x; c < C; c += blockDim.x) {
+      float mean = stats[c] * invM;
+      float var = fmaxf(stats[C + c] * invM - mean * mean, 0.f);
+      save_mean[c] = mean;
+      save_invstd[c] = rsqrtf(var + eps);
+      float ub = (M > 1) ? var * (float)M / (float)(M - 1) : var;
+      running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * mean;
+      running_var[c] = (1.f - momentum) * running_var[c] + momentum * ub;
+    }
+  }
+  const int cbeg = blockIdx.x * cslab;
+  const int lpr = cslab >> 3;
+  const int mstep = 256 / lpr;
+  const int tid = threadIdx.x;
+  if (tid >= mstep * lpr) return;
+  const int th_c = cbeg + (tid % lpr) * 8;
+  const int th_m = tid / lpr;
+  const long mbeg = (long)blockIdx.y * mchunk;
+  const long mend = min(M, mbeg + mchunk);
+  float ga[8], be[8];
+#pragma unroll
+  for (int e = 0; e < 8; e++) {
+    int c = th_c + e;
+    float mean, istd;
+    if (TRAIN) {
+      mean = stats[c] * invM;
+      float var = fmaxf(stats[C + c] * invM - mean * mean, 0.f);
+      istd = rsqrtf(var + eps);
+    } else {
+      mean = running_mean[c];
+      istd = rsqrtf(running_var[c] + eps);
+    }
+    ga[e] = gamma[c] * istd;
+    be[e] = beta[c] - mean * ga[e];   // y = ga*x + be  [+res] [act]
+  }
+  for (long m = mbeg + th_m; m < mend; m += mstep) {
+    const long i = m * C + th_c;
+    float v[8];
+    if (F32SRC) {
+#pragma unroll
+      for (int q = 0; q < 8; q += 4)
+        *(float4*)&v[q] = *(const float4*)(xf + i + q);
+      for (int z = 1; z < nsplit; z++)
+#pragma unroll
+        for (int q = 0; q < 8; q += 4) {
+          float4 w = *(const float4*)(xf + z * slab + i + q);
+          v[q] += w.x; v[q + 1] += w.y; v[q + 2] += w.z; v[q + 3] += w.w;
+        }
+      if (convout != nullptr) {
+        V8 co;
+#pragma unroll
+        for (int e = 0; e < 8; e++) co.e[e] = f2b(v[e]);
+        *(uint4*)(convout + i) = co.u;
+      }
+    } else {
+      V8 x8;
+      x8.u = *(const uint4*)(xb + i);
+#pragma unroll
+      for (int e = 0; e < 8; e++) v[e] = b2f(x8.e[e]);
+    }
+    V8 r8, out;
+    if (res != nullptr) r8.u = *(const uint4*)(res + i);
+#pragma unroll
+    for (int e = 0; e < 8; e++) {
+      float t = fmaf(ga[e], v[e], be[e]);
+      if (res != nullptr) t += b2f(r8.e[e]);
+      if (ACT == 1) t = fmaxf(t, 0.f);
+      else if (ACT == 2) t = fminf(fmaxf(t, 0.f), 6.f);  // ReLU6
+      out.e[e] = f2b(t);
     }
     *(uint4*)(y + i) = out.u;
   }
@@ -1415,15 +1512,61 @@ void launch_stats_bf16_det(const void* x, float* stats, long M, int C,
                                                   C);
 }
 
+static int bn_v8_iters();
+static int bn_v8_cslab(int C);
+
+// grid sizing for the c-blocked apply kernels: <=512-ch slabs x
+// m-chunks.  Unlike the reduce kernels (whose per-block LDS epilogue +
+// atomics reward long m-loops, bn_v8_iters), apply has no epilogue and
+// its f32 split-K slab-sum is LATENCY-bound — one m-row per thread,
+// i.e. maximum thread parallelism, measured 4x faster at the small
+// CIFAR shapes (30.2 -> ~7 us); the 768-block cap keeps big-M shapes
+// at the same fill as before.
+static dim3 bn_apply_grid(long M, int C, long* mchunk_out, int* cslab_out) {
+  int cslab = bn_v8_cslab(C), nslab = C / cslab;
+  int lpr = cslab >> 3, mstep = 256 / lpr;
+  int msplit = (int)min((long)max(1, 768 / nslab),
+                        max((long)1, (M + mstep - 1) / mstep));
+  long mchunk = (M + msplit - 1) / msplit;
+  msplit = (int)((M + mchunk - 1) / mchunk);
+  *mchunk_out = mchunk;
+  *cslab_out = cslab;
+  return dim3(nslab, msplit);
+}
+
 void launch_bn_apply(const void* x, const void* res, void* y,
                      const float* stats, const float* gamma,
                      const float* beta, float* rmean, float* rvar,
                      float* smean, float* sinvstd, long M, int C,
                      float momentum, float eps, int training, int act,
                      hipStream_t st) {
-  k_bn_apply<false><<<gsz(M * (long)C / 8 + 1), 256, 0, st>>>(
-      x, (const bf16*)res, (bf16*)y, nullptr, stats, gamma, beta, rmean,
-      rvar, smean, sinvstd, M, C, momentum, eps, training, act, 1);
+  if ((C & 7) == 0) {
+    long mchunk; int cslab;
+    dim3 grid = bn_apply_grid(M, C, &mchunk, &cslab);
+#define LF(AC, TR) k_bn_apply_v8<false, AC, TR><<<grid, 256, 0, st>>>( \
+    x, (const bf16*)res, (bf16*)y, nullptr, stats, gamma, beta, rmean, \
+    rvar, smean, sinvstd, M, C, momentum, eps, 1, mchunk, cslab)
+    if (training) {
+      if (act == 1) LF(1, true); else if (act == 2) LF(2, true);
+      else LF(0, true);
+    } else {
+      if (act == 1) LF(1, false); else if (act == 2) LF(2, false);
+      else LF(0, false);
+    }
+#undef LF
+    return;
+  }
+#define LF(AC, TR) k_bn_apply<false, AC, TR><<<gsz(M * (long)C / 8 + 1), \
+    256, 0, st>>>(x, (const bf16*)res, (bf16*)y, nullptr, stats, gamma, \
+    beta, rmean, rvar, smean, sinvstd, M, C, momentum, eps, 1)
+  if (training) {
+    if (act == 1) LF(1, true); else if (act == 2) LF(2, true);
+    else LF(0, true);
+  } else {
+    if (act == 1) LF(1, false); else if (act == 2) LF(2, false);
+    else LF(0, false);
+  }
+#undef LF
 }
 
 void launch_bn_apply_f32(const float* ws, const void* res, void* y,
@@ -1432,10 +1575,33 @@ void launch_bn_apply_f32(const float* ws, const void* res, void* y,
                          float* rvar, float* smean, float* sinvstd, long M,
                          int C, float momentum, float eps, int training,
                          int act, int nsplit, hipStream_t st) {
-  k_bn_apply<true><<<gsz(M * (long)C / 8 + 1), 256, 0, st>>>(
-      ws, (const bf16*)res, (bf16*)y, (bf16*)convout, stats, gamma, beta,
-      rmean, rvar, smean, sinvstd, M, C, momentum, eps, training, act,
-      nsplit);
+  if ((C & 7) == 0) {
+    long mchunk; int cslab;
+    dim3 grid = bn_apply_grid(M, C, &mchunk, &cslab);
+#define LF(AC, TR) k_bn_apply_v8<true, AC, TR><<<grid, 256, 0, st>>>( \
+    ws, (const bf16*)res, (bf16*)y, (bf16*)convout, stats, gamma, beta, \
+    rmean, rvar, smean, sinvstd, M, C, momentum, eps, nsplit, mchunk, cslab)
+    if (training) {
+      if (act == 1) LF(1, true); else if (act == 2) LF(2, true);
+      else LF(0, true);
+    } else {
+      if (act == 1) LF(1, false); else if (act == 2) LF(2, false);
+      else LF(0, false);
+    }
+#undef LF
+    return;
+  }
+#define LF(AC, TR) k_bn_apply<true, AC, TR><<<gsz(M * (long)C / 8 + 1), \
+    256, 0, st>>>(ws, (const bf16*)res, (bf16*)y, (bf16*)convout, stats, \
+    gamma, beta, rmean, rvar, smean, sinvstd, M, C, momentum, eps, nsplit)
+  if (training) {
+    if (act == 1) LF(1, true); else if (act == 2) LF(2, true);
+    else LF(0, true);
+  } else {
+    if (act == 1) LF(1, false); else if (act == 2) LF(2, false);
+    else LF(0, false);
+  }
+#undef LF
 }
 
 void launch_stats_reduce(const float* ws, float* stats, long M, int C,
