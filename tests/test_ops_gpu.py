@@ -359,15 +359,29 @@ def test_normalize_u8_kernel():
 
 
 def test_permute_krsc_rsck():
+    """Batched KRSC->RSCK transpose over a packed multi-conv buffer —
+    includes the stem's odd C=3 (scalar path) and 32-misaligned tails
+    alongside vector-eligible even shapes."""
     torch.manual_seed(0)
-    K, R, S, C = 32, 3, 3, 16
-    w = torch.randn(K, R, S, C, device="cuda").bfloat16().contiguous()
-    dst = torch.empty(R * S * C * K, device="cuda", dtype=torch.bfloat16)
-    meta = torch.tensor([[0, 0, K * R * S * C, (K << 16) | C]],
-                        dtype=torch.int32, device="cuda")
-    _C().permute_krsc_rsck(w.flatten(), dst, meta, K * R * S * C)
-    ref = w.float().permute(1, 2, 3, 0).reshape(R * S, C, K).flatten()
-    assert rel(dst, ref) == 0 or rel(dst, ref) < 1e-6
+    shapes = [(32, 3, 3, 16), (64, 7, 7, 3), (48, 1, 1, 24), (40, 3, 3, 8)]
+    ws, metas, total = [], [], 0
+    for K, R, S, C in shapes:
+        w = torch.randn(K, R, S, C, device="cuda").bfloat16().contiguous()
+        n = K * R * S * C
+        metas.append([total, total, n, (K << 16) | C])
+        ws.append(w)
+        total += n
+    src = torch.cat([w.flatten() for w in ws])
+    dst = torch.empty(total, device="cuda", dtype=torch.bfloat16)
+    meta = torch.tensor(metas, dtype=torch.int32, device="cuda")
+    _C().permute_krsc_rsck(src, dst, meta, max(m[2] for m in metas))
+    off = 0
+    for (K, R, S, C), w in zip(shapes, ws):
+        n = K * R * S * C
+        ref = w.float().permute(1, 2, 3, 0).reshape(R * S, C, K).flatten()
+        r = rel(dst[off:off + n], ref)
+        assert r == 0 or r < 1e-6, f"K{K} C{C}: rel={r}"
+        off += n
 
 
 # -------------------------------------------------------------- end-to-end --
