@@ -647,6 +647,69 @@ __global__ __launch_bounds__(256) void k_bnact_bwd_reduce(
   }
 }
 
+// c-blocked BN backward apply (bn_apply_v8's slab geometry): per-channel
+// coefficients fold to registers once per block and the math reduces to
+// two fmas per element: dconv = A·dz − D·x + E with A = γ·invstd,
+// D = A·invstd·Σdzx/M, E = D·mean − A·Σdz/M.  The flat per-tile kernel
+// below re-loaded 5-6 channel vectors per 8 elements.
+template <int MASK>
+__global__ __launch_bounds__(256) void k_bn_bwd_apply_v8(
+    const bf16* __restrict__ dy, const bf16* __restrict__ yout,
+    const bf16* __restrict__ x, const float* __restrict__ save_mean,
+    const float* __restrict__ save_invstd, const float* __restrict__ gamma,
+    const float* __restrict__ beta, const float* __restrict__ sum_dz,
+    const float* __restrict__ sum_dzx, bf16* __restrict__ dconv,
+    bf16* __restrict__ dres, long M, int C, long mchunk, int cslab) {
+  const float invM = 1.f / (float)M;
+  const int cbeg = blockIdx.x * cslab;
+  const int lpr = cslab >> 3;
+  const int mstep = 256 / lpr;
+  const int tid = threadIdx.x;
+  if (tid >= mstep * lpr) return;
+  const int th_c = cbeg + (tid % lpr) * 8;
+  const int th_m = tid / lpr;
+  const long mbeg = (long)blockIdx.y * mchunk;
+  const long mend = min(M, mbeg + mchunk);
+  float A[8], D[8], E[8], GB[8];
+#pragma unroll
+  for (int e = 0; e < 8; e++) {
+    int c = th_c + e;
+    float mean = save_mean[c], istd = save_invstd[c];
+    A[e] = gamma[c] * istd;
+    D[e] = A[e] * istd * sum_dzx[c] * invM;
+    E[e] = D[e] * mean - A[e] * sum_dz[c] * invM;
+    if (MASK == 2 || MASK == 4) GB[e] = beta[c] - mean * A[e];
+  }
+  for (long m = mbeg + th_m; m < mend; m += mstep) {
+    const long i = m * C + th_c;
+    V8 dy8, x8, y8, dc8, dr8;
+    dy8.u = *(const uint4*)(dy + i);
+    x8.u = *(const uint4*)(x + i);
+    if (MASK == 1 || MASK == 3) y8.u = *(const uint4*)(yout + i);
+#pragma unroll
+    for (int e = 0; e < 8; e++) {
+      float g = b2f(dy8.e[e]);
+      float xv = b2f(x8.e[e]);
+      if (MASK == 1) {
+        if (b2f(y8.e[e]) <= 0.f) g = 0.f;
+      } else if (MASK == 2) {
+        if (fmaf(A[e], xv, GB[e]) <= 0.f) g = 0.f;
+      } else if (MASK == 3) {
+        float yv = b2f(y8.e[e]);
+        if (yv <= 0.f || yv >= 6.f) g = 0.f;
+      } else if (MASK == 4) {
+        float bn = fmaf(A[e], xv, GB[e]);
+        if (bn <= 0.f || bn >= 6.f) g = 0.f;
+      }
+      dr8.e[e] = f2b(g);
+      float t = fmaf(A[e], g, E[e]);
+      dc8.e[e] = f2b(fmaf(-D[e], xv, t));
+    }
+    if (dres != nullptr) *(uint4*)(dres + i) = dr8.u;
+    *(uint4*)(dconv + i) = dc8.u;
+  }
+}
+
 // Pass 2: dconv = gamma·invstd·(dz − Σdz/M − xhat·Σdzx/M); optional dres =
 // dz.  V8 per thread (C % 8 == 0).  MASK compile-time like the reduce
 // kernels: with runtime mask_mode all five paths landed in ONE 694-instr
@@ -1795,6 +1858,23 @@ void launch_bn_bwd_apply(const void* dy, const void* yout, const void* x,
                          const float* sum_dz, const float* sum_dzx,
                          void* dconv, void* dres, long M, int C,
                          int mask_mode, hipStream_t st) {
+  if ((C & 7) == 0) {
+    long mchunk; int cslab;
+    dim3 grid = bn_apply_grid(M, C, &mchunk, &cslab);
+#define LA(MK) k_bn_bwd_apply_v8<MK><<<grid, 256, 0, st>>>( \
+    (const bf16*)dy, (const bf16*)yout, (const bf16*)x, smean, sinvstd, \
+    gamma, beta, sum_dz, sum_dzx, (bf16*)dconv, (bf16*)dres, M, C, \
+    mchunk, cslab)
+    switch (mask_mode) {
+      case 1: LA(1); break;
+      case 2: LA(2); break;
+      case 3: LA(3); break;
+      case 4: LA(4); break;
+      default: LA(0); break;
+    }
+#undef LA
+    return;
+  }
 #define LA(MK) k_bn_bwd_apply<MK><<<gsz(M * (long)C / 8 + 1), 256, 0, \
     st>>>((const bf16*)dy, (const bf16*)yout, (const bf16*)x, smean, \
     sinvstd, gamma, beta, sum_dz, sum_dzx, (bf16*)dconv, (bf16*)dres, M, C)
