@@ -7,7 +7,7 @@ import pytest
 import torch
 import torch.multiprocessing as mp
 
-from horizonml_amd.models import resnet18
+from horizonml_amd.models import mobilenet_v2, resnet18, resnet50
 from horizonml_amd.parallel.flat_reducer import (BackwardBucketScheduler,
                                                  FlatBucketReducer,
                                                  build_bucket_schedule,
@@ -41,6 +41,26 @@ def test_grad_units_tile_resnet18():
     assert len(units) == 10
     unit_params = [id(p) for _, ps in units for p in ps]
     assert unit_params == [id(p) for p in m.parameters()]
+
+
+@pytest.mark.parametrize("model_fn", [resnet50, mobilenet_v2])
+@pytest.mark.parametrize("n_buckets", [2, 4])
+def test_build_bucket_schedule_other_models(model_fn, n_buckets):
+    """The bucket schedule must tile the flat buffer exactly for every
+    shipped model family (bottleneck ResNet50, inverted-residual
+    MobileNetV2), not just the flagship ResNet18 layout."""
+    m = model_fn(num_classes=10)
+    slices, total = _fake_slices(m)
+    ranges, mods = build_bucket_schedule(m, slices, n_buckets)
+    units = grad_units(m)
+    unit_params = [id(p) for _, ps in units for p in ps]
+    assert unit_params == [id(p) for p in m.parameters()]
+    assert ranges[0][1] == total and ranges[-1][0] == 0
+    cover = 0
+    for lo, hi in sorted(ranges):
+        assert lo == cover
+        cover = hi
+    assert cover == total
 
 
 @pytest.mark.parametrize("n_buckets", [1, 2, 4, 8])
