@@ -1603,7 +1603,10 @@ void launch_bn_apply(const void* x, const void* res, void* y,
                      float* smean, float* sinvstd, long M, int C,
                      float momentum, float eps, int training, int act,
                      hipStream_t st) {
-  if ((C & 7) == 0) {
+  if ((C & 7) == 0 && bn_v8_cslab(C) <= 2048) {
+    // slab <= 2048 keeps lpr <= 256 (a C = 8 x large-prime extent has no
+    // viable slab divisor and would make mstep = 0) — such channel
+    // counts fall back to the flat elementwise kernel
     long mchunk; int cslab;
     dim3 grid = bn_apply_grid(M, C, &mchunk, &cslab);
 #define LF(AC, TR) k_bn_apply_v8<false, AC, TR><<<grid, 256, 0, st>>>( \
@@ -1638,7 +1641,10 @@ void launch_bn_apply_f32(const float* ws, const void* res, void* y,
                          float* rvar, float* smean, float* sinvstd, long M,
                          int C, float momentum, float eps, int training,
                          int act, int nsplit, hipStream_t st) {
-  if ((C & 7) == 0) {
+  if ((C & 7) == 0 && bn_v8_cslab(C) <= 2048) {
+    // slab <= 2048 keeps lpr <= 256 (a C = 8 x large-prime extent has no
+    // viable slab divisor and would make mstep = 0) — such channel
+    // counts fall back to the flat elementwise kernel
     long mchunk; int cslab;
     dim3 grid = bn_apply_grid(M, C, &mchunk, &cslab);
 #define LF(AC, TR) k_bn_apply_v8<true, AC, TR><<<grid, 256, 0, st>>>( \
@@ -1858,7 +1864,10 @@ void launch_bn_bwd_apply(const void* dy, const void* yout, const void* x,
                          const float* sum_dz, const float* sum_dzx,
                          void* dconv, void* dres, long M, int C,
                          int mask_mode, hipStream_t st) {
-  if ((C & 7) == 0) {
+  if ((C & 7) == 0 && bn_v8_cslab(C) <= 2048) {
+    // slab <= 2048 keeps lpr <= 256 (a C = 8 x large-prime extent has no
+    // viable slab divisor and would make mstep = 0) — such channel
+    // counts fall back to the flat elementwise kernel
     long mchunk; int cslab;
     dim3 grid = bn_apply_grid(M, C, &mchunk, &cslab);
 #define LA(MK) k_bn_bwd_apply_v8<MK><<<grid, 256, 0, st>>>( \
