@@ -662,6 +662,8 @@ def test_training_trajectory_tracks_cpu():
     (40, 24, 1, 1, 6, 11),    # 1x1
     (3, 40, 3, 1, 9, 6),      # scalar-gather path, odd spatial
     (56, 56, 3, 2, 7, 4),     # odd input spatial, stride 2
+    (16, 20, 3, 1, 8, 5),     # cout % 8 != 0: flat BN apply/bwd fallback
+    (20, 12, 1, 1, 6, 7),     # both dims % 8 != 0, 1x1
 ])
 def test_conv_shape_fuzz(cfg):
     """Off-grid shapes (non-multiple-of-64 channels, odd spatial, 5x5
