@@ -651,9 +651,47 @@ static inline int pick_splitk(int tiles, int Kd) {
   return sk;
 }
 
+static inline int pick_tile(int M, int N);
+static inline int pick_tile_dgrad(int M, int N);
+
+// Fill-split (HZ_SK_FILL / HZ_SK_FILL_DG, default 0 = off): when the
+// throughput 128x128 tile is picked but its grid lands under `fillmin`
+// blocks (r50@224 bs=32 mid-layers: 98-196 blocks on 256 CUs), split K
+// across blockIdx.z to recover occupancy instead of falling back to the
+// latency tile.  The split path pays the f32 workspace + consumer pass,
+// so this only makes sense for long-K shapes (kd floor re-used).
+static int sk_fill(int dgrad) {
+  static int vf = [] {
+    const char* e = getenv("HZ_SK_FILL");
+    return e ? atoi(e) : 0;
+  }();
+  static int vd = [] {
+    const char* e = getenv("HZ_SK_FILL_DG");
+    // swept on r50@224 bs=32 (same box): 8.536 -> 8.485 ms; the fwd
+    // variant (HZ_SK_FILL) measured flat and stays off — its split
+    // drags the stats pass + slab re-reads
+    return e ? atoi(e) : 256;
+  }();
+  return dgrad ? vd : vf;
+}
+
+static int fill_splitk(int blocks128, int Kd, int fillmin) {
+  int sk = cdiv_h(fillmin, blocks128);
+  int maxsk = cdiv_h(Kd, 32);
+  if (sk > maxsk) sk = maxsk;
+  if (sk > 8) sk = 8;
+  return sk;
+}
+
 extern "C" int conv_fwd_splitk(ConvP p) {
   int tiles = cdiv_h(p.K, 64) * cdiv_h(p.M, 64);
-  return pick_splitk(tiles, p.Kd);
+  int sk = pick_splitk(tiles, p.Kd);
+  if (sk == 1 && sk_fill(0) > 0 && p.Kd >= 512
+      && pick_tile(p.M, p.K) == 2) {
+    int blocks = cdiv_h(p.M, 128) * cdiv_h(p.K, 128);
+    if (blocks < sk_fill(0)) return fill_splitk(blocks, p.Kd, sk_fill(0));
+  }
+  return sk;
 }
 
 extern "C" int conv_dgrad_splitk(ConvP p) {
@@ -671,7 +709,14 @@ extern "C" int conv_dgrad_splitk(ConvP p) {
   }();
   int tiles = cdiv_h(p.C, 64) * cdiv_h(p.Nb * p.H * p.W, 64);
   int Kd = p.R * p.S * p.K;
-  if (tiles >= tile_min_dg || Kd < kd_min_dg) return 1;
+  if (tiles >= tile_min_dg || Kd < kd_min_dg) {
+    int M = p.Nb * p.H * p.W;
+    if (sk_fill(1) > 0 && Kd >= 512 && pick_tile_dgrad(M, p.C) == 2) {
+      int blocks = cdiv_h(M, 128) * cdiv_h(p.C, 128);
+      if (blocks < sk_fill(1)) return fill_splitk(blocks, Kd, sk_fill(1));
+    }
+    return 1;
+  }
   int sk = cdiv_h(256, tiles);
   int maxsk = cdiv_h(Kd, 32);
   if (sk > maxsk) sk = maxsk;
@@ -950,18 +995,21 @@ extern "C" void launch_conv_fwd_splitk(const void* x, const void* w,
                                        hipStream_t st) {
   int kchunk = cdiv_h(cdiv_h(p.Kd, splitk), 32) * 32;
   splitk = cdiv_h(p.Kd, kchunk);
-  dim3 grid(cdiv_h(p.K, 64), cdiv_h(p.M, 64), splitk);
   bool vec = (p.C % 8) == 0 && (p.Kd % 8) == 0;
   auto A = (const bf16*)x;
   auto B = (const bf16*)w;
-  if (vec)
-    k_conv_mfma<1, true, true, false, true>
-        <<<grid, 256, 0, st>>>(A, B, nullptr, ws, nullptr, p,
-                               make_magic(p, 1), p.K, kchunk, 0);
-  else
-    k_conv_mfma<1, false, false, false, true>
-        <<<grid, 256, 0, st>>>(A, B, nullptr, ws, nullptr, p,
-                               make_magic(p, 1), p.K, kchunk, 0);
+  MagicP mg = make_magic(p, 1);
+  // fill-split shapes keep their throughput tile; the classic small-M
+  // split shapes pick tile 0 (64x64) as before
+  int tile = pick_tile(p.M, p.K);
+#define SK_T(VA, VB, TM, TN)                                              k_conv_mfma<1, VA, VB, false, true, TM, TN><<<grid, 256, 0, st>>>(          A, B, nullptr, ws, nullptr, p, mg, p.K, kchunk, 0)
+#define DISPATCH(TM, TN)                                                  do {                                                                      dim3 grid(cdiv_h(p.K, TN), cdiv_h(p.M, TM), splitk);                    if (vec) SK_T(true, true, TM, TN);                                      else SK_T(false, false, TM, TN);                                      } while (0)
+  if (tile == 3) DISPATCH(256, 64);
+  else if (tile == 2) DISPATCH(128, 128);
+  else if (tile == 1) DISPATCH(128, 64);
+  else DISPATCH(64, 64);
+#undef DISPATCH
+#undef SK_T
 }
 
 extern "C" void launch_conv_dgrad(const void* dz, const void* w_rsck,
@@ -974,15 +1022,16 @@ extern "C" void launch_conv_dgrad(const void* dz, const void* w_rsck,
   if (splitk > 1) {
     int kchunk = cdiv_h(cdiv_h(p.Kd, splitk), 32) * 32;
     splitk = cdiv_h(p.Kd, kchunk);
-    dim3 grid(cdiv_h(p.C, 64), cdiv_h(p.M, 64), splitk);
-    if (vec)
-      k_conv_mfma<2, true, true, false, true>
-          <<<grid, 256, 0, st>>>(A, B, nullptr, ws, nullptr, p,
-                                 make_magic(p, 2), p.C, kchunk, 0);
-    else
-      k_conv_mfma<2, false, false, false, true>
-          <<<grid, 256, 0, st>>>(A, B, nullptr, ws, nullptr, p,
-                                 make_magic(p, 2), p.C, kchunk, 0);
+    MagicP mg = make_magic(p, 2);
+    int tile = pick_tile_dgrad(p.M, p.C);
+#define SKD_T(VA, VB, TM, TN)                                             k_conv_mfma<2, VA, VB, false, true, TM, TN><<<grid, 256, 0, st>>>(          A, B, nullptr, ws, nullptr, p, mg, p.C, kchunk, 0)
+#define DISPATCH(TM, TN)                                                  do {                                                                      dim3 grid(cdiv_h(p.C, TN), cdiv_h(p.M, TM), splitk);                    if (vec) SKD_T(true, true, TM, TN);                                     else SKD_T(false, false, TM, TN);                                     } while (0)
+    if (tile == 3) DISPATCH(256, 64);
+    else if (tile == 2) DISPATCH(128, 128);
+    else if (tile == 1) DISPATCH(128, 64);
+    else DISPATCH(64, 64);
+#undef DISPATCH
+#undef SKD_T
   } else {
     MagicP mg = make_magic(p, 2);
     int tile = pick_tile_dgrad(p.M, p.C);
