@@ -133,6 +133,24 @@ def test_conv_bn_relu_bwd(cfg):
     assert rel(gpu.bn_bias.grad, cpu.bn_bias.grad) < 5e-2
 
 
+def test_conv_bwd_fill_split_shape():
+    """Dgrad at the fill-split shape class (M=25088, N=128, kd=1152: the
+    picked 128x128 tile lands at 196 blocks < HZ_SK_FILL_DG=256, so the
+    backward splits K across blockIdx.z while keeping the throughput
+    tile) — numerics vs the CPU fp32 reference must hold through the f32
+    slab sum + consumer pass."""
+    torch.manual_seed(3)
+    cpu, gpu = _make_pair(128, 128, 3, 1)
+    x = torch.randn(32, 128, 28, 28)
+    xc = x.clone().requires_grad_(True)
+    xg = to_gpu_cl(x).requires_grad_(True)
+    cpu(xc).square().mean().backward()
+    gpu(xg).float().square().mean().backward()
+    assert rel(xg.grad, xc.grad) < 5e-2, f"dx rel={rel(xg.grad, xc.grad)}"
+    assert rel(gpu.weight.grad, cpu.weight.grad) < 5e-2
+    assert rel(gpu.bn_weight.grad, cpu.bn_weight.grad) < 5e-2
+
+
 def test_conv_residual_fused():
     cpu, gpu = _make_pair(64, 64, 3, 1)
     x = torch.randn(8, 64, 8, 8)
