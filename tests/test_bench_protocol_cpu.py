@@ -126,3 +126,19 @@ def test_bench_protocol_two_ranks():
     # record value is rounded to 2 decimals by the contract
     assert rec["value"] == pytest.approx(128 * 5 / out[0][2], abs=0.006)
     assert rec["config"]["comm"] == "bucketed4"
+
+
+def test_record_contract_variants():
+    """The --infer and --image-size record variants name their config the
+    way the judge reads them (model suffix + image field)."""
+    r = build_record(elapsed_s=0.2, steps=50, warmup=5, world=1,
+                     batch_size=128, model="resnet18", optimizer="adam",
+                     exec_mode="graph", final_loss=2.3, infer=True)
+    assert r["config"]["model"] == "resnet18_infer"
+    assert r["value"] == pytest.approx(128 * 50 / 0.2, rel=1e-6)
+    r224 = build_record(elapsed_s=1.0, steps=100, warmup=10, world=1,
+                        batch_size=32, model="resnet50", optimizer="adam",
+                        exec_mode="graph", final_loss=6.9, image_size=224)
+    assert r224["config"]["model"] == "resnet50_synthetic224"
+    assert r224["config"]["image"] == "3x224x224"
+    assert r224["ms_per_step"] == pytest.approx(10.0)
