@@ -180,8 +180,8 @@ def train_pp(ctx: DistContext, epochs: int, sample_size: int, logs_dir: str,
                 count += n
             prof.step_end()
 
+        t = prof.epoch_end()  # syncs: epoch_time includes the GPU tail
         epoch_time = time.time() - epoch_start
-        t = prof.epoch_end()
         gmem, gutil = sample_gpu_resources(ctx.device if ctx.is_gpu else None)
         # reference layout: real metrics on last rank, zeros elsewhere
         loss_v = (loss_sum / max(1, count)) if stage.is_last else 0.0

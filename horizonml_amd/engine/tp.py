@@ -124,9 +124,9 @@ def train_tp(ctx: DistContext, epochs: int, sample_size: int, logs_dir: str,
             if probe is not None:
                 probe.step()
             prof.step_end()
+        t = prof.epoch_end()  # syncs: epoch_time includes the GPU tail
         epoch_time = time.time() - epoch_start
         loss_v, acc_v = meters.epoch_values()
-        t = prof.epoch_end()
         gmem, gutil = sample_gpu_resources(ctx.device if ctx.is_gpu else None)
         m = EpochMetrics(
             epoch=epoch + 1, loss=loss_v, accuracy=acc_v,
