@@ -127,3 +127,33 @@ def test_fused_optimizer_state_roundtrip(tmp_path):
     import pytest
     with pytest.raises(ValueError):
         HorizonAdam(mgr).load_state_dict(sgd.state_dict())
+
+
+def test_hybrid_entrypoint_checkpoint_resume(tmp_path):
+    """Hybrid DPxPP checkpoint: per-rank files across both groups (each
+    stage owns distinct params, each DP replica its own file) and the
+    MIN-epoch agreement spans the WHOLE world, so a relaunch resumes all
+    four ranks on the same epoch."""
+    import os
+
+    import pandas as pd
+
+    from hybrid_parallel_train import run_hybrid_parallel
+
+    logs = str(tmp_path / "logs")
+    ckpt = str(tmp_path / "hy.ckpt")
+    run_hybrid_parallel(dp_size=2, pp_size=2, epochs=1, sample_size=32,
+                        logs_dir=logs, batch_size=16, model_name="resnet18",
+                        backend="gloo", synthetic=True,
+                        checkpoint_path=ckpt)
+    for r in range(4):
+        assert os.path.isfile(f"{ckpt}.rank{r}"), f"missing rank{r} file"
+    logs2 = str(tmp_path / "logs2")
+    run_hybrid_parallel(dp_size=2, pp_size=2, epochs=3, sample_size=32,
+                        logs_dir=logs2, batch_size=16,
+                        model_name="resnet18", backend="gloo",
+                        synthetic=True, checkpoint_path=ckpt)
+    for r in range(4):
+        df = pd.read_csv(f"{logs2}/worker_{r}_samples_32.csv")
+        assert list(df["epoch"]) == [2, 3], \
+            f"rank {r} resumed wrong epochs: {list(df['epoch'])}"
