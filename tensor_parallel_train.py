@@ -21,13 +21,14 @@ def run_tensor_parallel(world_size: int, epochs: int, sample_size: int,
                         logs_dir: str = "tensor_parallel_logs",
                         batch_size: int = 64, backend=None, synthetic=None,
                         lr: float = 1e-3, optimizer_name: str = "adam",
-                        tp_mode: str = "fc"):
+                        tp_mode: str = "fc", checkpoint_path=None):
     """Launcher parity with reference ``run_tensor_parallel``
     (``tensor_parallel_train.py:327-385``; TP keeps the 400 s timeout base)."""
     return run_workers(tp_worker, world_size, epochs, sample_size, logs_dir,
                        timeout_base=400,
                        extra_args=(batch_size, backend, synthetic, lr,
-                                   optimizer_name, tp_mode))
+                                   optimizer_name, tp_mode,
+                                   checkpoint_path))
 
 
 def main():

@@ -157,3 +157,29 @@ def test_hybrid_entrypoint_checkpoint_resume(tmp_path):
         df = pd.read_csv(f"{logs2}/worker_{r}_samples_32.csv")
         assert list(df["epoch"]) == [2, 3], \
             f"rank {r} resumed wrong epochs: {list(df['epoch'])}"
+
+
+def test_tp_entrypoint_checkpoint_resume(tmp_path):
+    """Tensor-parallel checkpoint: per-rank files (shard params differ by
+    rank); resume continues the epoch sequence on both ranks."""
+    import os
+
+    import pandas as pd
+
+    from tensor_parallel_train import run_tensor_parallel
+
+    logs = str(tmp_path / "logs")
+    ckpt = str(tmp_path / "tp.ckpt")
+    run_tensor_parallel(world_size=2, epochs=1, sample_size=32,
+                        logs_dir=logs, batch_size=16, backend="gloo",
+                        synthetic=True, checkpoint_path=ckpt)
+    for r in range(2):
+        assert os.path.isfile(f"{ckpt}.rank{r}"), f"missing rank{r} file"
+    logs2 = str(tmp_path / "logs2")
+    run_tensor_parallel(world_size=2, epochs=3, sample_size=32,
+                        logs_dir=logs2, batch_size=16, backend="gloo",
+                        synthetic=True, checkpoint_path=ckpt)
+    for r in range(2):
+        df = pd.read_csv(f"{logs2}/worker_{r}_samples_32.csv")
+        assert list(df["epoch"]) == [2, 3], \
+            f"rank {r} resumed wrong epochs: {list(df['epoch'])}"
