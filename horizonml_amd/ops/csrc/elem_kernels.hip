@@ -1,6 +1,7 @@
 // BatchNorm / pooling / classifier / loss / optimizer kernels for gfx950.
 // All activation tensors are bf16 NHWC (flattened [M][C], channels innermost,
 // coalesced along C); statistics and parameters are f32.
+#include <cstdio>
 #include <cstdlib>
 
 #include "common.h"
@@ -1991,8 +1992,13 @@ void launch_linear_bwd(const float* dy, const void* x, const void* w,
   else if (Out <= 16 && !g_det_kernels) {
     int bsplit = (B + 127) / 128;
     if (!accum) {  // fresh output buffers: atomics need zeroed targets
-      hipMemsetAsync(dw, 0, sizeof(float) * (long)Out * In, st);
-      if (db != nullptr) hipMemsetAsync(db, 0, sizeof(float) * Out, st);
+      hipError_t e1 = hipMemsetAsync(dw, 0, sizeof(float) * (long)Out * In,
+                                     st);
+      hipError_t e2 = (db != nullptr)
+          ? hipMemsetAsync(db, 0, sizeof(float) * Out, st) : hipSuccess;
+      if (e1 != hipSuccess || e2 != hipSuccess)
+        fprintf(stderr, "[horizonml] linear-wgrad memset failed: %s\n",
+                hipGetErrorString(e1 != hipSuccess ? e1 : e2));
     }
     dim3 grid((In + 255) / 256, bsplit);
     k_linear_bwd_dw_bsplit<<<grid, 256, 0, st>>>(dy, (const bf16*)x, dw,
