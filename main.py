@@ -39,12 +39,8 @@ def run_benchmarks(sample_sizes, world_size, epochs, synthetic=None,
                  "tensor_parallel_logs")]:
             print(f"=== {name} @ {n} samples ===", flush=True)
             try:
-                if name == "tensor_parallel":
-                    df = runner(world_size, epochs, n, logs,
-                                backend=backend, synthetic=synthetic)
-                else:
-                    df = runner(world_size, epochs, n, logs,
-                                backend=backend, synthetic=synthetic)
+                df = runner(world_size, epochs, n, logs,
+                            backend=backend, synthetic=synthetic)
             except Exception:  # noqa: BLE001 — strategy tolerance (main.py:44)
                 traceback.print_exc()
                 df = None
