@@ -157,3 +157,27 @@ def test_bucketed_equals_single_buffer_gloo():
     for rank, exact, maxdiff in results:
         assert exact, (f"rank {rank}: bucketed != single-buffer "
                        f"(max diff {maxdiff})")
+
+
+def test_partition_unit_sizes_edges():
+    from horizonml_amd.parallel.flat_reducer import partition_unit_sizes
+    # more buckets than units: clamps to one unit per bucket
+    assert partition_unit_sizes([5, 5], 8) == [1, 1]
+    # single unit
+    assert partition_unit_sizes([42], 4) == [1]
+    # skewed sizes still tile exactly
+    sizes = [100, 1, 1, 1, 1, 1]
+    counts = partition_unit_sizes(sizes, 3)
+    assert sum(counts) == len(sizes) and all(c >= 1 for c in counts)
+    # near-equal split of equal sizes
+    assert partition_unit_sizes([10] * 8, 4) == [2, 2, 2, 2]
+
+
+def test_pipeline_chunk_sizes_edges():
+    import torch
+
+    from horizonml_amd.parallel.pipeline import chunk_sizes
+    for n, m in [(10, 3), (3, 10), (16, 4), (1, 1), (7, 2)]:
+        assert chunk_sizes(n, m) == [c.shape[0] for c in
+                                     torch.arange(n).chunk(m)], (n, m)
+    assert chunk_sizes(0, 4) == []
