@@ -181,3 +181,40 @@ def test_pipeline_chunk_sizes_edges():
         assert chunk_sizes(n, m) == [c.shape[0] for c in
                                      torch.arange(n).chunk(m)], (n, m)
     assert chunk_sizes(0, 4) == []
+
+
+def test_partition_unit_sizes_invariants_fuzz():
+    from hypothesis import given, settings
+    from hypothesis import strategies as st
+
+    from horizonml_amd.parallel.flat_reducer import partition_unit_sizes
+
+    @settings(max_examples=200, deadline=None)
+    @given(st.lists(st.integers(min_value=1, max_value=10_000_000),
+                    min_size=1, max_size=64),
+           st.integers(min_value=1, max_value=32))
+    def check(sizes, n_buckets):
+        counts = partition_unit_sizes(sizes, n_buckets)
+        assert sum(counts) == len(sizes)
+        assert len(counts) == min(n_buckets, len(sizes))
+        assert all(c >= 1 for c in counts)
+
+    check()
+
+
+def test_chunk_sizes_matches_torch_fuzz():
+    import torch
+    from hypothesis import given, settings
+    from hypothesis import strategies as st
+
+    from horizonml_amd.parallel.pipeline import chunk_sizes
+
+    @settings(max_examples=200, deadline=None)
+    @given(st.integers(min_value=0, max_value=4096),
+           st.integers(min_value=1, max_value=64))
+    def check(n, m):
+        expect = ([c.shape[0] for c in torch.arange(n).chunk(m)]
+                  if n > 0 else [])
+        assert chunk_sizes(n, m) == expect
+
+    check()
