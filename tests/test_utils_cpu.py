@@ -100,3 +100,23 @@ def test_checkpoint_atomic_tmp(tmp_path):
     save_checkpoint(path, m, epoch=1)
     assert os.path.isfile(path)
     assert not os.path.exists(path + ".tmp")
+
+
+def test_split_counts_invariants_fuzz():
+    from hypothesis import given, settings
+    from hypothesis import strategies as st
+
+    from horizonml_amd.models.partition import split_counts
+
+    @settings(max_examples=200, deadline=None)
+    @given(st.integers(min_value=0, max_value=512),
+           st.integers(min_value=1, max_value=64))
+    def check(n_units, n_stages):
+        counts = split_counts(n_units, n_stages)
+        assert sum(counts) == n_units
+        assert len(counts) == n_stages
+        # balanced: max-min <= 1, and the remainder loads the front
+        assert max(counts) - min(counts) <= 1
+        assert counts == sorted(counts, reverse=True)
+
+    check()
