@@ -155,3 +155,21 @@ def test_dp_entrypoint_end_to_end(tmp_path):
     combined = os.path.join(str(tmp_path), "combined_results_64.csv")
     assert os.path.isfile(combined)
     assert pd.read_csv(combined).shape[0] == 2
+
+
+def test_dp_entrypoint_reproducible(tmp_path):
+    """Two identical CPU runs are bitwise-identical in the metrics that
+    depend only on math (loss/accuracy): seeded init + shared synthetic
+    subset + deterministic fp32 CPU ops — the reference's unseeded
+    per-rank randperm made runs unrepeatable."""
+    import pandas as pd
+
+    from data_parallel_train import run_data_parallel
+    dfs = []
+    for tag in ("a", "b"):
+        logs = str(tmp_path / tag)
+        run_data_parallel(2, 2, 64, logs, batch_size=16, synthetic=True,
+                          backend="gloo")
+        dfs.append(pd.read_csv(f"{logs}/worker_0_samples_64.csv"))
+    assert list(dfs[0]["loss"]) == list(dfs[1]["loss"])
+    assert list(dfs[0]["accuracy"]) == list(dfs[1]["accuracy"])
